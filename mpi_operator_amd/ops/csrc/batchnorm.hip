@@ -1,0 +1,282 @@
+// Fused BatchNorm(+ReLU) for NHWC bf16 activations, fp32 statistics.
+// Training fwd = {partial per-channel sum/sumsq (deterministic, no atomics)
+// → finalize (mean/invstd/scale/shift) → normalize+ReLU apply}.
+// Backward = {partial sum(dy·mask), sum(dy·mask·xhat) → finalize coeffs →
+// elementwise dx}. ReLU is folded into both directions (mask from y>0).
+// Reference op being replaced: the external image's cuDNN BN+ReLU
+// (SURVEY.md §2.3 N7).
+#include "common.h"
+
+// Per-thread fixed channel-group ownership: thread t owns channel block
+// cb = t % C8 and row-lane t / C8 — consecutive lanes read consecutive
+// 16 B groups (fully coalesced); accumulation stays in registers.
+// Requires C8 = C/8 <= 256 (C <= 2048 — every ResNet/BERT channel width).
+template <int WHAT> // 0: fwd stats (sum, sumsq); 1: bwd stats (dy*m, dy*m*xhat)
+__global__ void bn_partials_k(const ushort8 *__restrict__ x,
+                              const ushort8 *__restrict__ dy,
+                              const ushort8 *__restrict__ y,
+                              const float *__restrict__ mean,
+                              const float *__restrict__ invstd,
+                              float *__restrict__ partial, // [grid][2][C]
+                              long M, int C8, int relu) {
+  int C = C8 * 8;
+  int cb = threadIdx.x % C8;
+  int row_lane = threadIdx.x / C8;
+  int rows_per_block = blockDim.x / C8;
+  float a0[8] = {0}, a1[8] = {0};
+  float mn[8], is[8];
+  if (WHAT == 1 && row_lane < rows_per_block) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mn[j] = mean[cb * 8 + j];
+      is[j] = invstd[cb * 8 + j];
+    }
+  }
+  if (row_lane < rows_per_block) {
+    for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
+         row += (long)gridDim.x * rows_per_block) {
+      long off = row * C8 + cb;
+      float fx[8];
+      ushort8 vx = x[off];
+      bf8_to_f8(vx, fx);
+      if (WHAT == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          a0[j] += fx[j];            // fx here is x
+          a1[j] += fx[j] * fx[j];
+        }
+      } else {
+        ushort8 vdy = dy[off];
+        float fdy[8];
+        bf8_to_f8(vdy, fdy);
+        if (relu) {
+          ushort8 vy = y[off];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = (fx[j] - mn[j]) * is[j];
+          a0[j] += fdy[j];
+          a1[j] += fdy[j] * xhat;
+        }
+      }
+    }
+  }
+  // block reduce across row lanes via LDS
+  __shared__ float lds[2][256 * 8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    lds[0][threadIdx.x * 8 + j] = a0[j];
+    lds[1][threadIdx.x * 8 + j] = a1[j];
+  }
+  __syncthreads();
+  if (row_lane == 0) {
+    for (int rl = 1; rl < rows_per_block; ++rl) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        a0[j] += lds[0][(rl * C8 + cb) * 8 + j];
+        a1[j] += lds[1][(rl * C8 + cb) * 8 + j];
+      }
+    }
+    float *p0 = partial + (long)blockIdx.x * 2 * C + cb * 8;
+    float *p1 = p0 + C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      p0[j] = a0[j];
+      p1[j] = a1[j];
+    }
+  }
+}
+
+// finalize fwd: reduce partials → mean/invstd + scale/shift
+__global__ void bn_finalize_fwd_k(const float *__restrict__ partial, int grid,
+                                  int C, const float *__restrict__ gamma,
+                                  const float *__restrict__ beta, float inv_m,
+                                  float eps, float *__restrict__ mean,
+                                  float *__restrict__ invstd,
+                                  float *__restrict__ scale,
+                                  float *__restrict__ shift) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0, sq = 0;
+  for (int g = 0; g < grid; ++g) {
+    s += partial[(long)g * 2 * C + c];
+    sq += partial[(long)g * 2 * C + C + c];
+  }
+  float mu = s * inv_m;
+  float var = fmaxf(sq * inv_m - mu * mu, 0.f);
+  float is = rsqrtf(var + eps);
+  mean[c] = mu;
+  invstd[c] = is;
+  float sc = gamma[c] * is;
+  scale[c] = sc;
+  shift[c] = beta[c] - mu * sc;
+}
+
+// finalize bwd: dbeta/dgamma + the three per-channel dx coefficients
+__global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
+                                  int C, const float *__restrict__ gamma,
+                                  const float *__restrict__ invstd, float inv_m,
+                                  float *__restrict__ dbeta,
+                                  float *__restrict__ dgamma,
+                                  float *__restrict__ k1, // gamma*invstd
+                                  float *__restrict__ k2, // k1*dbeta/m
+                                  float *__restrict__ k3) { // k1*dgamma/m*invstd... see apply
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s0 = 0, s1 = 0;
+  for (int g = 0; g < grid; ++g) {
+    s0 += partial[(long)g * 2 * C + c];
+    s1 += partial[(long)g * 2 * C + C + c];
+  }
+  dbeta[c] = s0;
+  dgamma[c] = s1;
+  float g_is = gamma[c] * invstd[c];
+  k1[c] = g_is;
+  k2[c] = g_is * s0 * inv_m;
+  k3[c] = g_is * s1 * inv_m; // multiplied by xhat in apply
+}
+
+// apply scale/shift (+ReLU): fwd-train, fwd-eval share this
+__global__ void bn_apply_k(const ushort8 *__restrict__ x,
+                           const float *__restrict__ scale,
+                           const float *__restrict__ shift,
+                           ushort8 *__restrict__ y, long M, int C8, int relu) {
+  int cb = threadIdx.x % C8;
+  int row_lane = threadIdx.x / C8;
+  int rows_per_block = blockDim.x / C8;
+  if (row_lane >= rows_per_block) return;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale[cb * 8 + j];
+    sh[j] = shift[cb * 8 + j];
+  }
+  for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
+       row += (long)gridDim.x * rows_per_block) {
+    long off = row * C8 + cb;
+    ushort8 v = x[off];
+    float f[8];
+    bf8_to_f8(v, f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      f[j] = f[j] * sc[j] + sh[j];
+      if (relu) f[j] = fmaxf(f[j], 0.f);
+    }
+    y[off] = f8_to_bf8(f);
+  }
+}
+
+// dx = k1*dy·mask - k2 - k3*xhat, xhat = (x-mean)*invstd
+__global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
+                               const ushort8 *__restrict__ x,
+                               const ushort8 *__restrict__ y,
+                               const float *__restrict__ mean,
+                               const float *__restrict__ invstd,
+                               const float *__restrict__ k1,
+                               const float *__restrict__ k2,
+                               const float *__restrict__ k3,
+                               ushort8 *__restrict__ dx, long M, int C8,
+                               int relu) {
+  int cb = threadIdx.x % C8;
+  int row_lane = threadIdx.x / C8;
+  int rows_per_block = blockDim.x / C8;
+  if (row_lane >= rows_per_block) return;
+  float mn[8], is[8], a[8], b[8], c3[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = cb * 8 + j;
+    mn[j] = mean[c];
+    is[j] = invstd[c];
+    a[j] = k1[c];
+    b[j] = k2[c];
+    c3[j] = k3[c];
+  }
+  for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
+       row += (long)gridDim.x * rows_per_block) {
+    long off = row * C8 + cb;
+    ushort8 vdy = dy[off], vx = x[off];
+    float fdy[8], fx[8];
+    bf8_to_f8(vdy, fdy);
+    bf8_to_f8(vx, fx);
+    if (relu) {
+      ushort8 vy = y[off];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xhat = (fx[j] - mn[j]) * is[j];
+      fdy[j] = a[j] * fdy[j] - b[j] - c3[j] * xhat;
+    }
+    dx[off] = f8_to_bf8(fdy);
+  }
+}
+
+static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
+  rows_per_block = 256 / C8;
+  long g = (M + rows_per_block - 1) / rows_per_block;
+  grid = (int)(g > 512 ? 512 : (g < 1 ? 1 : g));
+}
+
+extern "C" hipError_t bn_fwd_train_launch(
+    const void *x, const float *gamma, const float *beta, float eps, int relu,
+    void *y, float *mean, float *invstd, float *scale, float *shift,
+    float *partial, long M, int C, hipStream_t s) {
+  int C8 = C / 8;
+  if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
+  int grid, rpb;
+  bn_geom(M, C8, grid, rpb);
+  bn_partials_k<0><<<grid, 256, 0, s>>>((const ushort8 *)x, nullptr, nullptr,
+                                        nullptr, nullptr, partial, M, C8, 0);
+  HIP_KERNEL_CHECK();
+  bn_finalize_fwd_k<<<cdiv_h(C, 256), 256, 0, s>>>(partial, grid, C, gamma,
+                                                   beta, 1.f / (float)M, eps,
+                                                   mean, invstd, scale, shift);
+  HIP_KERNEL_CHECK();
+  bn_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)x, scale, shift,
+                                  (ushort8 *)y, M, C8, relu);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+extern "C" hipError_t bn_fwd_eval_launch(const void *x, const float *scale,
+                                         const float *shift, int relu, void *y,
+                                         long M, int C, hipStream_t s) {
+  int C8 = C / 8;
+  if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
+  int grid, rpb;
+  bn_geom(M, C8, grid, rpb);
+  bn_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)x, scale, shift,
+                                  (ushort8 *)y, M, C8, relu);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
+                                    const void *y, const float *gamma,
+                                    const float *mean, const float *invstd,
+                                    int relu, void *dx, float *dgamma,
+                                    float *dbeta, float *k1, float *k2,
+                                    float *k3, float *partial, long M, int C,
+                                    hipStream_t s) {
+  int C8 = C / 8;
+  if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
+  int grid, rpb;
+  bn_geom(M, C8, grid, rpb);
+  bn_partials_k<1><<<grid, 256, 0, s>>>((const ushort8 *)x, (const ushort8 *)dy,
+                                        (const ushort8 *)y, mean, invstd,
+                                        partial, M, C8, relu);
+  HIP_KERNEL_CHECK();
+  bn_finalize_bwd_k<<<cdiv_h(C, 256), 256, 0, s>>>(
+      partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
+  HIP_KERNEL_CHECK();
+  bn_bwd_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x,
+                                      (const ushort8 *)y, mean, invstd, k1, k2,
+                                      k3, (ushort8 *)dx, M, C8, relu);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}

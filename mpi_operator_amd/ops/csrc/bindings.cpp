@@ -1,0 +1,391 @@
+// PyTorch bindings for the gfx950 kernels. The only TU that includes torch
+// headers — kernel TUs are pure HIP and are linked via the extern "C"
+// launchers declared below.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+namespace {
+
+using at::Tensor;
+
+#define CHK(call)                                                              \
+  do {                                                                         \
+    hipError_t e_ = (call);                                                    \
+    TORCH_CHECK(e_ == hipSuccess, "HIP kernel failure: ", hipGetErrorString(e_)); \
+  } while (0)
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+using HIPDeviceGuard = c10::hip::HIPGuard;
+
+void check_cl_bf16(const Tensor &t, const char *name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast),
+              name, " must be 4-D channels_last");
+}
+
+Tensor empty_cl_bf16(int64_t n, int64_t c, int64_t h, int64_t w, const Tensor &like) {
+  return at::empty({n, c, h, w},
+                   like.options().dtype(at::kBFloat16))
+      .contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+} // namespace
+
+// ---- extern "C" launchers from the .hip TUs ----
+extern "C" {
+hipError_t add_relu_fwd(const void *, const void *, void *, long, hipStream_t);
+hipError_t add_relu_bwd(const void *, const void *, void *, long, hipStream_t);
+hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
+hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
+hipError_t bias_add(void *, const float *, long, int, hipStream_t);
+hipError_t colsum_bf16(const void *, float *, long, int, hipStream_t);
+struct SgdDesc {
+  const void *grad;
+  float *master;
+  float *mom;
+  uint16_t *out;
+  long numel;
+};
+hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
+                           hipStream_t);
+hipError_t bn_fwd_train_launch(const void *, const float *, const float *,
+                               float, int, void *, float *, float *, float *,
+                               float *, float *, long, int, hipStream_t);
+hipError_t bn_fwd_eval_launch(const void *, const float *, const float *, int,
+                              void *, long, int, hipStream_t);
+hipError_t bn_bwd_launch(const void *, const void *, const void *,
+                         const float *, const float *, const float *, int,
+                         void *, float *, float *, float *, float *, float *,
+                         float *, long, int, hipStream_t);
+hipError_t maxpool_fwd_launch(const void *, void *, uint8_t *, int, int, int,
+                              int, int, int, int, int, int, hipStream_t);
+hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
+                              int, int, int, int, int, int, int, hipStream_t);
+hipError_t softmax_xent_fwd_launch(const void *, const long *, float *, float *,
+                                   int, int, hipStream_t);
+hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
+                                   int, float, hipStream_t);
+hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
+                   long, long, int, hipStream_t);
+hipError_t transpose2d_bf16(const void *, void *, int, int, hipStream_t);
+hipError_t im2col_t(const void *, void *, int, int, int, int, int, int, int,
+                    int, int, int, hipStream_t);
+hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
+                    int, int, int, int, int, int, int, hipStream_t);
+hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
+                      int, int, int, int, int, int, int, hipStream_t);
+hipError_t conv_wgrad_gemm(const void *, const void *, float *, float *, int,
+                           int, long, long, int, hipStream_t);
+}
+
+// ------------------------- conv -------------------------
+static Tensor conv2d_fwd(const Tensor &x, const Tensor &w, int64_t stride,
+                         int64_t pad) {
+  check_cl_bf16(x, "x");
+  check_cl_bf16(w, "w");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kout = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C, "conv channel mismatch");
+  TORCH_CHECK(C % 8 == 0, "conv requires C%8==0 (pad the stem input)");
+  int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
+  Tensor y = empty_cl_bf16(N, Kout, HO, WO, x);
+  CHK(conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W, C, Kout, R,
+               S, (int)stride, (int)pad, HO, WO, cur_stream()));
+  return y;
+}
+
+static Tensor conv2d_dgrad(const Tensor &dy, const Tensor &w, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad) {
+  check_cl_bf16(dy, "dy");
+  check_cl_bf16(w, "w");
+  const HIPDeviceGuard guard(dy.device());
+  int N = dy.size(0), Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
+  int C = w.size(1), R = w.size(2), S = w.size(3);
+  int RSC = R * S * C;
+  Tensor wT = at::empty({RSC, Kout}, w.options());
+  CHK(transpose2d_bf16(w.data_ptr(), wT.data_ptr(), Kout, RSC, cur_stream()));
+  Tensor dx = empty_cl_bf16(N, C, H, W, dy);
+  CHK(conv_dgrad(dy.data_ptr(), wT.data_ptr(), dx.data_ptr(), N, (int)H,
+                 (int)W, C, Kout, R, S, (int)stride, (int)pad, HO, WO,
+                 cur_stream()));
+  return dx;
+}
+
+static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
+                           int64_t S, int64_t stride, int64_t pad) {
+  check_cl_bf16(x, "x");
+  check_cl_bf16(dy, "dy");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
+  long M = (long)N * HO * WO;
+  long M8 = (M + 7) / 8;
+  int RSC = (int)R * S * C;
+  auto bf = x.options();
+  Tensor dyT = at::empty({Kout, M}, bf);
+  CHK(transpose2d_bf16(dy.data_ptr(), dyT.data_ptr(), (int)M, Kout, cur_stream()));
+  Tensor PT = at::empty({(long)RSC, M8 * 8}, bf);
+  CHK(im2col_t(x.data_ptr(), PT.data_ptr(), N, H, W, C, HO, WO, (int)R, (int)S,
+               (int)stride, (int)pad, cur_stream()));
+  int tiles = ((Kout + 127) / 128) * ((RSC + 127) / 128);
+  int splits = std::min(std::max(768 / tiles, 1), 64);
+  Tensor partial = at::empty({(long)splits, (long)Kout, (long)RSC},
+                             x.options().dtype(at::kFloat));
+  // dw fp32 with channels_last semantics: memory [Kout][R][S][C]
+  Tensor dw = at::empty({(int64_t)Kout, (int64_t)C, R, S},
+                        x.options().dtype(at::kFloat))
+                  .contiguous(at::MemoryFormat::ChannelsLast);
+  CHK(conv_wgrad_gemm(dyT.data_ptr(), PT.data_ptr(), partial.data_ptr<float>(),
+                      dw.data_ptr<float>(), Kout, RSC, M, M8 * 8, splits,
+                      cur_stream()));
+  return dw;
+}
+
+// ------------------------- batchnorm -------------------------
+static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
+                                        const Tensor &beta, double eps,
+                                        bool relu) {
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long M = (long)N * H * W;
+  auto f32 = x.options().dtype(at::kFloat);
+  Tensor y = empty_cl_bf16(N, C, H, W, x);
+  Tensor mean = at::empty({C}, f32), invstd = at::empty({C}, f32);
+  Tensor scale = at::empty({C}, f32), shift = at::empty({C}, f32);
+  Tensor partial = at::empty({512L * 2 * C}, f32);
+  CHK(bn_fwd_train_launch(x.data_ptr(), gamma.data_ptr<float>(),
+                          beta.data_ptr<float>(), (float)eps, relu ? 1 : 0,
+                          y.data_ptr(), mean.data_ptr<float>(),
+                          invstd.data_ptr<float>(), scale.data_ptr<float>(),
+                          shift.data_ptr<float>(), partial.data_ptr<float>(),
+                          M, C, cur_stream()));
+  return {y, mean, invstd};
+}
+
+static Tensor bn_fwd_eval(const Tensor &x, const Tensor &scale,
+                          const Tensor &shift, bool relu) {
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  Tensor y = empty_cl_bf16(N, C, H, W, x);
+  CHK(bn_fwd_eval_launch(x.data_ptr(), scale.contiguous().data_ptr<float>(),
+                         shift.contiguous().data_ptr<float>(), relu ? 1 : 0,
+                         y.data_ptr(), (long)N * H * W, C, cur_stream()));
+  return y;
+}
+
+static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
+                                  const Tensor &y, const Tensor &gamma,
+                                  const Tensor &mean, const Tensor &invstd,
+                                  bool relu) {
+  check_cl_bf16(dy, "dy");
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long M = (long)N * H * W;
+  auto f32 = x.options().dtype(at::kFloat);
+  Tensor dx = empty_cl_bf16(N, C, H, W, x);
+  Tensor dgamma = at::empty({C}, f32), dbeta = at::empty({C}, f32);
+  Tensor k1 = at::empty({C}, f32), k2 = at::empty({C}, f32), k3 = at::empty({C}, f32);
+  Tensor partial = at::empty({512L * 2 * C}, f32);
+  CHK(bn_bwd_launch(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+                    gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                    invstd.data_ptr<float>(), relu ? 1 : 0, dx.data_ptr(),
+                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                    k1.data_ptr<float>(), k2.data_ptr<float>(),
+                    k3.data_ptr<float>(), partial.data_ptr<float>(), M, C,
+                    cur_stream()));
+  return {dx, dgamma, dbeta};
+}
+
+// ------------------------- pooling -------------------------
+static std::vector<Tensor> maxpool_fwd(const Tensor &x, int64_t K,
+                                       int64_t stride, int64_t pad) {
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int HO = (H + 2 * (int)pad - (int)K) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - (int)K) / (int)stride + 1;
+  Tensor y = empty_cl_bf16(N, C, HO, WO, x);
+  Tensor idx = at::empty({N, HO, WO, C}, x.options().dtype(at::kByte));
+  CHK(maxpool_fwd_launch(x.data_ptr(), y.data_ptr(), idx.data_ptr<uint8_t>(),
+                         N, H, W, HO, WO, C, (int)K, (int)stride, (int)pad,
+                         cur_stream()));
+  return {y, idx};
+}
+
+static Tensor maxpool_bwd(const Tensor &dy, const Tensor &idx, int64_t H,
+                          int64_t W, int64_t K, int64_t stride, int64_t pad) {
+  check_cl_bf16(dy, "dy");
+  const HIPDeviceGuard guard(dy.device());
+  int N = dy.size(0), C = dy.size(1), HO = dy.size(2), WO = dy.size(3);
+  Tensor dx = empty_cl_bf16(N, C, H, W, dy);
+  CHK(maxpool_bwd_launch(dy.data_ptr(), idx.data_ptr<uint8_t>(), dx.data_ptr(),
+                         N, (int)H, (int)W, HO, WO, C, (int)K, (int)stride,
+                         (int)pad, cur_stream()));
+  return dx;
+}
+
+// ------------------------- gap -------------------------
+static Tensor gap_fwd_b(const Tensor &x) {
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  Tensor y = at::empty({N, C}, x.options());
+  CHK(gap_fwd(x.data_ptr(), y.data_ptr(), N, HW, C, cur_stream()));
+  return y;
+}
+
+static Tensor gap_bwd_b(const Tensor &dy, int64_t H, int64_t W) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16);
+  const HIPDeviceGuard guard(dy.device());
+  Tensor dyc = dy.contiguous();
+  int N = dy.size(0), C = dy.size(1);
+  Tensor dx = empty_cl_bf16(N, C, H, W, dy);
+  CHK(gap_bwd(dyc.data_ptr(), dx.data_ptr(), N, (int)(H * W), C, cur_stream()));
+  return dx;
+}
+
+// ------------------------- linear -------------------------
+static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  const HIPDeviceGuard guard(x.device());
+  Tensor xc = x.contiguous(), wc = w.contiguous();
+  int M = xc.size(0), K = xc.size(1), N = wc.size(0);
+  Tensor y = at::empty({M, N}, xc.options());
+  CHK(gemm_nt(xc.data_ptr(), wc.data_ptr(), y.data_ptr(), M, N, K, K, K, N, 0,
+              cur_stream()));
+  CHK(bias_add(y.data_ptr(), b.contiguous().data_ptr<float>(), M, N, cur_stream()));
+  return y;
+}
+
+static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
+                                      const Tensor &w) {
+  const HIPDeviceGuard guard(x.device());
+  Tensor dyc = dy.contiguous(), xc = x.contiguous(), wc = w.contiguous();
+  int M = xc.size(0), K = xc.size(1), N = wc.size(0);
+  auto f32 = x.options().dtype(at::kFloat);
+  // dx = dy @ w: NT-GEMM(dy [M,N], wT [K,N])
+  Tensor wT = at::empty({K, N}, wc.options());
+  CHK(transpose2d_bf16(wc.data_ptr(), wT.data_ptr(), N, K, cur_stream()));
+  Tensor dx = at::empty({M, K}, xc.options());
+  CHK(gemm_nt(dyc.data_ptr(), wT.data_ptr(), dx.data_ptr(), M, K, N, N, N, K,
+              0, cur_stream()));
+  // dw = dy^T @ x: NT-GEMM(dyT [N,M], xT [K,M]) → fp32
+  Tensor dyT = at::empty({N, M}, dyc.options());
+  CHK(transpose2d_bf16(dyc.data_ptr(), dyT.data_ptr(), M, N, cur_stream()));
+  Tensor xT = at::empty({K, M}, xc.options());
+  CHK(transpose2d_bf16(xc.data_ptr(), xT.data_ptr(), M, K, cur_stream()));
+  Tensor dw = at::empty({N, K}, f32);
+  CHK(gemm_nt(dyT.data_ptr(), xT.data_ptr(), dw.data_ptr(), N, K, M, M, M, K,
+              1, cur_stream()));
+  Tensor db = at::empty({N}, f32);
+  CHK(colsum_bf16(dyc.data_ptr(), db.data_ptr<float>(), M, N, cur_stream()));
+  return {dx, dw, db};
+}
+
+// ------------------------- softmax xent -------------------------
+static std::vector<Tensor> softmax_xent_fwd(const Tensor &logits,
+                                            const Tensor &target) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16);
+  const HIPDeviceGuard guard(logits.device());
+  int B = logits.size(0), V = logits.size(1);
+  auto f32 = logits.options().dtype(at::kFloat);
+  Tensor probs = at::empty({B, V}, f32);
+  Tensor loss = at::empty({}, f32);
+  CHK(softmax_xent_fwd_launch(logits.data_ptr(), target.data_ptr<long>(),
+                              probs.data_ptr<float>(), loss.data_ptr<float>(),
+                              B, V, cur_stream()));
+  return {loss, probs};
+}
+
+static Tensor softmax_xent_bwd(const Tensor &probs, const Tensor &target,
+                               double scale) {
+  const HIPDeviceGuard guard(probs.device());
+  int B = probs.size(0), V = probs.size(1);
+  Tensor d = at::empty({B, V}, probs.options().dtype(at::kBFloat16));
+  CHK(softmax_xent_bwd_launch(probs.data_ptr<float>(), target.data_ptr<long>(),
+                              d.data_ptr(), B, V, (float)scale, cur_stream()));
+  return d;
+}
+
+// ------------------------- add-relu -------------------------
+static Tensor add_relu_fwd_b(const Tensor &a, const Tensor &b) {
+  check_cl_bf16(a, "a");
+  const HIPDeviceGuard guard(a.device());
+  Tensor y = at::empty_like(a);
+  CHK(add_relu_fwd(a.data_ptr(), b.data_ptr(), y.data_ptr(), a.numel(), cur_stream()));
+  return y;
+}
+
+static Tensor add_relu_bwd_b(const Tensor &dy, const Tensor &y) {
+  const HIPDeviceGuard guard(dy.device());
+  Tensor dx = at::empty_like(dy);
+  CHK(add_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(), dy.numel(), cur_stream()));
+  return dx;
+}
+
+// ------------------------- sgd -------------------------
+static void sgd_step(std::vector<Tensor> masters, std::vector<Tensor> grads,
+                     std::vector<Tensor> momenta, std::vector<Tensor> outs,
+                     double lr, double mu, double wd, bool nesterov) {
+  TORCH_CHECK(!masters.empty());
+  const HIPDeviceGuard guard(masters[0].device());
+  std::vector<SgdDesc> by_dtype[2]; // 0: f32 grads, 1: bf16 grads
+  for (size_t i = 0; i < masters.size(); ++i) {
+    SgdDesc d;
+    d.grad = grads[i].data_ptr();
+    d.master = masters[i].data_ptr<float>();
+    d.mom = momenta[i].data_ptr<float>();
+    d.out = outs[i].numel() > 0 ? (uint16_t *)outs[i].data_ptr() : nullptr;
+    d.numel = masters[i].numel();
+    by_dtype[grads[i].scalar_type() == at::kBFloat16 ? 1 : 0].push_back(d);
+  }
+  for (int t = 0; t < 2; ++t)
+    if (!by_dtype[t].empty())
+      CHK(sgd_step_launch(by_dtype[t].data(), (int)by_dtype[t].size(), t,
+                          (float)lr, (float)mu, (float)wd, nesterov ? 1 : 0,
+                          cur_stream()));
+}
+
+// ------------------------- raw gemm (tests) -------------------------
+static Tensor gemm_nt_b(const Tensor &a, const Tensor &b, bool c_f32) {
+  const HIPDeviceGuard guard(a.device());
+  Tensor ac = a.contiguous(), bc = b.contiguous();
+  int M = ac.size(0), K = ac.size(1), N = bc.size(0);
+  TORCH_CHECK(bc.size(1) == K);
+  Tensor c = at::empty({M, N}, ac.options().dtype(c_f32 ? at::kFloat : at::kBFloat16));
+  CHK(gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K, K, K, N,
+              c_f32 ? 1 : 0, cur_stream()));
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_eval", &bn_fwd_eval);
+  m.def("bn_bwd", &bn_bwd);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("gap_fwd", &gap_fwd_b);
+  m.def("gap_bwd", &gap_bwd_b);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_bwd", &linear_bwd);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("add_relu_fwd", &add_relu_fwd_b);
+  m.def("add_relu_bwd", &add_relu_bwd_b);
+  m.def("sgd_step", &sgd_step);
+  m.def("gemm_nt", &gemm_nt_b);
+}

@@ -1,0 +1,252 @@
+// Elementwise / multi-tensor kernels: fused residual add+ReLU (fwd/bwd),
+// global average pool (fwd/bwd), fused multi-tensor SGD-momentum.
+// All memory-bound: vectorized 16 B/lane bf16 accesses (guide G13),
+// grid-stride with capped grids (guide G11).
+#include "common.h"
+
+static inline int ew_grid(long lane_tasks, int block = 256) {
+  long blocks = (lane_tasks + block - 1) / block;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+// ---------------- add + relu ----------------
+__global__ void add_relu_fwd_k(const ushort8 *__restrict__ a,
+                               const ushort8 *__restrict__ b,
+                               ushort8 *__restrict__ y, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8 va = a[i], vb = b[i];
+    float fa[8], fb[8];
+    bf8_to_f8(va, fa);
+    bf8_to_f8(vb, fb);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) fa[j] = fmaxf(fa[j] + fb[j], 0.f);
+    y[i] = f8_to_bf8(fa);
+  }
+}
+
+extern "C" hipError_t add_relu_fwd(const void *a, const void *b, void *y,
+                                   long n, hipStream_t s) {
+  long n8 = n / 8;
+  add_relu_fwd_k<<<ew_grid(n8), 256, 0, s>>>((const ushort8 *)a,
+                                             (const ushort8 *)b, (ushort8 *)y, n8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+__global__ void add_relu_bwd_k(const ushort8 *__restrict__ dy,
+                               const ushort8 *__restrict__ y,
+                               ushort8 *__restrict__ dx, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8 vd = dy[i], vy = y[i];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = bf2f(vy[j]) > 0.f ? vd[j] : uint16_t(0);
+    dx[i] = o;
+  }
+}
+
+extern "C" hipError_t add_relu_bwd(const void *dy, const void *y, void *dx,
+                                   long n, hipStream_t s) {
+  long n8 = n / 8;
+  add_relu_bwd_k<<<ew_grid(n8), 256, 0, s>>>((const ushort8 *)dy,
+                                             (const ushort8 *)y, (ushort8 *)dx, n8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+// ---------------- global average pool (NHWC) ----------------
+// y[n, c] = mean over HW of x[n, hw, c]; consecutive lanes take consecutive
+// 8-channel groups → coalesced 16 B/lane reads at every hw step.
+__global__ void gap_fwd_k(const ushort8 *__restrict__ x, ushort8 *__restrict__ y,
+                          int N, int HW, int C8) {
+  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < (long)N * C8;
+       t += (long)gridDim.x * blockDim.x) {
+    int n = t / C8, cb = t % C8;
+    const ushort8 *row = x + (long)n * HW * C8 + cb;
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int hw = 0; hw < HW; ++hw) {
+      ushort8 v = row[(long)hw * C8];
+      float f[8];
+      bf8_to_f8(v, f);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += f[j];
+    }
+    float inv = 1.f / HW;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] *= inv;
+    y[t] = f8_to_bf8(acc);
+  }
+}
+
+extern "C" hipError_t gap_fwd(const void *x, void *y, int N, int HW, int C,
+                              hipStream_t s) {
+  int C8 = C / 8;
+  gap_fwd_k<<<ew_grid((long)N * C8), 256, 0, s>>>((const ushort8 *)x,
+                                                  (ushort8 *)y, N, HW, C8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+__global__ void gap_bwd_k(const ushort8 *__restrict__ dy, ushort8 *__restrict__ dx,
+                          int N, int HW, int C8, float inv_hw) {
+  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < (long)N * HW * C8;
+       t += (long)gridDim.x * blockDim.x) {
+    int cb = t % C8;
+    int n = t / ((long)HW * C8);
+    ushort8 v = dy[(long)n * C8 + cb];
+    float f[8];
+    bf8_to_f8(v, f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] *= inv_hw;
+    dx[t] = f8_to_bf8(f);
+  }
+}
+
+extern "C" hipError_t gap_bwd(const void *dy, void *dx, int N, int HW, int C,
+                              hipStream_t s) {
+  int C8 = C / 8;
+  gap_bwd_k<<<ew_grid((long)N * HW * C8), 256, 0, s>>>(
+      (const ushort8 *)dy, (ushort8 *)dx, N, HW, C8, 1.f / HW);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+// ---------------- fused multi-tensor SGD-momentum ----------------
+// One launch covers up to SGD_MAX_T tensors (descriptors in kernarg space —
+// apex multi_tensor_apply-style, no per-step device metadata uploads).
+// Updates, per element: g' = grad(+wd*master); mom = mu*mom + g';
+// step = nesterov ? g' + mu*mom : mom; master -= lr*step; out_bf16 = master.
+constexpr int SGD_MAX_T = 40;
+constexpr int SGD_BLOCK = 256;
+constexpr int SGD_EPB = SGD_BLOCK * 32; // elements per block (8192)
+
+struct SgdDesc {
+  const void *grad;
+  float *master;
+  float *mom;
+  uint16_t *out; // nullptr if the param itself is fp32 (master IS the param)
+  long numel;
+};
+
+struct SgdArgs {
+  SgdDesc d[SGD_MAX_T];
+  int first_block[SGD_MAX_T + 1]; // prefix of per-tensor block counts
+  int nt;
+};
+
+template <bool GRAD_BF16>
+__global__ void sgd_step_k(SgdArgs args, float lr, float mu, float wd,
+                           int nesterov) {
+  // find this block's tensor (nt <= 40: linear scan, wave-uniform)
+  int bid = blockIdx.x;
+  int t = 0;
+  while (t + 1 <= args.nt - 1 && bid >= args.first_block[t + 1]) ++t;
+  const SgdDesc &D = args.d[t];
+  long base = (long)(bid - args.first_block[t]) * SGD_EPB;
+  for (long i = base + threadIdx.x * 8L; i < min(base + SGD_EPB, D.numel);
+       i += SGD_BLOCK * 8L) {
+    bool full = (i + 8 <= D.numel);
+    int m = full ? 8 : (int)(D.numel - i);
+    float g[8], mst[8], mm[8];
+    if (GRAD_BF16) {
+      const uint16_t *gp = (const uint16_t *)D.grad + i;
+      if (full) {
+        ushort8 v = *(const ushort8 *)gp;
+        bf8_to_f8(v, g);
+      } else
+        for (int j = 0; j < m; ++j) g[j] = bf2f(gp[j]);
+    } else {
+      const float *gp = (const float *)D.grad + i;
+      if (full) {
+        float4v a = *(const float4v *)gp, b = *(const float4v *)(gp + 4);
+        g[0] = a[0]; g[1] = a[1]; g[2] = a[2]; g[3] = a[3];
+        g[4] = b[0]; g[5] = b[1]; g[6] = b[2]; g[7] = b[3];
+      } else
+        for (int j = 0; j < m; ++j) g[j] = gp[j];
+    }
+    for (int j = 0; j < m; ++j) {
+      mst[j] = D.master[i + j];
+      mm[j] = D.mom[i + j];
+      float gj = g[j] + wd * mst[j];
+      mm[j] = mu * mm[j] + gj;
+      float step = nesterov ? gj + mu * mm[j] : mm[j];
+      mst[j] -= lr * step;
+      D.master[i + j] = mst[j];
+      D.mom[i + j] = mm[j];
+    }
+    if (D.out) {
+      uint16_t *op = D.out + i;
+      if (full) {
+        *(ushort8 *)op = f8_to_bf8(mst);
+      } else
+        for (int j = 0; j < m; ++j) op[j] = f2bf(mst[j]);
+    }
+  }
+}
+
+extern "C" hipError_t sgd_step_launch(const SgdDesc *descs, int nt,
+                                      int grad_is_bf16, float lr, float mu,
+                                      float wd, int nesterov, hipStream_t s) {
+  for (int start = 0; start < nt; start += SGD_MAX_T) {
+    SgdArgs a;
+    a.nt = (nt - start < SGD_MAX_T) ? nt - start : SGD_MAX_T;
+    int blocks = 0;
+    for (int i = 0; i < a.nt; ++i) {
+      a.d[i] = descs[start + i];
+      a.first_block[i] = blocks;
+      blocks += (int)((a.d[i].numel + SGD_EPB - 1) / SGD_EPB);
+    }
+    a.first_block[a.nt] = blocks;
+    if (grad_is_bf16)
+      sgd_step_k<true><<<blocks, SGD_BLOCK, 0, s>>>(a, lr, mu, wd, nesterov);
+    else
+      sgd_step_k<false><<<blocks, SGD_BLOCK, 0, s>>>(a, lr, mu, wd, nesterov);
+    HIP_KERNEL_CHECK();
+  }
+  return hipSuccess;
+}
+
+// ---------------- linear-layer helpers ----------------
+__global__ void bias_add_k(ushort8 *__restrict__ y, const float *__restrict__ b,
+                           long M, int N8) {
+  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < M * N8;
+       t += (long)gridDim.x * blockDim.x) {
+    int nb = t % N8;
+    ushort8 v = y[t];
+    float f[8];
+    bf8_to_f8(v, f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] += b[nb * 8 + j];
+    y[t] = f8_to_bf8(f);
+  }
+}
+
+extern "C" hipError_t bias_add(void *y, const float *b, long M, int N,
+                               hipStream_t s) {
+  int N8 = N / 8;
+  bias_add_k<<<ew_grid(M * N8), 256, 0, s>>>((ushort8 *)y, b, M, N8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+// db[n] = sum_m dy[m][n] (small M — classifier head)
+__global__ void colsum_k(const uint16_t *__restrict__ dy, float *__restrict__ db,
+                         long M, int N) {
+  for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += gridDim.x * blockDim.x) {
+    float a = 0;
+    for (long m = 0; m < M; ++m) a += bf2f(dy[m * N + n]);
+    db[n] = a;
+  }
+}
+
+extern "C" hipError_t colsum_bf16(const void *dy, float *db, long M, int N,
+                                  hipStream_t s) {
+  colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}

@@ -1,0 +1,232 @@
+"""Autograd-integrated ops with HIP (gfx950) dispatch.
+
+Dispatch rule (see ``_backend``): CUDA/ROCm tensors run the hand-written HIP
+kernels — a missing extension raises; CPU tensors run the PyTorch reference.
+
+GPU layout/dtype contract (MI355X-first):
+  - activations: logical NCHW, channels-last memory (NHWC), bf16
+  - conv weights: logical [K,C,R,S], channels-last memory ([K][R][S][C]), bf16
+  - conv dgrad/wgrad: dx bf16, dw fp32 (accumulated exactly in fp32)
+  - BN gamma/beta and all per-channel stats: fp32
+
+Reference parity: these ops cover the tf_cnn_benchmarks ResNet hot path the
+reference delegates to external CUDA images.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import reference as ref
+from ._backend import hip_ext
+
+
+def _cl(t):  # channels_last view check for 4-D activations
+    return t.is_contiguous(memory_format=torch.channels_last)
+
+
+class Conv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride: int, padding: int):
+        ctx.stride, ctx.padding = stride, padding
+        ctx.save_for_backward(x, w)
+        if x.is_cuda:
+            assert _cl(x) and _cl(w), "conv2d: GPU tensors must be channels_last"
+            return hip_ext().conv2d_fwd(x, w, stride, padding)
+        return ref.conv2d_fwd(x, w, stride, padding)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        st, pad = ctx.stride, ctx.padding
+        dx = dw = None
+        if dy.is_cuda:
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            if ctx.needs_input_grad[0]:
+                dx = hip_ext().conv2d_dgrad(dy, w, x.shape[2], x.shape[3], st, pad)
+            if ctx.needs_input_grad[1]:
+                dw = hip_ext().conv2d_wgrad(x, dy, w.shape[2], w.shape[3], st, pad)
+                dw = dw.to(w.dtype)
+        else:
+            if ctx.needs_input_grad[0]:
+                dx = ref.conv2d_dgrad(dy, w, x.shape, st, pad)
+            if ctx.needs_input_grad[1]:
+                dw = ref.conv2d_wgrad(x, dy, w.shape, st, pad).to(w.dtype)
+        return dx, dw, None, None
+
+
+def conv2d(x, w, stride: int = 1, padding: int = 0):
+    return Conv2dFn.apply(x, w, stride, padding)
+
+
+class BNReLUFn(torch.autograd.Function):
+    """Fused BatchNorm(+ReLU) with batch stats (training mode)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps: float, relu: bool):
+        if x.is_cuda:
+            assert _cl(x), "bn_relu: GPU tensors must be channels_last"
+            y, mean, invstd = hip_ext().bn_fwd_train(x, gamma, beta, eps, relu)
+        else:
+            y, mean, invstd = ref.bn_relu_fwd_train(x, gamma, beta, eps, relu)
+        ctx.relu = relu
+        ctx.save_for_backward(x, y, gamma, mean, invstd)
+        ctx.mark_non_differentiable(mean, invstd)
+        return y, mean, invstd
+
+    @staticmethod
+    def backward(ctx, dy, _dmean=None, _dinvstd=None):
+        x, y, gamma, mean, invstd = ctx.saved_tensors
+        if dy.is_cuda:
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            dx, dgamma, dbeta = hip_ext().bn_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
+        else:
+            dx, dgamma, dbeta = ref.bn_relu_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
+        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, None
+
+
+def bn_relu_train(x, gamma, beta, eps: float = 1e-5, relu: bool = True):
+    return BNReLUFn.apply(x, gamma, beta, eps, relu)
+
+
+def bn_relu_eval(x, gamma, beta, running_mean, running_var, eps: float = 1e-5, relu: bool = True):
+    if x.is_cuda:
+        invstd = (running_var.float() + eps).rsqrt()
+        scale = gamma.float() * invstd
+        shift = beta.float() - running_mean.float() * scale
+        return hip_ext().bn_fwd_eval(x, scale, shift, relu)
+    return ref.bn_relu_fwd_eval(x, gamma, beta, running_mean, running_var, eps, relu)
+
+
+class MaxPool2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel: int, stride: int, padding: int):
+        ctx.kernel, ctx.stride, ctx.padding = kernel, stride, padding
+        ctx.x_shape = x.shape
+        if x.is_cuda:
+            assert _cl(x)
+            y, idx = hip_ext().maxpool_fwd(x, kernel, stride, padding)
+        else:
+            y, idx = ref.max_pool2d_fwd(x, kernel, stride, padding)
+        ctx.save_for_backward(idx)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        if dy.is_cuda:
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            dx = hip_ext().maxpool_bwd(dy, idx, ctx.x_shape[2], ctx.x_shape[3],
+                                       ctx.kernel, ctx.stride, ctx.padding)
+        else:
+            dx = ref.max_pool2d_bwd(dy, idx, ctx.x_shape, ctx.kernel, ctx.stride, ctx.padding)
+        return dx, None, None, None
+
+
+def max_pool2d(x, kernel: int = 3, stride: int = 2, padding: int = 1):
+    return MaxPool2dFn.apply(x, kernel, stride, padding)
+
+
+class GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.x_shape = x.shape
+        if x.is_cuda:
+            assert _cl(x)
+            return hip_ext().gap_fwd(x)
+        return ref.global_avg_pool_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if dy.is_cuda:
+            return hip_ext().gap_bwd(dy.contiguous(), ctx.x_shape[2], ctx.x_shape[3])
+        return ref.global_avg_pool_bwd(dy, ctx.x_shape)
+
+
+def global_avg_pool(x):
+    return GlobalAvgPoolFn.apply(x)
+
+
+class LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        if x.is_cuda:
+            return hip_ext().linear_fwd(x, w, b)
+        return ref.linear_fwd(x, w, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        if dy.is_cuda:
+            dy = dy.contiguous()
+            dx, dw, db = hip_ext().linear_bwd(dy, x, w)
+            return dx, dw.to(w.dtype), db
+        dx = dy @ w
+        dw = dy.transpose(0, 1).float() @ x.float()
+        db = dy.float().sum(0)
+        return dx, dw.to(w.dtype), db
+
+
+def linear(x, w, b):
+    return LinearFn.apply(x, w, b)
+
+
+class SoftmaxCrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        if logits.is_cuda:
+            loss, probs = hip_ext().softmax_xent_fwd(logits.contiguous(), target)
+        else:
+            loss, probs = ref.softmax_cross_entropy_fwd(logits, target)
+        ctx.save_for_backward(probs, target)
+        ctx.dtype = logits.dtype
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        probs, target = ctx.saved_tensors
+        scale = float(dloss)
+        if probs.is_cuda:
+            return hip_ext().softmax_xent_bwd(probs, target, scale).to(ctx.dtype), None
+        return ref.softmax_cross_entropy_bwd(probs, target, scale).to(ctx.dtype), None
+
+
+def softmax_cross_entropy(logits, target):
+    return SoftmaxCrossEntropyFn.apply(logits, target)
+
+
+def sgd_momentum_step(params32, grads, momenta, bf16_outs, lr, momentum, weight_decay, nesterov=False):
+    """Fused multi-tensor SGD: fp32 masters + momentum + bf16 working-copy refresh."""
+    if params32 and params32[0].is_cuda:
+        hip_ext().sgd_step(list(params32), list(grads), list(momenta),
+                           [o if o is not None else torch.empty(0) for o in bf16_outs],
+                           lr, momentum, weight_decay, nesterov)
+    else:
+        ref.sgd_momentum_step(params32, grads, momenta, bf16_outs, lr, momentum, weight_decay, nesterov)
+
+
+class AddReLUFn(torch.autograd.Function):
+    """Fused residual add + ReLU (the Bottleneck join — hot elementwise)."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        if a.is_cuda:
+            assert _cl(a) and _cl(b)
+            y = hip_ext().add_relu_fwd(a, b)
+        else:
+            y = torch.relu(a + b)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        if dy.is_cuda:
+            dx = hip_ext().add_relu_bwd(dy.contiguous(memory_format=torch.channels_last), y)
+        else:
+            dx = dy * (y > 0)
+        return dx, dx
+
+
+def add_relu(a, b):
+    return AddReLUFn.apply(a, b)

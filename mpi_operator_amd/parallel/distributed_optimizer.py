@@ -1,0 +1,168 @@
+"""DistributedOptimizer — Horovod-tensor-fusion equivalent over RCCL/xGMI.
+
+Reference behavior being rebuilt: Horovod's DistributedOptimizer wraps the
+local optimizer so that per-step gradients are fused into buckets and
+allreduced (sum → average) while backward is still producing later
+gradients (reference tensorflow_mnist.py:133).
+
+MI355X-first design decisions:
+  - Gradients live directly inside persistent flat bucket buffers
+    (``p.grad`` is a strided view of the bucket), so bucket assembly costs
+    zero copies and zero extra HBM traffic — important at 8 TB/s where an
+    extra grad read/write pass is pure loss.
+  - Buckets fire their ``all_reduce`` the moment their last grad
+    accumulates (post-accumulate-grad hooks), overlapping communication
+    with the rest of backward. RCCL runs the collective on its own HIP
+    stream; ``step()`` only waits on the handles.
+  - Default bucket size 28 MiB — a multiple of 7×4 MiB so ring/direct
+    shards split evenly over the 7 point-to-point xGMI links of an MI355X
+    (SURVEY.md §5.8), large enough to amortize per-collective launch cost,
+    small enough to overlap with backward.
+  - bf16 grads allreduce in bf16 (half the link bytes); fp32 stays fp32.
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+_XGMI_BUCKET_BYTES = 28 * 1024 * 1024
+
+
+class _Bucket:
+    __slots__ = ("params", "flat", "handle", "ready", "views")
+
+    def __init__(self):
+        self.params: list[torch.nn.Parameter] = []
+        self.flat: torch.Tensor | None = None
+        self.handle = None
+        self.ready = 0
+        self.views: list[torch.Tensor] = []
+
+
+def _strided_view(flat: torch.Tensor, offset: int, p: torch.Tensor) -> torch.Tensor:
+    """A view into ``flat`` with exactly p's (possibly channels-last) layout."""
+    return flat.as_strided(p.shape, p.stride(), storage_offset=offset)
+
+
+class DistributedOptimizer:
+    """Wraps any torch optimizer with overlapped bucketized grad allreduce.
+
+    Usage (Horovod-shaped):
+        opt = make_local_optimizer(model.parameters())
+        opt = DistributedOptimizer(opt, model.named_parameters())
+        ...
+        loss.backward(); opt.step(); opt.zero_grad()
+    """
+
+    def __init__(self, optimizer, named_parameters=None, bucket_bytes: int = _XGMI_BUCKET_BYTES,
+                 average: bool = True, process_group=None):
+        self.optimizer = optimizer
+        self.average = average
+        self.group = process_group
+        self._hooks = []
+        params = [p for g in optimizer.param_groups for p in g["params"] if p.requires_grad]
+        # reverse order ≈ backward completion order (output-side params first)
+        self._build_buckets(list(reversed(params)), bucket_bytes)
+        self._register_hooks()
+
+    # -- bucket construction ------------------------------------------------
+    def _build_buckets(self, params, bucket_bytes):
+        self.buckets: list[_Bucket] = []
+        self._param_bucket: dict[int, _Bucket] = {}
+        cur = _Bucket()
+        cur_bytes = 0
+        by_dtype: dict[torch.dtype, _Bucket] = {}
+
+        def flush(dtype):
+            nonlocal by_dtype
+            b = by_dtype.pop(dtype, None)
+            if b is not None and b.params:
+                self.buckets.append(b)
+
+        sizes: dict[int, int] = {}
+        for p in params:
+            dt = p.dtype
+            b = by_dtype.get(dt)
+            if b is None:
+                b = by_dtype[dt] = _Bucket()
+                sizes[id(b)] = 0
+            b.params.append(p)
+            self._param_bucket[id(p)] = b
+            sizes[id(b)] += p.numel() * p.element_size()
+            if sizes[id(b)] >= bucket_bytes:
+                flush(dt)
+        for dt in list(by_dtype):
+            flush(dt)
+
+        # allocate flats and wire p.grad views
+        for b in self.buckets:
+            n = sum(p.numel() for p in b.params)
+            p0 = b.params[0]
+            b.flat = torch.zeros(n, dtype=p0.dtype, device=p0.device)
+            off = 0
+            for p in b.params:
+                v = _strided_view(b.flat, off, p)
+                p.grad = v
+                b.views.append(v)
+                off += p.numel()
+
+    def _register_hooks(self):
+        for b in self.buckets:
+            for p in b.params:
+                h = p.register_post_accumulate_grad_hook(self._make_hook(b))
+                self._hooks.append(h)
+
+    def _make_hook(self, bucket: _Bucket):
+        def hook(_param):
+            bucket.ready += 1
+            if bucket.ready == len(bucket.params):
+                self._launch(bucket)
+        return hook
+
+    # -- comm ---------------------------------------------------------------
+    def _launch(self, bucket: _Bucket):
+        if dist.is_available() and dist.is_initialized():
+            bucket.handle = dist.all_reduce(bucket.flat, async_op=True, group=self.group)
+
+    def _world(self):
+        if dist.is_available() and dist.is_initialized():
+            return dist.get_world_size(self.group)
+        return 1
+
+    def synchronize(self):
+        """Wait for all in-flight bucket allreduces; average."""
+        w = self._world()
+        for b in self.buckets:
+            if b.ready != len(b.params) and b.ready > 0:
+                # backward did not produce every grad of this bucket — fire now
+                self._launch(b)
+            if b.handle is not None:
+                b.handle.wait()
+                b.handle = None
+            if self.average and w > 1:
+                b.flat.div_(w)
+            b.ready = 0
+
+    # -- optimizer protocol -------------------------------------------------
+    def step(self, closure=None):
+        self.synchronize()
+        return self.optimizer.step(closure)
+
+    def zero_grad(self, set_to_none: bool = False):
+        # grads are persistent bucket views — zero in place, never detach
+        for b in self.buckets:
+            b.flat.zero_()
+            b.ready = 0
+            for p, v in zip(b.params, b.views):
+                if p.grad is not v:
+                    p.grad = v  # re-attach if user code replaced it
+
+    @property
+    def param_groups(self):
+        return self.optimizer.param_groups
+
+    def state_dict(self):
+        return self.optimizer.state_dict()
+
+    def load_state_dict(self, sd):
+        self.optimizer.load_state_dict(sd)

@@ -1,0 +1,84 @@
+"""Training-loop driver: the in-repo equivalent of the reference's
+tf_cnn_benchmarks/Horovod step loop (reference README.md:96-143,
+tensorflow_mnist.py:165-171) — synthetic-data ResNet training with
+images/sec accounting, DistributedOptimizer overlap, and rank-0 logging."""
+from __future__ import annotations
+
+import time
+
+import torch
+
+from . import parallel as hvd
+from .optim import FusedSGD
+from .parallel import DistributedOptimizer
+
+
+class SyntheticImageData:
+    """Synthetic ImageNet-shaped data, random-init (no network access for
+    datasets — mirrors tf_cnn_benchmarks --data_name=synthetic)."""
+
+    def __init__(self, batch: int, image: int = 224, classes: int = 1000,
+                 device="cpu", dtype=torch.float32, channels_last=False, seed: int = 1234):
+        g = torch.Generator(device="cpu").manual_seed(seed + hvd.rank())
+        x = torch.rand(batch, 3, image, image, generator=g) * 2 - 1
+        y = torch.randint(0, classes, (batch,), generator=g)
+        self.x = x.to(device=device, dtype=dtype)
+        if channels_last:
+            self.x = self.x.contiguous(memory_format=torch.channels_last)
+        self.y = y.to(device)
+
+    def __iter__(self):
+        while True:
+            yield self.x, self.y
+
+
+def make_trainer(model, lr: float = 0.1, momentum: float = 0.9, weight_decay: float = 1e-4,
+                 bucket_bytes: int | None = None):
+    """LR scaled by world size, as the Horovod example does
+    (reference tensorflow_mnist.py:123-130)."""
+    opt = FusedSGD(model.parameters(), lr=lr * hvd.size(), momentum=momentum,
+                   weight_decay=weight_decay)
+    kw = {}
+    if bucket_bytes is not None:
+        kw["bucket_bytes"] = bucket_bytes
+    dopt = DistributedOptimizer(opt, model.named_parameters(), **kw)
+    hvd.broadcast_parameters(model, root_rank=0)
+    return dopt
+
+
+def train_step(model, dopt, x, y):
+    logits = model(x)
+    loss = model.loss(logits, y)
+    dopt.zero_grad()
+    loss.backward()
+    dopt.step()
+    return loss
+
+
+def timed_steps(model, dopt, data_iter, steps: int, warmup: int, device) -> dict:
+    """Run warmup+steps; barrier+sync bracketed timing of exactly `steps`."""
+    it = iter(data_iter)
+    use_cuda = torch.cuda.is_available() and str(device).startswith("cuda")
+    for _ in range(warmup):
+        x, y = next(it)
+        train_step(model, dopt, x, y)
+    hvd.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    loss = None
+    for _ in range(steps):
+        x, y = next(it)
+        loss = train_step(model, dopt, x, y)
+    hvd.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    # MAX over ranks (slowest rank defines the job's step time)
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if hvd.size() > 1:
+        import torch.distributed as dist
+        t = t.to("cuda") if use_cuda else t
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return {"elapsed": float(t.item()), "loss": float(loss.item()), "steps": steps}
