@@ -82,6 +82,7 @@ hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
                       int, int, int, int, int, int, int, hipStream_t);
 hipError_t conv_wgrad_gemm(const void *, const void *, float *, float *, int,
                            int, long, long, int, hipStream_t);
+hipError_t mfma_probe(const void *, const void *, float *, hipStream_t);
 }
 
 // ------------------------- conv -------------------------
@@ -388,4 +389,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_bwd", &add_relu_bwd_b);
   m.def("sgd_step", &sgd_step);
   m.def("gemm_nt", &gemm_nt_b);
+  m.def("mfma_probe", [](const Tensor &a, const Tensor &b) {
+    const HIPDeviceGuard guard(a.device());
+    Tensor d = at::empty({16, 16}, a.options().dtype(at::kFloat));
+    CHK(mfma_probe(a.contiguous().data_ptr(), b.contiguous().data_ptr(),
+                   d.data_ptr<float>(), cur_stream()));
+    return d;
+  });
 }

@@ -89,3 +89,26 @@ extern "C" hipError_t im2col_t(const void *x, void *pt, int N, int H, int W,
                                             pad, M8);
   return hipGetLastError();
 }
+
+// Single-wave MFMA layout probe: loads A[16][32], B[16][32] (row-major, the
+// assumed per-lane fragment mapping) and writes D[16][16] via the assumed
+// C-map. Feeding basis/unique-valued operands from the test reveals the true
+// hardware mapping if the assumption is wrong.
+__global__ void mfma_probe_k(const uint16_t *__restrict__ a,
+                             const uint16_t *__restrict__ b,
+                             float *__restrict__ d) {
+  int lane = threadIdx.x & 63;
+  bf16x8 af = us8_to_bf8v(*(const ushort8 *)(a + (lane & 15) * 32 + (lane >> 4) * 8));
+  bf16x8 bf_ = us8_to_bf8v(*(const ushort8 *)(b + (lane & 15) * 32 + (lane >> 4) * 8));
+  float4v acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf_, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    d[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+extern "C" hipError_t mfma_probe(const void *a, const void *b, float *d,
+                                 hipStream_t s) {
+  mfma_probe_k<<<1, 64, 0, s>>>((const uint16_t *)a, (const uint16_t *)b, d);
+  return hipGetLastError();
+}
