@@ -3,8 +3,8 @@
 // launchers declared below.
 #include <torch/extension.h>
 
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
 #include <vector>
@@ -19,8 +19,10 @@ using at::Tensor;
     TORCH_CHECK(e_ == hipSuccess, "HIP kernel failure: ", hipGetErrorString(e_)); \
   } while (0)
 
-hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
-using HIPDeviceGuard = c10::hip::HIPGuard;
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+using HIPDeviceGuard = c10::hip::HIPGuardMasqueradingAsCUDA;
 
 void check_cl_bf16(const Tensor &t, const char *name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
