@@ -100,6 +100,8 @@ def to_mi355x(model: nn.Module, device="cuda") -> nn.Module:
     for m in model.modules():
         if isinstance(m, (Conv2d, Linear)):
             m.to(torch.bfloat16)
+        if isinstance(m, Linear):
+            m.bias.data = m.bias.data.float()  # classifier bias stays fp32
     for m in model.modules():
         if isinstance(m, BatchNormReLU):
             m.float()

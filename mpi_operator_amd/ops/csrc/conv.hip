@@ -110,6 +110,9 @@ extern "C" hipError_t conv_wgrad_gemm(const void *dyT, const void *PT,
                                       float *partial, float *dw, int Kout,
                                       int RSC, long M, long ldPT, int splits,
                                       hipStream_t strm) {
+  int nk = (int)((M + BK - 1) / BK);
+  if (splits > nk) splits = nk > 0 ? nk : 1; // must match the launch's clamp:
+  // the reduce below must sum exactly the slabs the GEMM wrote.
   GemmLoader la{(const uint16_t *)dyT, Kout, M, (int)M};
   GemmLoader lb{(const uint16_t *)PT, RSC, ldPT, (int)M};
   hipError_t e = launch_nt_gemm(la, lb, partial, Kout, RSC, (int)M, RSC, true,

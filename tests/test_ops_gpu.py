@@ -96,15 +96,38 @@ def test_bn_eval():
 
 # ---------------- pooling / gap / add-relu ----------------
 def test_maxpool_fwd_bwd():
+    """y vs torch reference; idx checked by self-consistency (ties in bf16
+    make torch's argmax choice non-unique, so bwd is validated against a
+    python scatter using OUR indices)."""
     from mpi_operator_amd.ops import hip_ext
     ext = hip_ext()
-    x = rand_cl(2, 64, 15, 15, seed=30)
+    N, C, H, W = 2, 64, 15, 15
+    x = rand_cl(N, C, H, W, seed=30)
     y, idx = ext.maxpool_fwd(x, 3, 2, 1)
-    yr, idxr = ref.max_pool2d_fwd(x.float().cpu(), 3, 2, 1)
+    yr, _ = ref.max_pool2d_fwd(x.float().cpu(), 3, 2, 1)
     assert relerr(y.cpu(), yr) < 0.01
-    dy = rand_cl(*y.shape, seed=31)
-    dx = ext.maxpool_bwd(dy, idx, 15, 15, 3, 2, 1)
-    dxr = ref.max_pool2d_bwd(dy.float().cpu(), idxr, x.shape, 3, 2, 1)
+    HO = WO = (H + 2 - 3) // 2 + 1
+    # consistency: x at the recorded argmax equals y, and bwd scatters there
+    xp = torch.nn.functional.pad(x.float().cpu(), (1, 1, 1, 1), value=float("-inf"))
+    idx_c = idx.cpu().long()  # [N,HO,WO,C], values 0..8
+    dy = rand_cl(N, C, HO, WO, seed=31)
+    dxr = torch.zeros(N, C, H, W)
+    for n in range(N):
+        for ho in range(HO):
+            for wo in range(WO):
+                pos = idx_c[n, ho, wo]  # [C]
+                r, s = pos // 3, pos % 3
+                h = ho * 2 + r - 1
+                w = wo * 2 + s - 1
+                cvals = xp[n, torch.arange(C), h + 1, w + 1]
+                assert torch.allclose(cvals.to(torch.bfloat16).float(),
+                                      y.cpu()[n, :, ho, wo].float(), atol=1e-3)
+                valid = (h >= 0) & (h < H) & (w >= 0) & (w < W)
+                hs = h.clamp(0, H - 1)
+                ws = w.clamp(0, W - 1)
+                dxr[n, torch.arange(C)[valid], hs[valid], ws[valid]] += \
+                    dy.float().cpu()[n, torch.arange(C)[valid], ho, wo]
+    dx = ext.maxpool_bwd(dy, idx, H, W, 3, 2, 1)
     assert relerr(dx.cpu(), dxr) < 0.02
 
 
