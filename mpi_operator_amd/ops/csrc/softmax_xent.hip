@@ -70,8 +70,10 @@ extern "C" hipError_t softmax_xent_bwd_launch(const float *probs,
   long tasks = (long)B * V;
   long blocks = (tasks + 255) / 256;
   if (blocks > 2048) blocks = 2048;
+  // loss is a batch MEAN: dlogits = (p - onehot) * dloss / B
   softmax_xent_bwd_k<<<(int)blocks, 256, 0, s>>>(probs, target,
-                                                 (uint16_t *)dlogits, B, V, scale);
+                                                 (uint16_t *)dlogits, B, V,
+                                                 scale / B);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
