@@ -75,7 +75,7 @@ hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, float, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
                    long, long, int, hipStream_t);
-hipError_t transpose2d_bf16(const void *, void *, int, int, hipStream_t);
+hipError_t transpose2d_bf16(const void *, void *, int, int, long, hipStream_t);
 hipError_t im2col_t(const void *, void *, int, int, int, int, int, int, int,
                     int, int, int, hipStream_t);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
@@ -113,8 +113,9 @@ static Tensor conv2d_dgrad(const Tensor &dy, const Tensor &w, int64_t H,
   int N = dy.size(0), Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
   int C = w.size(1), R = w.size(2), S = w.size(3);
   int RSC = R * S * C;
+  TORCH_CHECK(Kout % 8 == 0, "dgrad requires out-channels %8==0");
   Tensor wT = at::empty({RSC, Kout}, w.options());
-  CHK(transpose2d_bf16(w.data_ptr(), wT.data_ptr(), Kout, RSC, cur_stream()));
+  CHK(transpose2d_bf16(w.data_ptr(), wT.data_ptr(), Kout, RSC, Kout, cur_stream()));
   Tensor dx = empty_cl_bf16(N, C, H, W, dy);
   CHK(conv_dgrad(dy.data_ptr(), wT.data_ptr(), dx.data_ptr(), N, (int)H,
                  (int)W, C, Kout, R, S, (int)stride, (int)pad, HO, WO,
@@ -133,8 +134,9 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   long M8 = (M + 7) / 8;
   int RSC = (int)R * S * C;
   auto bf = x.options();
-  Tensor dyT = at::empty({Kout, M}, bf);
-  CHK(transpose2d_bf16(dy.data_ptr(), dyT.data_ptr(), (int)M, Kout, cur_stream()));
+  Tensor dyT = at::zeros({Kout, M8 * 8}, bf); // zero pad: reduce dim %8
+  CHK(transpose2d_bf16(dy.data_ptr(), dyT.data_ptr(), (int)M, Kout, M8 * 8,
+                       cur_stream()));
   Tensor PT = at::empty({(long)RSC, M8 * 8}, bf);
   CHK(im2col_t(x.data_ptr(), PT.data_ptr(), N, H, W, C, HO, WO, (int)R, (int)S,
                (int)stride, (int)pad, cur_stream()));
@@ -147,7 +149,7 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
                         x.options().dtype(at::kFloat))
                   .contiguous(at::MemoryFormat::ChannelsLast);
   CHK(conv_wgrad_gemm(dyT.data_ptr(), PT.data_ptr(), partial.data_ptr<float>(),
-                      dw.data_ptr<float>(), Kout, RSC, M, M8 * 8, splits,
+                      dw.data_ptr<float>(), Kout, RSC, M8 * 8, M8 * 8, splits,
                       cur_stream()));
   return dw;
 }
@@ -264,6 +266,7 @@ static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
   const HIPDeviceGuard guard(x.device());
   Tensor xc = x.contiguous(), wc = w.contiguous();
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
+  TORCH_CHECK(K % 8 == 0, "linear requires in_features %8==0");
   Tensor y = at::empty({M, N}, xc.options());
   CHK(gemm_nt(xc.data_ptr(), wc.data_ptr(), y.data_ptr(), M, N, K, K, K, N, 0,
               cur_stream()));
@@ -277,20 +280,27 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   Tensor dyc = dy.contiguous(), xc = x.contiguous(), wc = w.contiguous();
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   auto f32 = x.options().dtype(at::kFloat);
-  // dx = dy @ w: NT-GEMM(dy [M,N], wT [K,N])
-  Tensor wT = at::empty({K, N}, wc.options());
-  CHK(transpose2d_bf16(wc.data_ptr(), wT.data_ptr(), N, K, cur_stream()));
+  // NT-GEMM reduce dims must be %8 (16 B load granule): zero-pad N and M.
+  long Np = (N + 7) / 8 * 8, Mp = (M + 7) / 8 * 8;
+  // dx = dy @ w: NT-GEMM(dy [M,Np], wT [K,Np])
+  Tensor wT = at::zeros({K, Np}, wc.options());
+  CHK(transpose2d_bf16(wc.data_ptr(), wT.data_ptr(), N, K, Np, cur_stream()));
+  Tensor dyp = dyc;
+  if (Np != N) {
+    dyp = at::zeros({M, Np}, dyc.options());
+    dyp.narrow(1, 0, N).copy_(dyc);
+  }
   Tensor dx = at::empty({M, K}, xc.options());
-  CHK(gemm_nt(dyc.data_ptr(), wT.data_ptr(), dx.data_ptr(), M, K, N, N, N, K,
-              0, cur_stream()));
-  // dw = dy^T @ x: NT-GEMM(dyT [N,M], xT [K,M]) → fp32
-  Tensor dyT = at::empty({N, M}, dyc.options());
-  CHK(transpose2d_bf16(dyc.data_ptr(), dyT.data_ptr(), M, N, cur_stream()));
-  Tensor xT = at::empty({K, M}, xc.options());
-  CHK(transpose2d_bf16(xc.data_ptr(), xT.data_ptr(), M, K, cur_stream()));
+  CHK(gemm_nt(dyp.data_ptr(), wT.data_ptr(), dx.data_ptr(), M, K, (int)Np, Np,
+              Np, K, 0, cur_stream()));
+  // dw = dy^T @ x: NT-GEMM(dyT [N,Mp], xT [K,Mp]) → fp32
+  Tensor dyT = at::zeros({N, Mp}, dyc.options());
+  CHK(transpose2d_bf16(dyc.data_ptr(), dyT.data_ptr(), M, N, Mp, cur_stream()));
+  Tensor xT = at::zeros({K, Mp}, xc.options());
+  CHK(transpose2d_bf16(xc.data_ptr(), xT.data_ptr(), M, K, Mp, cur_stream()));
   Tensor dw = at::empty({N, K}, f32);
-  CHK(gemm_nt(dyT.data_ptr(), xT.data_ptr(), dw.data_ptr(), N, K, M, M, M, K,
-              1, cur_stream()));
+  CHK(gemm_nt(dyT.data_ptr(), xT.data_ptr(), dw.data_ptr(), N, K, (int)Mp, Mp,
+              Mp, K, 1, cur_stream()));
   Tensor db = at::empty({N}, f32);
   CHK(colsum_bf16(dyc.data_ptr(), db.data_ptr<float>(), M, N, cur_stream()));
   return {dx, dw, db};
@@ -366,6 +376,7 @@ static Tensor gemm_nt_b(const Tensor &a, const Tensor &b, bool c_f32) {
   Tensor ac = a.contiguous(), bc = b.contiguous();
   int M = ac.size(0), K = ac.size(1), N = bc.size(0);
   TORCH_CHECK(bc.size(1) == K);
+  TORCH_CHECK(K % 8 == 0, "gemm_nt requires K%8==0");
   Tensor c = at::empty({M, N}, ac.options().dtype(c_f32 ? at::kFloat : at::kBFloat16));
   CHK(gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K, K, K, N,
               c_f32 ? 1 : 0, cur_stream()));

@@ -211,24 +211,16 @@ extern "C" hipError_t sgd_step_launch(const SgdDesc *descs, int nt,
 }
 
 // ---------------- linear-layer helpers ----------------
-__global__ void bias_add_k(ushort8 *__restrict__ y, const float *__restrict__ b,
-                           long M, int N8) {
-  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < M * N8;
-       t += (long)gridDim.x * blockDim.x) {
-    int nb = t % N8;
-    ushort8 v = y[t];
-    float f[8];
-    bf8_to_f8(v, f);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) f[j] += b[nb * 8 + j];
-    y[t] = f8_to_bf8(f);
-  }
+__global__ void bias_add_k(uint16_t *__restrict__ y, const float *__restrict__ b,
+                           long M, int N) {
+  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < M * N;
+       t += (long)gridDim.x * blockDim.x)
+    y[t] = f2bf(bf2f(y[t]) + b[t % N]);
 }
 
 extern "C" hipError_t bias_add(void *y, const float *b, long M, int N,
                                hipStream_t s) {
-  int N8 = N / 8;
-  bias_add_k<<<ew_grid(M * N8), 256, 0, s>>>((ushort8 *)y, b, M, N8);
+  bias_add_k<<<ew_grid(M * N), 256, 0, s>>>((uint16_t *)y, b, M, N);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

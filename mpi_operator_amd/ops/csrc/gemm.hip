@@ -10,9 +10,12 @@ extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, c_f32 != 0, s);
 }
 
-// bf16 2-D transpose: out[j][i] = in[i][j]. LDS 64x64 tile, padded rows.
+// bf16 2-D transpose: out[j][i] = in[i][j], output leading dim ldo >= R
+// (callers pad ldo to a multiple of 8 — the NT-GEMM's reduce-dim granule —
+// and pre-zero the output so pad columns contribute 0). LDS 64x64 tile.
 __global__ void transpose2d_k(const uint16_t *__restrict__ in,
-                              uint16_t *__restrict__ out, int R, int C) {
+                              uint16_t *__restrict__ out, int R, int C,
+                              long ldo) {
   __shared__ uint16_t t[64][65];
   int bi = blockIdx.x * 64, bj = blockIdx.y * 64;
   // 256 threads: load 64x64 (each thread 16 elems, 8-wide rows)
@@ -34,16 +37,17 @@ __global__ void transpose2d_k(const uint16_t *__restrict__ in,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int gi = bi + lc + j;
-        if (gi < R) out[(long)go * R + gi] = t[lc + j][rr];
+        if (gi < R) out[(long)go * ldo + gi] = t[lc + j][rr];
       }
     }
   }
 }
 
 extern "C" hipError_t transpose2d_bf16(const void *in, void *out, int R, int C,
-                                       hipStream_t s) {
+                                       long ldo, hipStream_t s) {
   dim3 grid((R + 63) / 64, (C + 63) / 64);
-  transpose2d_k<<<grid, 256, 0, s>>>((const uint16_t *)in, (uint16_t *)out, R, C);
+  transpose2d_k<<<grid, 256, 0, s>>>((const uint16_t *)in, (uint16_t *)out, R,
+                                     C, ldo);
   return hipGetLastError();
 }
 

@@ -1,43 +1,65 @@
-"""Ad-hoc on-GPU debugging of failing ops (softmax_xent)."""
+"""Pinpoint the step-2 divergence in the e2e ResNet50 train test: print the
+max-|grad| per layer each step and the first activation explosion."""
 import os
 import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: E402
 
-from mpi_operator_amd.ops import hip_ext  # noqa: E402
-from mpi_operator_amd.ops import reference as ref  # noqa: E402
+from mpi_operator_amd import models  # noqa: E402
+from mpi_operator_amd.optim import FusedSGD  # noqa: E402
 
 
 def main():
-    ext = hip_ext()
-    torch.manual_seed(41)
-    B, V = 64, 1000
-    logits = ((torch.rand(B, V, device="cuda") * 2 - 1) * 4).to(torch.bfloat16)
-    tgt = torch.randint(0, V, (B,), device="cuda")
-    loss, probs = ext.softmax_xent_fwd(logits, tgt)
-    lr_, pr = ref.softmax_cross_entropy_fwd(logits.float().cpu(), tgt.cpu())
-    print("gpu loss:", loss.item(), " cpu loss:", lr_.item())
-    perr = (probs.cpu() - pr).abs()
-    print("probs maxerr:", perr.max().item(), "at", divmod(perr.argmax().item(), V))
-    lf = logits.float().cpu()
-    row_ls = torch.logsumexp(lf, dim=1)
-    row_loss = row_ls - lf[torch.arange(B), tgt.cpu()]
-    print("cpu mean row loss:", row_loss.mean().item())
-    print("first 4 row losses cpu:", row_loss[:4].tolist())
-    # per-row gpu logsumexp from probs: p = e^(x-m)/sum → can't recover; instead
-    # run B=1 kernels to isolate
-    for b in range(3):
-        l1, p1 = ext.softmax_xent_fwd(logits[b:b + 1].contiguous(), tgt[b:b + 1])
-        print(f"row {b}: gpu {l1.item():.5f} cpu {row_loss[b].item():.5f}")
-    # small case
-    lg = torch.tensor([[1.0, 2.0, 3.0, 0.5] * 2], device="cuda").to(torch.bfloat16)
-    t0 = torch.tensor([2], device="cuda")
-    l2, p2 = ext.softmax_xent_fwd(lg, t0)
-    lr2, pr2 = ref.softmax_cross_entropy_fwd(lg.float().cpu(), t0.cpu())
-    print("small: gpu", l2.item(), "cpu", lr2.item())
-    print("small probs gpu:", p2.cpu().numpy())
-    print("small probs cpu:", pr2.numpy())
+    torch.manual_seed(50)
+    m = models.to_mi355x(models.resnet50(num_classes=100), "cuda")
+    m.train()
+    x = torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 100, (4,), device="cuda")
+    opt = FusedSGD(m.parameters(), lr=0.02, momentum=0.9)
+
+    acts = {}
+
+    def hook(name):
+        def h(_mod, _in, out):
+            o = out[0] if isinstance(out, tuple) else out
+            if torch.is_tensor(o):
+                acts[name] = o.float().abs().max().item()
+        return h
+
+    for name, mod in m.named_modules():
+        if len(list(mod.children())) == 0:
+            mod.register_forward_hook(hook(name))
+
+    for step in range(4):
+        opt.zero_grad()
+        loss = m.loss(m(x), y)
+        loss.backward()
+        bad_act = [(n, v) for n, v in acts.items() if v > 1e4 or v != v]
+        gmax = sorted(((p.grad.float().abs().max().item(), n)
+                       for n, p in m.named_parameters() if p.grad is not None),
+                      reverse=True)
+        pmax = sorted(((p.float().abs().max().item(), n)
+                       for n, p in m.named_parameters()), reverse=True)
+        print(f"step {step}: loss={float(loss):.4f} "
+              f"top grads={[(n.split('.')[-2:], round(v, 3)) for v, n in gmax[:4]]} "
+              f"top params={[(n.split('.')[-2:], round(v, 3)) for v, n in pmax[:3]]}")
+        if bad_act:
+            print("  BAD ACTS:", bad_act[:8])
+        opt.step()
+        if float(loss) != float(loss) or float(loss) > 1e6:
+            # rerun forward to find first exploding activation
+            acts.clear()
+            with torch.no_grad():
+                m(x)
+            first_bad = None
+            for n, v in acts.items():
+                if v > 1e4 or v != v:
+                    first_bad = (n, v)
+                    break
+            print("  first bad act after step:", first_bad)
+            break
 
 
 if __name__ == "__main__":
