@@ -90,21 +90,42 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
   }
 }
 
-// finalize fwd: reduce partials → mean/invstd + scale/shift
+// finalize: 8 lanes cooperate per channel (grid entries split across lanes,
+// shfl-reduced) — the serial-per-channel version was one-wave latency-bound
+// at 124 µs/call and 31% of the whole step.
+DEV_INLINE void lane8_sums(const float *__restrict__ partial, int grid, int C,
+                           int c, int lane8, float &s0, float &s1) {
+  s0 = 0.f;
+  s1 = 0.f;
+  for (int g = lane8; g < grid; g += 8) {
+    s0 += partial[(long)g * 2 * C + c];
+    s1 += partial[(long)g * 2 * C + C + c];
+  }
+#pragma unroll
+  for (int off = 4; off > 0; off >>= 1) {
+    s0 += __shfl_down(s0, off, 8);
+    s1 += __shfl_down(s1, off, 8);
+  }
+}
+
+// fwd: mean/invstd + scale/shift + fused running-stats update (saves ~5 eager
+// tensor ops per BN layer per step on the torch side).
 __global__ void bn_finalize_fwd_k(const float *__restrict__ partial, int grid,
                                   int C, const float *__restrict__ gamma,
                                   const float *__restrict__ beta, float inv_m,
                                   float eps, float *__restrict__ mean,
                                   float *__restrict__ invstd,
                                   float *__restrict__ scale,
-                                  float *__restrict__ shift) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+                                  float *__restrict__ shift,
+                                  float *__restrict__ running_mean,
+                                  float *__restrict__ running_var,
+                                  float momentum, float unbias) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = t / 8, lane8 = t % 8;
   if (c >= C) return;
-  float s = 0, sq = 0;
-  for (int g = 0; g < grid; ++g) {
-    s += partial[(long)g * 2 * C + c];
-    sq += partial[(long)g * 2 * C + C + c];
-  }
+  float s, sq;
+  lane8_sums(partial, grid, C, c, lane8, s, sq);
+  if (lane8 != 0) return;
   float mu = s * inv_m;
   float var = fmaxf(sq * inv_m - mu * mu, 0.f);
   float is = rsqrtf(var + eps);
@@ -113,9 +134,13 @@ __global__ void bn_finalize_fwd_k(const float *__restrict__ partial, int grid,
   float sc = gamma[c] * is;
   scale[c] = sc;
   shift[c] = beta[c] - mu * sc;
+  if (running_mean) {
+    running_mean[c] = running_mean[c] * (1.f - momentum) + mu * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + var * unbias * momentum;
+  }
 }
 
-// finalize bwd: dbeta/dgamma + the three per-channel dx coefficients
+// bwd: dbeta/dgamma + the three per-channel dx coefficients
 __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
                                   int C, const float *__restrict__ gamma,
                                   const float *__restrict__ invstd, float inv_m,
@@ -123,20 +148,19 @@ __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
                                   float *__restrict__ dgamma,
                                   float *__restrict__ k1, // gamma*invstd
                                   float *__restrict__ k2, // k1*dbeta/m
-                                  float *__restrict__ k3) { // k1*dgamma/m*invstd... see apply
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+                                  float *__restrict__ k3) { // k1*dgamma/m (×xhat in apply)
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = t / 8, lane8 = t % 8;
   if (c >= C) return;
-  float s0 = 0, s1 = 0;
-  for (int g = 0; g < grid; ++g) {
-    s0 += partial[(long)g * 2 * C + c];
-    s1 += partial[(long)g * 2 * C + C + c];
-  }
+  float s0, s1;
+  lane8_sums(partial, grid, C, c, lane8, s0, s1);
+  if (lane8 != 0) return;
   dbeta[c] = s0;
   dgamma[c] = s1;
   float g_is = gamma[c] * invstd[c];
   k1[c] = g_is;
   k2[c] = g_is * s0 * inv_m;
-  k3[c] = g_is * s1 * inv_m; // multiplied by xhat in apply
+  k3[c] = g_is * s1 * inv_m;
 }
 
 // apply scale/shift (+ReLU): fwd-train, fwd-eval share this
@@ -219,13 +243,14 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
-  grid = (int)(g > 512 ? 512 : (g < 1 ? 1 : g));
+  grid = (int)(g > 256 ? 256 : (g < 1 ? 1 : g));
 }
 
 extern "C" hipError_t bn_fwd_train_launch(
     const void *x, const float *gamma, const float *beta, float eps, int relu,
     void *y, float *mean, float *invstd, float *scale, float *shift,
-    float *partial, long M, int C, hipStream_t s) {
+    float *partial, float *running_mean, float *running_var, float momentum,
+    long M, int C, hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
@@ -233,9 +258,10 @@ extern "C" hipError_t bn_fwd_train_launch(
   bn_partials_k<0><<<grid, 256, 0, s>>>((const ushort8 *)x, nullptr, nullptr,
                                         nullptr, nullptr, partial, M, C8, 0);
   HIP_KERNEL_CHECK();
-  bn_finalize_fwd_k<<<cdiv_h(C, 256), 256, 0, s>>>(partial, grid, C, gamma,
-                                                   beta, 1.f / (float)M, eps,
-                                                   mean, invstd, scale, shift);
+  float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  bn_finalize_fwd_k<<<cdiv_h((long)C * 8, 256), 256, 0, s>>>(
+      partial, grid, C, gamma, beta, 1.f / (float)M, eps, mean, invstd, scale,
+      shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
   bn_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)x, scale, shift,
                                   (ushort8 *)y, M, C8, relu);
@@ -271,7 +297,7 @@ extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
                                         (const ushort8 *)y, mean, invstd,
                                         partial, M, C8, relu);
   HIP_KERNEL_CHECK();
-  bn_finalize_bwd_k<<<cdiv_h(C, 256), 256, 0, s>>>(
+  bn_finalize_bwd_k<<<cdiv_h((long)C * 8, 256), 256, 0, s>>>(
       partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
   HIP_KERNEL_CHECK();
   bn_bwd_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x,

@@ -32,9 +32,9 @@ void check_cl_bf16(const Tensor &t, const char *name) {
 }
 
 Tensor empty_cl_bf16(int64_t n, int64_t c, int64_t h, int64_t w, const Tensor &like) {
-  return at::empty({n, c, h, w},
-                   like.options().dtype(at::kBFloat16))
-      .contiguous(at::MemoryFormat::ChannelsLast);
+  // memory_format at allocation — .contiguous() on a fresh empty COPIES.
+  return at::empty({n, c, h, w}, like.options().dtype(at::kBFloat16),
+                   at::MemoryFormat::ChannelsLast);
 }
 
 } // namespace
@@ -58,7 +58,8 @@ hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
                            hipStream_t);
 hipError_t bn_fwd_train_launch(const void *, const float *, const float *,
                                float, int, void *, float *, float *, float *,
-                               float *, float *, long, int, hipStream_t);
+                               float *, float *, float *, float *, float, long,
+                               int, hipStream_t);
 hipError_t bn_fwd_eval_launch(const void *, const float *, const float *, int,
                               void *, long, int, hipStream_t);
 hipError_t bn_bwd_launch(const void *, const void *, const void *,
@@ -146,8 +147,8 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
                              x.options().dtype(at::kFloat));
   // dw fp32 with channels_last semantics: memory [Kout][R][S][C]
   Tensor dw = at::empty({(int64_t)Kout, (int64_t)C, R, S},
-                        x.options().dtype(at::kFloat))
-                  .contiguous(at::MemoryFormat::ChannelsLast);
+                        x.options().dtype(at::kFloat),
+                        at::MemoryFormat::ChannelsLast);
   CHK(conv_wgrad_gemm(dyT.data_ptr(), PT.data_ptr(), partial.data_ptr<float>(),
                       dw.data_ptr<float>(), Kout, RSC, M8 * 8, M8 * 8, splits,
                       cur_stream()));
@@ -157,7 +158,9 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
 // ------------------------- batchnorm -------------------------
 static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
                                         const Tensor &beta, double eps,
-                                        bool relu) {
+                                        bool relu, const Tensor &running_mean,
+                                        const Tensor &running_var,
+                                        double momentum) {
   check_cl_bf16(x, "x");
   const HIPDeviceGuard guard(x.device());
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
@@ -167,12 +170,15 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
   Tensor mean = at::empty({C}, f32), invstd = at::empty({C}, f32);
   Tensor scale = at::empty({C}, f32), shift = at::empty({C}, f32);
   Tensor partial = at::empty({512L * 2 * C}, f32);
+  float *rm = running_mean.defined() && running_mean.numel() == C
+                  ? running_mean.data_ptr<float>() : nullptr;
+  float *rv = rm ? running_var.data_ptr<float>() : nullptr;
   CHK(bn_fwd_train_launch(x.data_ptr(), gamma.data_ptr<float>(),
                           beta.data_ptr<float>(), (float)eps, relu ? 1 : 0,
                           y.data_ptr(), mean.data_ptr<float>(),
                           invstd.data_ptr<float>(), scale.data_ptr<float>(),
                           shift.data_ptr<float>(), partial.data_ptr<float>(),
-                          M, C, cur_stream()));
+                          rm, rv, (float)momentum, M, C, cur_stream()));
   return {y, mean, invstd};
 }
 

@@ -59,15 +59,28 @@ def conv2d(x, w, stride: int = 1, padding: int = 0):
 
 
 class BNReLUFn(torch.autograd.Function):
-    """Fused BatchNorm(+ReLU) with batch stats (training mode)."""
+    """Fused BatchNorm(+ReLU) with batch stats (training mode). On GPU the
+    finalize kernel also updates running_mean/var in place (buffers, no
+    autograd) — saving the eager per-layer stat updates."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps: float, relu: bool):
+    def forward(ctx, x, gamma, beta, eps: float, relu: bool,
+                running_mean=None, running_var=None, momentum: float = 0.1):
         if x.is_cuda:
             assert _cl(x), "bn_relu: GPU tensors must be channels_last"
-            y, mean, invstd = hip_ext().bn_fwd_train(x, gamma, beta, eps, relu)
+            empty = torch.empty(0)
+            y, mean, invstd = hip_ext().bn_fwd_train(
+                x, gamma, beta, eps, relu,
+                running_mean if running_mean is not None else empty,
+                running_var if running_var is not None else empty, momentum)
         else:
             y, mean, invstd = ref.bn_relu_fwd_train(x, gamma, beta, eps, relu)
+            if running_mean is not None:
+                with torch.no_grad():
+                    n = x.numel() / x.shape[1]
+                    var = invstd.pow(-2) - eps
+                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                    running_var.mul_(1 - momentum).add_(var * n / max(n - 1, 1), alpha=momentum)
         ctx.relu = relu
         ctx.save_for_backward(x, y, gamma, mean, invstd)
         ctx.mark_non_differentiable(mean, invstd)
@@ -81,11 +94,13 @@ class BNReLUFn(torch.autograd.Function):
             dx, dgamma, dbeta = hip_ext().bn_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
         else:
             dx, dgamma, dbeta = ref.bn_relu_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
-        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, None
+        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, None,
+                None, None, None)
 
 
-def bn_relu_train(x, gamma, beta, eps: float = 1e-5, relu: bool = True):
-    return BNReLUFn.apply(x, gamma, beta, eps, relu)
+def bn_relu_train(x, gamma, beta, eps: float = 1e-5, relu: bool = True,
+                  running_mean=None, running_var=None, momentum: float = 0.1):
+    return BNReLUFn.apply(x, gamma, beta, eps, relu, running_mean, running_var, momentum)
 
 
 def bn_relu_eval(x, gamma, beta, running_mean, running_var, eps: float = 1e-5, relu: bool = True):

@@ -47,13 +47,9 @@ class BatchNormReLU(nn.Module):
 
     def forward(self, x):
         if self.training:
-            y, mean, invstd = Fx.bn_relu_train(x, self.weight, self.bias, self.eps, self.relu)
-            with torch.no_grad():
-                var = invstd.float().pow(-2) - self.eps
-                n = x.numel() / x.shape[1]
-                unbiased = var * (n / max(n - 1, 1))
-                self.running_mean.mul_(1 - self.momentum).add_(mean.float(), alpha=self.momentum)
-                self.running_var.mul_(1 - self.momentum).add_(unbiased, alpha=self.momentum)
+            y, _, _ = Fx.bn_relu_train(x, self.weight, self.bias, self.eps,
+                                       self.relu, self.running_mean,
+                                       self.running_var, self.momentum)
             return y
         return Fx.bn_relu_eval(x, self.weight, self.bias, self.running_mean,
                                self.running_var, self.eps, self.relu)

@@ -65,11 +65,18 @@ def test_bn_fwd_bwd(C, relu):
     x = rand_cl(N, C, H, W, seed=20, scale=3.0)
     gamma = torch.rand(C, device="cuda") + 0.5
     beta = torch.randn(C, device="cuda") * 0.2
-    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, 1e-5, relu)
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, 1e-5, relu, rm, rv, 0.3)
     yr, mr, ir = ref.bn_relu_fwd_train(x.float().cpu(), gamma.cpu(), beta.cpu(), 1e-5, relu)
     assert relerr(mean.cpu(), mr) < 1e-3
     assert relerr(invstd.cpu(), ir) < 1e-3
     assert relerr(y.cpu(), yr) < 0.02
+    # fused running-stats update
+    n = N * H * W
+    var_b = ir.pow(-2) - 1e-5
+    assert relerr(rm.cpu(), 0.3 * mr) < 1e-3
+    assert relerr(rv.cpu(), 0.7 + 0.3 * var_b * n / (n - 1)) < 1e-3
 
     dy = rand_cl(N, C, H, W, seed=21)
     dx, dgamma, dbeta = ext.bn_bwd(dy, x, y, gamma, mean, invstd, relu)
