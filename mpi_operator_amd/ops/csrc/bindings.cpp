@@ -83,8 +83,8 @@ hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, hipStream_t);
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
                       int, int, int, int, int, int, int, hipStream_t);
-hipError_t conv_wgrad_gemm(const void *, const void *, float *, float *, int,
-                           int, long, long, int, hipStream_t);
+hipError_t conv_wgrad_gemm(const void *, const void *, float *, void *, int,
+                           int, long, long, int, int, hipStream_t);
 hipError_t mfma_probe(const void *, const void *, float *, hipStream_t);
 }
 
@@ -145,12 +145,13 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   int splits = std::min(std::max(768 / tiles, 1), 64);
   Tensor partial = at::empty({(long)splits, (long)Kout, (long)RSC},
                              x.options().dtype(at::kFloat));
-  // dw fp32 with channels_last semantics: memory [Kout][R][S][C]
+  // dw bf16 (fp32-accumulated in the split-K slabs, rounded once at the
+  // reduce) with channels_last memory [Kout][R][S][C]
   Tensor dw = at::empty({(int64_t)Kout, (int64_t)C, R, S},
-                        x.options().dtype(at::kFloat),
+                        x.options().dtype(at::kBFloat16),
                         at::MemoryFormat::ChannelsLast);
   CHK(conv_wgrad_gemm(dyT.data_ptr(), PT.data_ptr(), partial.data_ptr<float>(),
-                      dw.data_ptr<float>(), Kout, RSC, M8 * 8, M8 * 8, splits,
+                      dw.data_ptr(), Kout, RSC, M8 * 8, M8 * 8, splits, 1,
                       cur_stream()));
   return dw;
 }

@@ -86,30 +86,37 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *wT, void *dx,
   return launch_nt_gemm(la, lb, dx, (int)M, C, K, C, false, strm);
 }
 
+template <bool OUT_BF16>
 __global__ void splitk_reduce_k(const float *__restrict__ partial, int splits,
-                                long len, float *__restrict__ out) {
+                                long len, void *__restrict__ out) {
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < len;
        i += (long)gridDim.x * blockDim.x) {
     float a = 0;
     for (int s = 0; s < splits; ++s) a += partial[(long)s * len + i];
-    out[i] = a;
+    if (OUT_BF16)
+      ((uint16_t *)out)[i] = f2bf(a);
+    else
+      ((float *)out)[i] = a;
   }
 }
 
 extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
-                                    float *out, hipStream_t s) {
+                                    void *out, int out_bf16, hipStream_t s) {
   long blocks = (len + 255) / 256;
   if (blocks > 2048) blocks = 2048;
-  splitk_reduce_k<<<(int)blocks, 256, 0, s>>>(partial, splits, len, out);
+  if (out_bf16)
+    splitk_reduce_k<true><<<(int)blocks, 256, 0, s>>>(partial, splits, len, out);
+  else
+    splitk_reduce_k<false><<<(int)blocks, 256, 0, s>>>(partial, splits, len, out);
   return hipGetLastError();
 }
 
 // wgrad GEMM over transposed operands (dyT [Kout][M], PT [RSC][M8*8]):
 // partial fp32 slabs [splits][Kout][RSC] → splitk_reduce → dw fp32.
 extern "C" hipError_t conv_wgrad_gemm(const void *dyT, const void *PT,
-                                      float *partial, float *dw, int Kout,
+                                      float *partial, void *dw, int Kout,
                                       int RSC, long M, long ldPT, int splits,
-                                      hipStream_t strm) {
+                                      int dw_bf16, hipStream_t strm) {
   int nk = (int)((M + BK - 1) / BK);
   if (splits > nk) splits = nk > 0 ? nk : 1; // must match the launch's clamp:
   // the reduce below must sum exactly the slabs the GEMM wrote.
@@ -118,5 +125,5 @@ extern "C" hipError_t conv_wgrad_gemm(const void *dyT, const void *PT,
   hipError_t e = launch_nt_gemm(la, lb, partial, Kout, RSC, (int)M, RSC, true,
                                 strm, splits);
   if (e != hipSuccess) return e;
-  return splitk_reduce(partial, splits, (long)Kout * RSC, dw, strm);
+  return splitk_reduce(partial, splits, (long)Kout * RSC, dw, dw_bf16, strm);
 }
