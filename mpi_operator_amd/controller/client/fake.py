@@ -74,6 +74,7 @@ class FakeResourceClient(ResourceClient):
             obj.setdefault("apiVersion", self.gvr.api_version)
             obj.setdefault("kind", self.gvr.kind)
             self.p.store[key] = obj
+            self.p.broadcast(self.gvr, "ADDED", obj)
             return copy.deepcopy(obj)
 
     def update(self, namespace, obj):
@@ -89,6 +90,7 @@ class FakeResourceClient(ResourceClient):
             old = self.p.store[key]
             obj["metadata"]["resourceVersion"] = str(int(old["metadata"].get("resourceVersion", "0")) + 1)
             self.p.store[key] = obj
+            self.p.broadcast(self.gvr, "MODIFIED", obj)
             return copy.deepcopy(obj)
 
     def update_status(self, namespace, obj):
@@ -105,6 +107,7 @@ class FakeResourceClient(ResourceClient):
             cur["status"] = obj.get("status", {})
             cur["metadata"]["resourceVersion"] = str(int(cur["metadata"].get("resourceVersion", "0")) + 1)
             self.p.store[key] = cur
+            self.p.broadcast(self.gvr, "MODIFIED", cur)
             return copy.deepcopy(cur)
 
     def delete(self, namespace, name):
@@ -115,7 +118,39 @@ class FakeResourceClient(ResourceClient):
             key = self._key(namespace, name)
             if key not in self.p.store:
                 raise NotFound(f"{self.gvr.resource} {namespace}/{name}")
-            del self.p.store[key]
+            obj = self.p.store.pop(key)
+            self.p.broadcast(self.gvr, "DELETED", obj)
+
+    def watch(self, namespace, resource_version=None, timeout_s: int = 60):
+        """Watch-stream parity with RestResourceClient.watch: yields
+        {"type": ..., "object": ...} for every mutation, starting with
+        synthetic ADDED events for the current store contents (the
+        list+watch informer bootstrap)."""
+        q: "list" = []
+        cond = threading.Condition(self.p.lock)
+        with self.p.lock:
+            for (kres, ns, _), o in self.p.store.items():
+                if kres == self.gvr.resource + "." + self.gvr.group \
+                        and (namespace is None or ns == namespace):
+                    q.append({"type": "ADDED", "object": copy.deepcopy(o)})
+            self.p.watchers.setdefault(self.gvr, []).append((namespace, q, cond))
+        deadline = __import__("time").time() + timeout_s
+        try:
+            while True:
+                with cond:
+                    while not q:
+                        remaining = deadline - __import__("time").time()
+                        if remaining <= 0:
+                            return
+                        cond.wait(min(remaining, 0.2))
+                    ev = q.pop(0)
+                yield ev
+        finally:
+            with self.p.lock:
+                try:
+                    self.p.watchers[self.gvr].remove((namespace, q, cond))
+                except ValueError:
+                    pass
 
 
 class FakeKubeClient(KubeClient):
@@ -128,6 +163,19 @@ class FakeKubeClient(KubeClient):
         self.uid_counter = itertools.count(1)
         self.lock = threading.RLock()
         self._clients: dict[GVR, FakeResourceClient] = {}
+        # watch plumbing: gvr -> [(namespace filter, event list, condition)]
+        self.watchers: dict = {}
+
+    def broadcast(self, gvr, ev_type: str, obj: dict) -> None:
+        """Deliver a watch event to every subscriber of gvr (called with
+        self.lock held by the mutating resource client)."""
+        import mpi_operator_amd.controller.api.types as _t
+        ns = _t.namespace(obj)
+        for want_ns, q, cond in self.watchers.get(gvr, []):
+            if want_ns is None or want_ns == ns:
+                with cond:
+                    q.append({"type": ev_type, "object": copy.deepcopy(obj)})
+                    cond.notify_all()
 
     def resource(self, gvr: GVR) -> FakeResourceClient:
         if gvr not in self._clients:
