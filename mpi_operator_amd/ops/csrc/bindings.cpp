@@ -76,15 +76,18 @@ hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, float, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
                    long, long, int, hipStream_t);
+hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
+                      long, long, int, hipStream_t);
+hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
+                      long, long, int, hipStream_t);
 hipError_t transpose2d_bf16(const void *, void *, int, int, long, hipStream_t);
-hipError_t im2col_t(const void *, void *, int, int, int, int, int, int, int,
-                    int, int, int, hipStream_t);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, hipStream_t);
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
                       int, int, int, int, int, int, int, hipStream_t);
-hipError_t conv_wgrad_gemm(const void *, const void *, float *, void *, int,
-                           int, long, long, int, int, hipStream_t);
+hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
+                               int, int, int, int, int, int, int, int, int,
+                               int, int, int, int, hipStream_t);
 hipError_t mfma_probe(const void *, const void *, float *, hipStream_t);
 }
 
@@ -113,12 +116,9 @@ static Tensor conv2d_dgrad(const Tensor &dy, const Tensor &w, int64_t H,
   const HIPDeviceGuard guard(dy.device());
   int N = dy.size(0), Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
   int C = w.size(1), R = w.size(2), S = w.size(3);
-  int RSC = R * S * C;
   TORCH_CHECK(Kout % 8 == 0, "dgrad requires out-channels %8==0");
-  Tensor wT = at::empty({RSC, Kout}, w.options());
-  CHK(transpose2d_bf16(w.data_ptr(), wT.data_ptr(), Kout, RSC, Kout, cur_stream()));
   Tensor dx = empty_cl_bf16(N, C, H, W, dy);
-  CHK(conv_dgrad(dy.data_ptr(), wT.data_ptr(), dx.data_ptr(), N, (int)H,
+  CHK(conv_dgrad(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N, (int)H,
                  (int)W, C, Kout, R, S, (int)stride, (int)pad, HO, WO,
                  cur_stream()));
   return dx;
@@ -131,16 +131,7 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   const HIPDeviceGuard guard(x.device());
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
-  long M = (long)N * HO * WO;
-  long M8 = (M + 7) / 8;
   int RSC = (int)R * S * C;
-  auto bf = x.options();
-  Tensor dyT = at::zeros({Kout, M8 * 8}, bf); // zero pad: reduce dim %8
-  CHK(transpose2d_bf16(dy.data_ptr(), dyT.data_ptr(), (int)M, Kout, M8 * 8,
-                       cur_stream()));
-  Tensor PT = at::empty({(long)RSC, M8 * 8}, bf);
-  CHK(im2col_t(x.data_ptr(), PT.data_ptr(), N, H, W, C, HO, WO, (int)R, (int)S,
-               (int)stride, (int)pad, cur_stream()));
   int tiles = ((Kout + 127) / 128) * ((RSC + 127) / 128);
   int splits = std::min(std::max(768 / tiles, 1), 64);
   Tensor partial = at::empty({(long)splits, (long)Kout, (long)RSC},
@@ -150,9 +141,10 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   Tensor dw = at::empty({(int64_t)Kout, (int64_t)C, R, S},
                         x.options().dtype(at::kBFloat16),
                         at::MemoryFormat::ChannelsLast);
-  CHK(conv_wgrad_gemm(dyT.data_ptr(), PT.data_ptr(), partial.data_ptr<float>(),
-                      dw.data_ptr(), Kout, RSC, M8 * 8, M8 * 8, splits, 1,
-                      cur_stream()));
+  CHK(conv_wgrad_implicit(dy.data_ptr(), x.data_ptr(),
+                          partial.data_ptr<float>(), dw.data_ptr(), N, H, W, C,
+                          Kout, (int)R, (int)S, (int)stride, (int)pad, HO, WO,
+                          splits, 1, cur_stream()));
   return dw;
 }
 
@@ -277,7 +269,11 @@ static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
   Tensor y = at::empty({M, N}, xc.options());
   CHK(gemm_nt(xc.data_ptr(), wc.data_ptr(), y.data_ptr(), M, N, K, K, K, N, 0,
               cur_stream()));
-  CHK(bias_add(y.data_ptr(), b.contiguous().data_ptr<float>(), M, N, cur_stream()));
+  // bias kernel accumulates in fp32; accept bf16 bias (BERT) or fp32
+  // (classifier keeps an fp32 bias for exactness)
+  Tensor bf = b.scalar_type() == at::kFloat ? b.contiguous()
+                                            : b.to(at::kFloat).contiguous();
+  CHK(bias_add(y.data_ptr(), bf.data_ptr<float>(), M, N, cur_stream()));
   return y;
 }
 
@@ -287,27 +283,22 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   Tensor dyc = dy.contiguous(), xc = x.contiguous(), wc = w.contiguous();
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   auto f32 = x.options().dtype(at::kFloat);
-  // NT-GEMM reduce dims must be %8 (16 B load granule): zero-pad N and M.
-  long Np = (N + 7) / 8 * 8, Mp = (M + 7) / 8 * 8;
-  // dx = dy @ w: NT-GEMM(dy [M,Np], wT [K,Np])
-  Tensor wT = at::zeros({K, Np}, wc.options());
-  CHK(transpose2d_bf16(wc.data_ptr(), wT.data_ptr(), N, K, Np, cur_stream()));
+  // dx = dy @ w: A = dy [M][N] k-contiguous; B = w [N rows][K cols]
+  // k-strided (TN-staged in LDS — no transpose buffer). The NT operand's
+  // reduce dim must be %8 (16 B load granule): zero-pad dy's N if ragged.
+  long Np = (N + 7) / 8 * 8;
   Tensor dyp = dyc;
   if (Np != N) {
     dyp = at::zeros({M, Np}, dyc.options());
     dyp.narrow(1, 0, N).copy_(dyc);
   }
   Tensor dx = at::empty({M, K}, xc.options());
-  CHK(gemm_nt(dyp.data_ptr(), wT.data_ptr(), dx.data_ptr(), M, K, (int)Np, Np,
-              Np, K, 0, cur_stream()));
-  // dw = dy^T @ x: NT-GEMM(dyT [N,Mp], xT [K,Mp]) → fp32
-  Tensor dyT = at::zeros({N, Mp}, dyc.options());
-  CHK(transpose2d_bf16(dyc.data_ptr(), dyT.data_ptr(), M, N, Mp, cur_stream()));
-  Tensor xT = at::zeros({K, Mp}, xc.options());
-  CHK(transpose2d_bf16(xc.data_ptr(), xT.data_ptr(), M, K, Mp, cur_stream()));
+  CHK(gemm_nt_tn(dyp.data_ptr(), wc.data_ptr(), dx.data_ptr(), M, K, N, Np, K,
+                 K, 0, cur_stream()));
+  // dw = dy^T @ x: both operands k-strided (k = batch row m) → fp32
   Tensor dw = at::empty({N, K}, f32);
-  CHK(gemm_nt(dyT.data_ptr(), xT.data_ptr(), dw.data_ptr(), N, K, (int)Mp, Mp,
-              Mp, K, 1, cur_stream()));
+  CHK(gemm_tn_tn(dyc.data_ptr(), xc.data_ptr(), dw.data_ptr(), N, K, M, N, K,
+                 K, 1, cur_stream()));
   Tensor db = at::empty({N}, f32);
   CHK(colsum_bf16(dyc.data_ptr(), db.data_ptr<float>(), M, N, cur_stream()));
   return {dx, dw, db};

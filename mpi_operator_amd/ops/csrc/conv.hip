@@ -6,11 +6,13 @@
 //   dgrad: dx[m=(n,h,w)][c]      = sum_{k=(r,s,q)}  dy[opatch(m,k)] * wT[k][c]
 //   wgrad: dw[kout][(r,s,c)]     = sum_{m}          dyT[kout][m] * PT[(r,s,c)][m]
 //
-// fwd/dgrad instantiate the NT-GEMM tile template with gather loaders
-// (padding/stride validity → zero-fill, so edges and stride-2 dgrad need no
-// special cases); wgrad runs split-K NT-GEMM over transposed dy and a
-// transposed im2col buffer, accumulating exactly in fp32.
+// fwd/dgrad/wgrad all run on the mixed-staging MFMA GEMM (mix_gemm.h):
+// gather loaders turn padding/stride validity into zero-fill, and k-strided
+// operands (wgrad's dy and implicit im2col, dgrad's weight) are transposed
+// in the LDS write pass — nothing is ever materialized or pre-transposed.
+// wgrad is split-K with exact fp32 partial slabs + a reduce kernel.
 #include "mfma_tile.h"
+#include "mix_gemm.h"
 
 struct ConvFwdALoader {
   const uint16_t *x;
@@ -60,30 +62,46 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
                                hipStream_t strm) {
   long M = (long)N * HO * WO;
   int K = R * S * C;
+  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+    // 1x1 conv IS a GEMM on the NHWC image viewed [M][C] — skip the
+    // gather arithmetic entirely (≈47% of ResNet bottleneck FLOPs)
+    GemmLoader la{(const uint16_t *)x, (int)M, (long)C, C};
+    GemmLoader lb{(const uint16_t *)w, Kout, (long)C, C};
+    return launch_nt_gemm(la, lb, y, (int)M, Kout, C, Kout, false, strm);
+  }
   ConvFwdALoader la{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, M, K, S * C};
   GemmLoader lb{(const uint16_t *)w, Kout, (long)K, K};
   return launch_nt_gemm(la, lb, y, (int)M, Kout, K, Kout, false, strm);
 }
 
-// dgrad: wT is the [RSC][Kout] transpose of w (transpose2d_bf16 of w[K][RSC])
-extern "C" hipError_t conv_dgrad(const void *dy, const void *wT, void *dx,
+// dgrad: w used directly (channels_last [Kout][RSC]); the k-strided B view
+// w[q][rs·C + c0..7] is transposed in the LDS write pass (TN staging).
+extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
                                  int N, int H, int W, int C, int Kout, int R,
                                  int S, int stride, int pad, int HO, int WO,
                                  hipStream_t strm) {
   long M = (long)N * H * W;
   int K = R * S * Kout;
+  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+    // 1x1 dgrad: dx[M][C] = dy[M][Q] · w[Q][C] (w TN-staged, no gather)
+    GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
+    TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
+    return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                           dx, (int)M, C, Kout, C, false, strm);
+  }
   ConvDgradALoader la{(const uint16_t *)dy, H, W, Kout, HO, WO, S, stride, pad, M, K};
-  // B row = c; k=(r,s,q): wT[(r*S+s)*C + c][q] → addr = ((r*S+s)*C + c)*Q + q
-  struct DgradBLoader {
-    const uint16_t *wT;
-    int C, Q, K;
-    DEV_INLINE ushort8 load(int c, int k) const {
-      if (c >= C || k >= K) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+  // TN B: k=(r,s,q) with q fastest; element (c, k) = w[q][(r*S+s)*C + c]
+  struct DgradWTn {
+    const uint16_t *w;
+    int C, Q, K, RSC;
+    DEV_INLINE ushort8 load(int k, int c0) const {
+      if (k >= K || c0 >= C) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
       int q = k % Q, rs = k / Q;
-      return *(const ushort8 *)(wT + ((long)rs * C + c) * Q + q);
+      return *(const ushort8 *)(w + (long)q * RSC + rs * C + c0);
     }
-  } lb{(const uint16_t *)wT, C, Kout, K};
-  return launch_nt_gemm(la, lb, dx, (int)M, C, K, C, false, strm);
+  } lb{(const uint16_t *)w, C, Kout, K, R * S * C};
+  return launch_mix_gemm(NtStage<ConvDgradALoader>{la}, TnStage<DgradWTn>{lb},
+                         dx, (int)M, C, K, C, false, strm);
 }
 
 template <bool OUT_BF16>
@@ -111,19 +129,32 @@ extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
   return hipGetLastError();
 }
 
-// wgrad GEMM over transposed operands (dyT [Kout][M], PT [RSC][M8*8]):
-// partial fp32 slabs [splits][Kout][RSC] → splitk_reduce → dw fp32.
-extern "C" hipError_t conv_wgrad_gemm(const void *dyT, const void *PT,
-                                      float *partial, void *dw, int Kout,
-                                      int RSC, long M, long ldPT, int splits,
-                                      int dw_bf16, hipStream_t strm) {
+// Implicit wgrad: dw[q][rsc] = Σ_m dy[m][q]·xcol[m][rsc], both operands
+// k-strided (k = output pixel m) and TN-staged straight from dy / x —
+// no transposes, no im2col buffer. fp32 partial slabs → splitk_reduce.
+extern "C" hipError_t conv_wgrad_implicit(const void *dy, const void *x,
+                                          float *partial, void *dw, int N,
+                                          int H, int W, int C, int Kout, int R,
+                                          int S, int stride, int pad, int HO,
+                                          int WO, int splits, int dw_bf16,
+                                          hipStream_t strm) {
+  long M = (long)N * HO * WO;
+  int RSC = R * S * C;
   int nk = (int)((M + BK - 1) / BK);
   if (splits > nk) splits = nk > 0 ? nk : 1; // must match the launch's clamp:
   // the reduce below must sum exactly the slabs the GEMM wrote.
-  GemmLoader la{(const uint16_t *)dyT, Kout, M, (int)M};
-  GemmLoader lb{(const uint16_t *)PT, RSC, ldPT, (int)M};
-  hipError_t e = launch_nt_gemm(la, lb, partial, Kout, RSC, (int)M, RSC, true,
-                                strm, splits);
+  TnRowMajor la{(const uint16_t *)dy, Kout, (int)M, Kout};
+  hipError_t e;
+  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+    // 1x1 wgrad: the im2col column of x IS x itself — plain TN view
+    TnRowMajor lb{(const uint16_t *)x, (long)C, (int)M, C};
+    e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
+                        partial, Kout, RSC, (int)M, RSC, true, strm, splits);
+  } else {
+    TnXcol lb{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, M, RSC};
+    e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnXcol>{lb},
+                        partial, Kout, RSC, (int)M, RSC, true, strm, splits);
+  }
   if (e != hipSuccess) return e;
   return splitk_reduce(partial, splits, (long)Kout * RSC, dw, dw_bf16, strm);
 }

@@ -1,6 +1,8 @@
-// Plain NT-GEMM (C = A·B^T with both operands reduce-contiguous) plus the
-// transpose helpers that feed it for linear-layer backward and conv wgrad.
+// GEMM entry points over the MFMA tile templates: NT (both operands
+// k-contiguous), NT×TN and TN×TN (k-strided operands transposed in the LDS
+// write pass — linear backward runs with zero materialized transposes).
 #include "mfma_tile.h"
+#include "mix_gemm.h"
 
 extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
                               int N, int K, long lda, long ldb, long ldc,
@@ -8,6 +10,27 @@ extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
   GemmLoader la{(const uint16_t *)a, M, lda, K};
   GemmLoader lb{(const uint16_t *)b, N, ldb, K};
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, c_f32 != 0, s);
+}
+
+// C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx
+extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
+                                 int N, int K, long lda, long ldb, long ldc,
+                                 int c_f32, hipStream_t s) {
+  GemmLoader la{(const uint16_t *)a, M, lda, K};
+  TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
+  return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, c,
+                         M, N, K, ldc, c_f32 != 0, s);
+}
+
+// C[M][N] = Σ_k A(k-strided rows, M cols) · B(k-strided rows, N cols)
+// — linear dw (A = dy [Mbatch][N_out] viewed k-strided, B = x likewise)
+extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
+                                 int N, int K, long lda, long ldb, long ldc,
+                                 int c_f32, hipStream_t s) {
+  TnRowMajor la{(const uint16_t *)a, lda, K, M};
+  TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
+  return launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb}, c,
+                         M, N, K, ldc, c_f32 != 0, s);
 }
 
 // bf16 2-D transpose: out[j][i] = in[i][j], output leading dim ldo >= R
@@ -48,49 +71,6 @@ extern "C" hipError_t transpose2d_bf16(const void *in, void *out, int R, int C,
   dim3 grid((R + 63) / 64, (C + 63) / 64);
   transpose2d_k<<<grid, 256, 0, s>>>((const uint16_t *)in, (uint16_t *)out, R,
                                      C, ldo);
-  return hipGetLastError();
-}
-
-// Transposed im2col for wgrad: PT[(r,s,c)][m] from NHWC input x.
-// m = (n,ho,wo) output-pixel index; each thread owns one (rsc, m0..m0+7)
-// strip; lanes are consecutive in rsc (c fastest) so the per-j gathers
-// coalesce across the wave while the 16 B writes coalesce along m.
-__global__ void im2col_t_k(const uint16_t *__restrict__ x,
-                           ushort8 *__restrict__ pt, int N, int H, int W,
-                           int C, int HO, int WO, int R, int S, int stride,
-                           int pad, long M8) {
-  long RSC = (long)R * S * C;
-  for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < RSC * M8;
-       t += (long)gridDim.x * blockDim.x) {
-    long m0 = (t / RSC) * 8;
-    int rsc = t % RSC;
-    int c = rsc % C, s_ = (rsc / C) % S, r = rsc / (C * S);
-    ushort8 v;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      long m = m0 + j;
-      int wo = m % WO;
-      int ho = (m / WO) % HO;
-      int n = m / ((long)WO * HO);
-      int h = ho * stride + r - pad, w = wo * stride + s_ - pad;
-      bool ok = (n < N) && h >= 0 && h < H && w >= 0 && w < W;
-      v[j] = ok ? x[((long)(n * H + h) * W + w) * C + c] : 0;
-    }
-    pt[(long)rsc * M8 + (m0 >> 3)] = v;
-  }
-}
-
-extern "C" hipError_t im2col_t(const void *x, void *pt, int N, int H, int W,
-                               int C, int HO, int WO, int R, int S, int stride,
-                               int pad, hipStream_t strm) {
-  long M = (long)N * HO * WO;
-  long M8 = (M + 7) / 8;
-  long tasks = (long)R * S * C * M8;
-  long blocks = (tasks + 255) / 256;
-  if (blocks > 4096) blocks = 4096;
-  im2col_t_k<<<(int)blocks, 256, 0, strm>>>((const uint16_t *)x, (ushort8 *)pt,
-                                            N, H, W, C, HO, WO, R, S, stride,
-                                            pad, M8);
   return hipGetLastError();
 }
 

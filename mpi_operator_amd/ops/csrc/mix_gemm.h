@@ -1,0 +1,221 @@
+// Mixed-staging MFMA GEMM for gfx950: C[M][N] = sum_k A(m,k)·B(n,k) where
+// EITHER operand may be stored k-contiguous (NT: rows along the output dim,
+// 16 B loads along k) or k-strided (TN: memory rows ARE k, 16 B loads along
+// the output dim, transposed during the LDS write pass).
+//
+// This removes every materialized transpose / im2col in the framework:
+//   conv wgrad   : dw[q][rsc] = Σ_m dy[m][q]·xcol[m][rsc]   (TN × TN-gather)
+//   conv dgrad   : dx[m][c]   = Σ_k dy-gather · w[q][rs·C+c] (NT-gather × TN)
+//   linear dx    : dx[m][k]   = Σ_n dy[m][n]·w[n][k]         (NT × TN)
+//   linear dw    : dw[n][k]   = Σ_m dy[m][n]·x[m][k]         (TN × TN)
+// (profiled before: im2col_t_k 15% + transpose2d_k 7.4% of step time —
+//  both are pure data movement this file deletes.)
+//
+// LDS image: [128 rows][9 slots of 16 B] (pitch 144 B). The 9-slot pitch
+// makes ds_read_b128 conflict-free WITHOUT an XOR swizzle: a 16-lane read
+// group covers rows r..r+15 at one slot, and banks (36·row + 4·slot) mod 64
+// are distinct because 36·row mod 64 has period 16. The TN write pass packs
+// two k's per lane into one ds_write_b32 (its ≤2-way write conflicts are
+// free: LDS-array cycles stay under the instruction's own 4).
+//
+// Schedule: register staging, issue-early / write-late (load tile t+1's
+// global data during tile t's MFMAs, write it to LDS after them) — the
+// measured-best register-staging form for this 2-barrier structure.
+#pragma once
+#include "mfma_tile.h"
+
+constexpr int MXP = 9; // slots per LDS image row (pitch 144 B)
+
+// ---- staging policies ----
+// NT: loader.load(row, k0) -> 8 bf16 along k (row-major k-contiguous).
+// TN: loader.load(k, col0) -> 8 bf16 along the output dim (k-strided).
+
+template <class L> struct NtStage {
+  L l;
+  ushort8 r[4];
+  DEV_INLINE void load(int tid, int base, int kb) {
+    int s_row = tid >> 3, s_slot = tid & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      r[i] = l.load(base + s_row + 32 * i, kb + s_slot * 8);
+  }
+  DEV_INLINE void write(int tid, ushort8 *img) const {
+    int s_row = tid >> 3, s_slot = tid & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) img[(s_row + 32 * i) * MXP + s_slot] = r[i];
+  }
+};
+
+template <class L> struct TnStage {
+  L l;
+  ushort8 r[4]; // [it][k-parity]: 2 pair-loads of 2 adjacent k each
+  DEV_INLINE void load(int tid, int base, int kb) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;  // pair index in [0,512)
+      int k0 = (p & 31) * 2;   // even k within the 64-deep tile
+      int col0 = (p >> 5) * 8; // 8 output-dim columns
+      r[it * 2] = l.load(kb + k0, base + col0);
+      r[it * 2 + 1] = l.load(kb + k0 + 1, base + col0);
+    }
+  }
+  DEV_INLINE void write(int tid, ushort8 *img) const {
+    uint32_t *img32 = (uint32_t *)img;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;
+      int k0 = (p & 31) * 2;
+      int col0 = (p >> 5) * 8;
+      const ushort8 &va = r[it * 2], &vb = r[it * 2 + 1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        img32[(col0 + j) * (MXP * 4) + (k0 >> 1)] =
+            (uint32_t)va[j] | ((uint32_t)vb[j] << 16);
+    }
+  }
+};
+
+template <class SA, class SB, bool C_F32>
+__global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
+    SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, long ldc,
+    int tiles_n, int kt_per_split, long split_stride) {
+  int tile = blockIdx.x;
+  int split = blockIdx.y;
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * BM, col0 = tn * BN;
+  int tid = threadIdx.x;
+  int lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 1, wc = wave & 1;
+
+  __shared__ ushort8 lds[2][2][BM * MXP]; // [buf][A|B][image]
+
+  float4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
+
+  int nk_total = (K + BK - 1) / BK;
+  int t0 = split * kt_per_split;
+  int nk = min(kt_per_split, nk_total - t0);
+  if (nk < 0) nk = 0;
+
+  if (nk > 0) {
+    sa.load(tid, row0, t0 * BK);
+    sb.load(tid, col0, t0 * BK);
+    sa.write(tid, lds[0][0]);
+    sb.write(tid, lds[0][1]);
+  }
+  __syncthreads();
+
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+    if (t + 1 < nk) { // issue-early: HBM latency hides under the MFMAs
+      sa.load(tid, row0, (t0 + t + 1) * BK);
+      sb.load(tid, col0, (t0 + t + 1) * BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      bf16x8 af[4], bf_[4];
+      int slot = kk * 4 + (lane >> 4);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = us8_to_bf8v(lds[buf][0][(wr * 64 + mi * 16 + (lane & 15)) * MXP + slot]);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf_[ni] = us8_to_bf8v(lds[buf][1][(wc * 64 + ni * 16 + (lane & 15)) * MXP + slot]);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < nk) { // write-late
+      sa.write(tid, lds[buf ^ 1][0]);
+      sb.write(tid, lds[buf ^ 1][1]);
+    }
+    __syncthreads();
+  }
+
+  cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        if (C_F32)
+          ((float *)cptr)[(long)row * ldc + col] = acc[mi][ni][r];
+        else
+          ((uint16_t *)cptr)[(long)row * ldc + col] = f2bf(acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
+template <class SA, class SB>
+static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
+                                  int N, int K, long ldc, bool c_f32,
+                                  hipStream_t s, int splits = 1) {
+  int tiles_m = (M + BM - 1) / BM, tiles_n = (N + BN - 1) / BN;
+  int nk = (K + BK - 1) / BK;
+  if (splits > nk) splits = nk > 0 ? nk : 1;
+  int kts = (nk + splits - 1) / splits;
+  long split_stride = (long)M * ldc;
+  dim3 grid(tiles_m * tiles_n, splits);
+  if (c_f32)
+    mix_gemm_k<SA, SB, true><<<grid, NT_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride);
+  else
+    mix_gemm_k<SA, SB, false><<<grid, NT_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride);
+  return hipGetLastError();
+}
+
+// ---- TN loaders ----
+
+// k-strided view of a row-major [K][cols] matrix (dy, x, w as TN operand).
+struct TnRowMajor {
+  const uint16_t *p;
+  long ld;  // elements per k-row
+  int kdim; // rows (= reduce extent)
+  int cols;
+  DEV_INLINE ushort8 load(int k, int c0) const {
+    if (k >= kdim || c0 >= cols) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    if (c0 + 8 <= cols) return *(const ushort8 *)(p + (long)k * ld + c0);
+    ushort8 v{0, 0, 0, 0, 0, 0, 0, 0}; // ragged tail (cols not %8)
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (c0 + j < cols) v[j] = p[(long)k * ld + c0 + j];
+    return v;
+  }
+};
+
+// Implicit-im2col column gather for wgrad: k = output pixel m = (n,ho,wo),
+// col = (r,s,c) with c fastest; returns x[n][h][w][c0..c0+7] or zeros where
+// the patch leaves the image (requires C%8==0 so an 8-col run stays inside
+// one (r,s) cell — guaranteed by the NHWC8 input padding).
+struct TnXcol {
+  const uint16_t *x;
+  int H, W, C, HO, WO, S, stride, pad;
+  long M;
+  int RSC;
+  DEV_INLINE ushort8 load(int m, int rsc0) const {
+    if (m >= M || rsc0 >= RSC) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    int c = rsc0 % C, rs = rsc0 / C;
+    int s_ = rs % S, r = rs / S;
+    int wo = m % WO;
+    long t = m / WO;
+    int ho = t % HO;
+    int n = t / HO;
+    int h = ho * stride + r - pad, w = wo * stride + s_ - pad;
+    if ((unsigned)h >= (unsigned)H || (unsigned)w >= (unsigned)W)
+      return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    return *(const ushort8 *)(x + ((long)(n * H + h) * W + w) * C + c);
+  }
+};

@@ -165,6 +165,7 @@ class LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         ctx.save_for_backward(x, w)
+        ctx.b_dtype = b.dtype
         if x.is_cuda:
             return hip_ext().linear_fwd(x, w, b)
         return ref.linear_fwd(x, w, b)
@@ -175,11 +176,11 @@ class LinearFn(torch.autograd.Function):
         if dy.is_cuda:
             dy = dy.contiguous()
             dx, dw, db = hip_ext().linear_bwd(dy, x, w)
-            return dx, dw.to(w.dtype), db
+            return dx, dw.to(w.dtype), db.to(ctx.b_dtype)
         dx = dy @ w
         dw = dy.transpose(0, 1).float() @ x.float()
         db = dy.float().sum(0)
-        return dx, dw.to(w.dtype), db
+        return dx, dw.to(w.dtype), db.to(ctx.b_dtype)
 
 
 def linear(x, w, b):
