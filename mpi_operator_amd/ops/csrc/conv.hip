@@ -14,45 +14,167 @@
 #include "mfma_tile.h"
 #include "mix_gemm.h"
 
-struct ConvFwdALoader {
+// Custom NT stager for conv fwd: the 4 staged rows (output pixels) are
+// FIXED for the whole kernel, so their (n, ho, wo) decomposition is hoisted
+// into init(); per k-step only the (r,s,c) split of one k remains — the
+// innermost gather is 2 adds + 2 bounds checks + 1 load.
+struct ConvFwdStage {
   const uint16_t *x;
-  int H, W, C, HO, WO, S, stride, pad;
+  int H, W, C, HO, WO, S, stride, pad, K;
   long M;
-  int K, SC;
-  DEV_INLINE ushort8 load(int m, int k) const {
-    if (m >= M || k >= K) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+  ushort8 r[4];
+  long nbase_[4]; // n*H*W (row-invariant part of the address)
+  int hb_[4], wb_[4];
+  bool ok_[4];
+  DEV_INLINE void init(int tid, int base) {
+    int s_row = tid >> 3;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      long m = base + s_row + 32 * i;
+      ok_[i] = m < M;
+      int wo = (int)(m % WO);
+      long t = m / WO;
+      int ho = (int)(t % HO);
+      int n = (int)(t / HO);
+      nbase_[i] = (long)n * H * W;
+      hb_[i] = ho * stride - pad;
+      wb_[i] = wo * stride - pad;
+    }
+  }
+  DEV_INLINE void load(int tid, int, int kb) {
+    int k = kb + (tid & 7) * 8;
     int c = k % C, rs = k / C;
-    int s_ = rs % S, r = rs / S;
-    int wo = m % WO;
-    long t = m / WO;
-    int ho = t % HO;
-    int n = t / HO;
-    int h = ho * stride + r - pad, w = wo * stride + s_ - pad;
-    if ((unsigned)h >= (unsigned)H || (unsigned)w >= (unsigned)W)
-      return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-    return *(const ushort8 *)(x + ((long)(n * H + h) * W + w) * C + c);
+    int s_ = rs % S, rr = rs / S;
+    bool kok = k < K;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int h = hb_[i] + rr, w = wb_[i] + s_;
+      if (kok && ok_[i] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        r[i] = *(const ushort8 *)(x + (nbase_[i] + (long)h * W + w) * C + c);
+      else
+        r[i] = ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+  DEV_INLINE void write(int tid, ushort8 *img) const {
+    int s_row = tid >> 3, s_slot = tid & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) img[(s_row + 32 * i) * MXP + s_slot] = r[i];
   }
 };
 
-struct ConvDgradALoader {
+// NT stager for dgrad: row (input pixel) decomposition hoisted to init();
+// STRIDE is a template parameter so the innermost %/÷ are shifts, not the
+// full integer division a runtime stride emits.
+template <int STRIDE> struct ConvDgradStage {
   const uint16_t *dy;
-  int H, W, Q /*Kout*/, HO, WO, S, stride, pad;
+  int H, W, Q /*Kout*/, HO, WO, S, pad, K;
   long M;
-  int K;
-  DEV_INLINE ushort8 load(int m, int k) const {
-    if (m >= M || k >= K) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+  ushort8 r[4];
+  long nbase_[4]; // n*HO*WO
+  int hb_[4], wb_[4];
+  bool ok_[4];
+  DEV_INLINE void init(int tid, int base) {
+    int s_row = tid >> 3;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      long m = base + s_row + 32 * i;
+      ok_[i] = m < M;
+      int w_ = (int)(m % W);
+      long t = m / W;
+      int h_ = (int)(t % H);
+      int n = (int)(t / H);
+      nbase_[i] = (long)n * HO * WO;
+      hb_[i] = h_ + pad;
+      wb_[i] = w_ + pad;
+    }
+  }
+  DEV_INLINE void load(int tid, int, int kb) {
+    int k = kb + (tid & 7) * 8;
     int q = k % Q, rs = k / Q;
-    int s_ = rs % S, r = rs / S;
-    int w_ = m % W;
-    long t = m / W;
-    int h_ = t % H;
-    int n = t / H;
-    int hn = h_ + pad - r, wn = w_ + pad - s_;
-    if (hn < 0 || wn < 0 || hn % stride || wn % stride)
-      return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-    int ho = hn / stride, wo = wn / stride;
-    if (ho >= HO || wo >= WO) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-    return *(const ushort8 *)(dy + ((long)(n * HO + ho) * WO + wo) * Q + q);
+    int s_ = rs % S, rr = rs / S;
+    bool kok = k < K;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int hn = hb_[i] - rr, wn = wb_[i] - s_;
+      int ho = hn / STRIDE, wo = wn / STRIDE;
+      bool ok = kok && ok_[i] && hn >= 0 && wn >= 0 &&
+                (STRIDE == 1 || (hn % STRIDE == 0 && wn % STRIDE == 0)) &&
+                ho < HO && wo < WO;
+      r[i] = ok ? *(const ushort8 *)(dy + (nbase_[i] + (long)ho * WO + wo) * Q + q)
+                : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+  DEV_INLINE void write(int tid, ushort8 *img) const {
+    int s_row = tid >> 3, s_slot = tid & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) img[(s_row + 32 * i) * MXP + s_slot] = r[i];
+  }
+};
+
+// TN stager for wgrad's implicit-im2col operand: each thread's two column
+// octets (r,s,c) are fixed for the whole kernel — decomposed once in
+// init(); per k-step only the even output-pixel m is decomposed (odd m is
+// derived by carry), so the inner gather is adds + bounds + load.
+struct XcolStage {
+  const uint16_t *x;
+  int H, W, C, HO, WO, S, stride, pad, RSC;
+  long M;
+  ushort8 r[4];
+  int k0_[2], coff_[2], rr_[2], ss_[2];
+  bool cok_[2];
+  DEV_INLINE void init(int tid, int base) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;
+      k0_[it] = (p & 31) * 2;
+      int rsc0 = base + (p >> 5) * 8;
+      cok_[it] = rsc0 < RSC;
+      int c = rsc0 % C, rs = rsc0 / C;
+      ss_[it] = rs % S;
+      rr_[it] = rs / S;
+      coff_[it] = c;
+    }
+  }
+  DEV_INLINE ushort8 gather(int it, int n, int ho, int wo, bool mok) const {
+    int h = ho * stride + rr_[it] - pad, w = wo * stride + ss_[it] - pad;
+    if (mok && cok_[it] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+      return *(const ushort8 *)(x + ((long)(n * H + h) * W + w) * C + coff_[it]);
+    return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+  DEV_INLINE void load(int tid, int, int kb) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      long m = kb + k0_[it]; // even
+      int wo = (int)(m % WO);
+      long t = m / WO;
+      int ho = (int)(t % HO);
+      int n = (int)(t / HO);
+      r[it * 2] = gather(it, n, ho, wo, m < M);
+      // odd m = even + 1: carry-propagate instead of re-dividing
+      int wo2 = wo + 1, ho2 = ho, n2 = n;
+      if (wo2 == WO) {
+        wo2 = 0;
+        if (++ho2 == HO) {
+          ho2 = 0;
+          ++n2;
+        }
+      }
+      r[it * 2 + 1] = gather(it, n2, ho2, wo2, m + 1 < M);
+    }
+  }
+  DEV_INLINE void write(int tid, ushort8 *img) const {
+    uint32_t *img32 = (uint32_t *)img;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;
+      int k0 = (p & 31) * 2;
+      int col0 = (p >> 5) * 8;
+      const ushort8 &va = r[it * 2], &vb = r[it * 2 + 1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        img32[(col0 + j) * (MXP * 4) + (k0 >> 1)] =
+            (uint32_t)va[j] | ((uint32_t)vb[j] << 16);
+    }
   }
 };
 
@@ -69,10 +191,21 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
     GemmLoader lb{(const uint16_t *)w, Kout, (long)C, C};
     return launch_nt_gemm(la, lb, y, (int)M, Kout, C, Kout, false, strm);
   }
-  ConvFwdALoader la{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, M, K, S * C};
+  ConvFwdStage sa{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, K, M};
   GemmLoader lb{(const uint16_t *)w, Kout, (long)K, K};
-  return launch_nt_gemm(la, lb, y, (int)M, Kout, K, Kout, false, strm);
+  return launch_mix_gemm(sa, NtStage<GemmLoader>{lb}, y, (int)M, Kout, K, Kout,
+                         false, strm);
 }
+
+struct DgradWTn {
+  const uint16_t *w;
+  int C, Q, K, RSC;
+  DEV_INLINE ushort8 load(int k, int c0) const {
+    if (k >= K || c0 >= C) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    int q = k % Q, rs = k / Q;
+    return *(const ushort8 *)(w + (long)q * RSC + rs * C + c0);
+  }
+};
 
 // dgrad: w used directly (channels_last [Kout][RSC]); the k-strided B view
 // w[q][rs·C + c0..7] is transposed in the LDS write pass (TN staging).
@@ -89,19 +222,16 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
     return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
                            dx, (int)M, C, Kout, C, false, strm);
   }
-  ConvDgradALoader la{(const uint16_t *)dy, H, W, Kout, HO, WO, S, stride, pad, M, K};
   // TN B: k=(r,s,q) with q fastest; element (c, k) = w[q][(r*S+s)*C + c]
-  struct DgradWTn {
-    const uint16_t *w;
-    int C, Q, K, RSC;
-    DEV_INLINE ushort8 load(int k, int c0) const {
-      if (k >= K || c0 >= C) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-      int q = k % Q, rs = k / Q;
-      return *(const ushort8 *)(w + (long)q * RSC + rs * C + c0);
-    }
-  } lb{(const uint16_t *)w, C, Kout, K, R * S * C};
-  return launch_mix_gemm(NtStage<ConvDgradALoader>{la}, TnStage<DgradWTn>{lb},
-                         dx, (int)M, C, K, C, false, strm);
+  DgradWTn lb{(const uint16_t *)w, C, Kout, K, R * S * C};
+  if (stride == 1) {
+    ConvDgradStage<1> sa{(const uint16_t *)dy, H, W, Kout, HO, WO, S, pad, K, M};
+    return launch_mix_gemm(sa, TnStage<DgradWTn>{lb}, dx, (int)M, C, K, C,
+                           false, strm);
+  }
+  ConvDgradStage<2> sa{(const uint16_t *)dy, H, W, Kout, HO, WO, S, pad, K, M};
+  return launch_mix_gemm(sa, TnStage<DgradWTn>{lb}, dx, (int)M, C, K, C, false,
+                         strm);
 }
 
 template <bool OUT_BF16>
@@ -151,9 +281,9 @@ extern "C" hipError_t conv_wgrad_implicit(const void *dy, const void *x,
     e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
                         partial, Kout, RSC, (int)M, RSC, true, strm, splits);
   } else {
-    TnXcol lb{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, M, RSC};
-    e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnXcol>{lb},
-                        partial, Kout, RSC, (int)M, RSC, true, strm, splits);
+    XcolStage sb{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, RSC, M};
+    e = launch_mix_gemm(TnStage<TnRowMajor>{la}, sb, partial, Kout, RSC,
+                        (int)M, RSC, true, strm, splits);
   }
   if (e != hipSuccess) return e;
   return splitk_reduce(partial, splits, (long)Kout * RSC, dw, dw_bf16, strm);

@@ -33,6 +33,7 @@ constexpr int MXP = 9; // slots per LDS image row (pitch 144 B)
 template <class L> struct NtStage {
   L l;
   ushort8 r[4];
+  DEV_INLINE void init(int, int) {}
   DEV_INLINE void load(int tid, int base, int kb) {
     int s_row = tid >> 3, s_slot = tid & 7;
 #pragma unroll
@@ -49,6 +50,7 @@ template <class L> struct NtStage {
 template <class L> struct TnStage {
   L l;
   ushort8 r[4]; // [it][k-parity]: 2 pair-loads of 2 adjacent k each
+  DEV_INLINE void init(int, int) {}
   DEV_INLINE void load(int tid, int base, int kb) {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
@@ -100,6 +102,8 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   int nk = min(kt_per_split, nk_total - t0);
   if (nk < 0) nk = 0;
 
+  sa.init(tid, row0);
+  sb.init(tid, col0);
   if (nk > 0) {
     sa.load(tid, row0, t0 * BK);
     sb.load(tid, col0, t0 * BK);
