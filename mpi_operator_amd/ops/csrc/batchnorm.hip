@@ -97,14 +97,14 @@ DEV_INLINE void lane8_sums(const float *__restrict__ partial, int grid, int C,
                            int c, int lane8, float &s0, float &s1) {
   s0 = 0.f;
   s1 = 0.f;
-  for (int g = lane8; g < grid; g += 8) {
+  for (int g = lane8; g < grid; g += 32) {
     s0 += partial[(long)g * 2 * C + c];
     s1 += partial[(long)g * 2 * C + C + c];
   }
 #pragma unroll
-  for (int off = 4; off > 0; off >>= 1) {
-    s0 += __shfl_down(s0, off, 8);
-    s1 += __shfl_down(s1, off, 8);
+  for (int off = 16; off > 0; off >>= 1) {
+    s0 += __shfl_down(s0, off, 32);
+    s1 += __shfl_down(s1, off, 32);
   }
 }
 
@@ -121,7 +121,7 @@ __global__ void bn_finalize_fwd_k(const float *__restrict__ partial, int grid,
                                   float *__restrict__ running_var,
                                   float momentum, float unbias) {
   int t = blockIdx.x * blockDim.x + threadIdx.x;
-  int c = t / 8, lane8 = t % 8;
+  int c = t / 32, lane8 = t % 32;
   if (c >= C) return;
   float s, sq;
   lane8_sums(partial, grid, C, c, lane8, s, sq);
@@ -150,7 +150,7 @@ __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
                                   float *__restrict__ k2, // k1*dbeta/m
                                   float *__restrict__ k3) { // k1*dgamma/m (×xhat in apply)
   int t = blockIdx.x * blockDim.x + threadIdx.x;
-  int c = t / 8, lane8 = t % 8;
+  int c = t / 32, lane8 = t % 32;
   if (c >= C) return;
   float s0, s1;
   lane8_sums(partial, grid, C, c, lane8, s0, s1);
@@ -240,10 +240,19 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
   }
 }
 
+// Partials grid is capped by the slab ([512][2][C]); the elementwise
+// apply kernels have no slab and get a bandwidth-sized grid (256 CUs want
+// >=2 blocks/CU in flight to cover HBM latency).
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
-  grid = (int)(g > 256 ? 256 : (g < 1 ? 1 : g));
+  grid = (int)(g > 512 ? 512 : (g < 1 ? 1 : g));
+}
+
+static int bn_apply_grid(long M, int C8) {
+  int rpb = 256 / C8;
+  long g = (M + rpb - 1) / rpb;
+  return (int)(g > 4096 ? 4096 : (g < 1 ? 1 : g));
 }
 
 extern "C" hipError_t bn_fwd_train_launch(
@@ -259,12 +268,12 @@ extern "C" hipError_t bn_fwd_train_launch(
                                         nullptr, nullptr, partial, M, C8, 0);
   HIP_KERNEL_CHECK();
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
-  bn_finalize_fwd_k<<<cdiv_h((long)C * 8, 256), 256, 0, s>>>(
+  bn_finalize_fwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
       partial, grid, C, gamma, beta, 1.f / (float)M, eps, mean, invstd, scale,
       shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
-  bn_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)x, scale, shift,
-                                  (ushort8 *)y, M, C8, relu);
+  bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
+      (const ushort8 *)x, scale, shift, (ushort8 *)y, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -276,8 +285,8 @@ extern "C" hipError_t bn_fwd_eval_launch(const void *x, const float *scale,
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
-  bn_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)x, scale, shift,
-                                  (ushort8 *)y, M, C8, relu);
+  bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
+      (const ushort8 *)x, scale, shift, (ushort8 *)y, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -297,12 +306,12 @@ extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
                                         (const ushort8 *)y, mean, invstd,
                                         partial, M, C8, relu);
   HIP_KERNEL_CHECK();
-  bn_finalize_bwd_k<<<cdiv_h((long)C * 8, 256), 256, 0, s>>>(
+  bn_finalize_bwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
       partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
   HIP_KERNEL_CHECK();
-  bn_bwd_apply_k<<<grid, 256, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x,
-                                      (const ushort8 *)y, mean, invstd, k1, k2,
-                                      k3, (ushort8 *)dx, M, C8, relu);
+  bn_bwd_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
+      (const ushort8 *)dy, (const ushort8 *)x, (const ushort8 *)y, mean,
+      invstd, k1, k2, k3, (ushort8 *)dx, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

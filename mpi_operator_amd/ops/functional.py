@@ -84,6 +84,9 @@ class BNReLUFn(torch.autograd.Function):
         ctx.relu = relu
         ctx.save_for_backward(x, y, gamma, mean, invstd)
         ctx.mark_non_differentiable(mean, invstd)
+        # without this, autograd materializes a zero tensor for dmean and
+        # dinvstd on EVERY backward (2 fill launches × every BN layer × step)
+        ctx.set_materialize_grads(False)
         return y, mean, invstd
 
     @staticmethod
