@@ -107,6 +107,7 @@ def main():
                 "image": image,
                 "parallelism": f"dp{n_gpus}",
                 "impl": args.impl,
+                "graph": res.get("graph", False),
                 "loss": res["loss"],
             },
         }
