@@ -73,7 +73,7 @@ hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
 hipError_t softmax_xent_fwd_launch(const void *, const long *, float *, float *,
                                    int, int, hipStream_t);
 hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
-                                   int, float, hipStream_t);
+                                   int, const float *, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
                    long, long, int, hipStream_t);
 hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
@@ -320,12 +320,15 @@ static std::vector<Tensor> softmax_xent_fwd(const Tensor &logits,
 }
 
 static Tensor softmax_xent_bwd(const Tensor &probs, const Tensor &target,
-                               double scale) {
+                               const Tensor &dloss) {
   const HIPDeviceGuard guard(probs.device());
   int B = probs.size(0), V = probs.size(1);
+  TORCH_CHECK(dloss.is_cuda() && dloss.scalar_type() == at::kFloat &&
+              dloss.numel() == 1, "dloss must be a device fp32 scalar");
   Tensor d = at::empty({B, V}, probs.options().dtype(at::kBFloat16));
   CHK(softmax_xent_bwd_launch(probs.data_ptr<float>(), target.data_ptr<long>(),
-                              d.data_ptr(), B, V, (float)scale, cur_stream()));
+                              d.data_ptr(), B, V, dloss.data_ptr<float>(),
+                              cur_stream()));
   return d;
 }
 

@@ -77,11 +77,15 @@ template <class L> struct TnStage {
   }
 };
 
+typedef __attribute__((ext_vector_type(16))) float float16v;
+
 template <class SA, class SB, bool C_F32>
 __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, long ldc,
-    int tiles_n, int kt_per_split, long split_stride) {
+    int tiles_n, int kt_per_split, long split_stride, int xcd_cpx) {
   int tile = blockIdx.x;
+  if (xcd_cpx) // T1: contiguous tile chunk per XCD (L2 reuse of panels)
+    tile = (tile & 7) * xcd_cpx + (tile >> 3);
   int split = blockIdx.y;
   int tm = tile / tiles_n, tn = tile % tiles_n;
   int row0 = tm * BM, col0 = tn * BN;
@@ -91,11 +95,9 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
 
   __shared__ ushort8 lds[2][2][BM * MXP]; // [buf][A|B][image]
 
-  float4v acc[4][4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
+  // 32x32x16 MFMA (higher ceiling than 16x16x32): each wave computes a
+  // 64x64 quadrant as 2x2 fragments of 32x32, 16 fp32 accumulators each.
+  float16v acc[2][2] = {};
 
   int nk_total = (K + BK - 1) / BK;
   int t0 = split * kt_per_split;
@@ -119,20 +121,21 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
       sb.load(tid, col0, (t0 + t + 1) * BK);
     }
 #pragma unroll
-    for (int kk = 0; kk < BK / 32; ++kk) {
-      bf16x8 af[4], bf_[4];
-      int slot = kk * 4 + (lane >> 4);
+    for (int kk = 0; kk < BK / 16; ++kk) { // 4 k-steps of 16
+      bf16x8 af[2], bf_[2];
+      // lane l: row = l&31, k = kk*16 + (l>>5)*8 .. +7 → slot kk*2+(l>>5)
+      int slot = kk * 2 + (lane >> 5);
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        af[mi] = us8_to_bf8v(lds[buf][0][(wr * 64 + mi * 16 + (lane & 15)) * MXP + slot]);
+      for (int mi = 0; mi < 2; ++mi)
+        af[mi] = us8_to_bf8v(lds[buf][0][(wr * 64 + mi * 32 + (lane & 31)) * MXP + slot]);
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        bf_[ni] = us8_to_bf8v(lds[buf][1][(wc * 64 + ni * 16 + (lane & 15)) * MXP + slot]);
+      for (int ni = 0; ni < 2; ++ni)
+        bf_[ni] = us8_to_bf8v(lds[buf][1][(wc * 64 + ni * 32 + (lane & 31)) * MXP + slot]);
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
     }
     if (t + 1 < nk) { // write-late
@@ -142,16 +145,17 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     __syncthreads();
   }
 
+  // 32x32 C/D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
   cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = col0 + wc * 64 + ni * 32 + (lane & 31);
       if (col >= N) continue;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = row0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
         if (row >= M) continue;
         if (C_F32)
           ((float *)cptr)[(long)row * ldc + col] = acc[mi][ni][r];
@@ -171,14 +175,28 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
   if (splits > nk) splits = nk > 0 ? nk : 1;
   int kts = (nk + splits - 1) / splits;
   long split_stride = (long)M * ldc;
-  dim3 grid(tiles_m * tiles_n, splits);
+  int nwg = tiles_m * tiles_n;
+  // T1 XCD swizzle: give each XCD a contiguous chunk of tiles so neighbor
+  // tiles (sharing operand panels) hit the same per-XCD L2. Needs nwg%8==0
+  // and enough tiles to matter.
+  int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
+  dim3 grid(nwg, splits);
   if (c_f32)
     mix_gemm_k<SA, SB, true><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride);
+        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride, cpx);
   else
     mix_gemm_k<SA, SB, false><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride);
+        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride, cpx);
   return hipGetLastError();
+}
+
+// Plain NT×NT entry (both operands k-contiguous row-major).
+template <class LA, class LB>
+static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
+                                 int N, int K, long ldc, bool c_f32,
+                                 hipStream_t s, int splits = 1) {
+  return launch_mix_gemm(NtStage<LA>{la}, NtStage<LB>{lb}, c, M, N, K, ldc,
+                         c_f32, s, splits);
 }
 
 // ---- TN loaders ----

@@ -38,10 +38,14 @@ __global__ void softmax_xent_fwd_k(const uint16_t *__restrict__ logits,
   }
 }
 
+// dscale is read from DEVICE memory (the incoming dloss scalar): reading it
+// on the host would force a per-step D2H sync and break hipGraph capture.
 __global__ void softmax_xent_bwd_k(const float *__restrict__ probs,
                                    const long *__restrict__ target,
                                    uint16_t *__restrict__ dlogits, int B, int V,
-                                   float scale) {
+                                   const float *__restrict__ dscale,
+                                   float inv_b) {
+  float scale = *dscale * inv_b;
   for (long t = blockIdx.x * blockDim.x + threadIdx.x; t < (long)B * V;
        t += (long)gridDim.x * blockDim.x) {
     int b = t / V;
@@ -65,7 +69,7 @@ extern "C" hipError_t softmax_xent_fwd_launch(const void *logits,
 
 extern "C" hipError_t softmax_xent_bwd_launch(const float *probs,
                                               const long *target, void *dlogits,
-                                              int B, int V, float scale,
+                                              int B, int V, const float *dscale,
                                               hipStream_t s) {
   long tasks = (long)B * V;
   long blocks = (tasks + 255) / 256;
@@ -73,7 +77,7 @@ extern "C" hipError_t softmax_xent_bwd_launch(const float *probs,
   // loss is a batch MEAN: dlogits = (p - onehot) * dloss / B
   softmax_xent_bwd_k<<<(int)blocks, 256, 0, s>>>(probs, target,
                                                  (uint16_t *)dlogits, B, V,
-                                                 scale / B);
+                                                 dscale, 1.f / B);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

@@ -204,10 +204,13 @@ class SoftmaxCrossEntropyFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss):
         probs, target = ctx.saved_tensors
-        scale = float(dloss)
         if probs.is_cuda:
-            return hip_ext().softmax_xent_bwd(probs, target, scale).to(ctx.dtype), None
-        return ref.softmax_cross_entropy_bwd(probs, target, scale).to(ctx.dtype), None
+            # dloss stays on-device: reading it on the host would insert a
+            # per-step D2H sync and break hipGraph capture of the train step
+            d = hip_ext().softmax_xent_bwd(probs, target,
+                                           dloss.reshape(1).float().contiguous())
+            return d.to(ctx.dtype), None
+        return ref.softmax_cross_entropy_bwd(probs, target, float(dloss)).to(ctx.dtype), None
 
 
 def softmax_cross_entropy(logits, target):

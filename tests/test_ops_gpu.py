@@ -193,7 +193,7 @@ def test_softmax_xent():
     lr_, pr = ref.softmax_cross_entropy_fwd(logits.float().cpu(), tgt.cpu())
     assert abs(loss.item() - lr_.item()) < 2e-3 * abs(lr_.item()) + 1e-3
     assert relerr(probs.cpu(), pr) < 0.01
-    d = ext.softmax_xent_bwd(probs, tgt, 1.0)
+    d = ext.softmax_xent_bwd(probs, tgt, torch.ones(1, device="cuda"))
     dr = ref.softmax_cross_entropy_bwd(pr, tgt.cpu(), 1.0)
     assert relerr(d.cpu(), dr) < 0.02
 
