@@ -33,33 +33,45 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
     }
   }
   if (row_lane < rows_per_block) {
+    long stride = (long)gridDim.x * rows_per_block;
+    // two rows per iteration: two independent load chains in flight
+    // (single-chain version sat ~80% wave-parked on HBM latency)
     for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
-         row += (long)gridDim.x * rows_per_block) {
-      long off = row * C8 + cb;
-      float fx[8];
+         row += 2 * stride) {
+      long r2 = row + stride;
+      bool has2 = r2 < M;
+      long off = row * C8 + cb, off2 = r2 * C8 + cb;
+      float fx[8], fx2[8];
       ushort8 vx = x[off];
+      ushort8 vx2 = has2 ? x[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
       bf8_to_f8(vx, fx);
+      bf8_to_f8(vx2, fx2);
       if (WHAT == 0) {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          a0[j] += fx[j];            // fx here is x
-          a1[j] += fx[j] * fx[j];
+          a0[j] += fx[j] + fx2[j];
+          a1[j] += fx[j] * fx[j] + fx2[j] * fx2[j];
         }
       } else {
         ushort8 vdy = dy[off];
-        float fdy[8];
+        ushort8 vdy2 = has2 ? dy[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+        float fdy[8], fdy2[8];
         bf8_to_f8(vdy, fdy);
+        bf8_to_f8(vdy2, fdy2);
         if (relu) {
           ushort8 vy = y[off];
+          ushort8 vy2 = has2 ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
+          for (int j = 0; j < 8; ++j) {
             if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+            if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+          }
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float xhat = (fx[j] - mn[j]) * is[j];
-          a0[j] += fdy[j];
-          a1[j] += fdy[j] * xhat;
+          a0[j] += fdy[j] + fdy2[j];
+          a1[j] += fdy[j] * ((fx[j] - mn[j]) * is[j]) +
+                   fdy2[j] * ((fx2[j] - mn[j]) * is[j]);
         }
       }
     }
@@ -178,18 +190,27 @@ __global__ void bn_apply_k(const ushort8 *__restrict__ x,
     sc[j] = scale[cb * 8 + j];
     sh[j] = shift[cb * 8 + j];
   }
+  long stride = (long)gridDim.x * rows_per_block;
   for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
-       row += (long)gridDim.x * rows_per_block) {
-    long off = row * C8 + cb;
+       row += 2 * stride) {
+    long r2 = row + stride;
+    long off = row * C8 + cb, off2 = r2 * C8 + cb;
     ushort8 v = x[off];
-    float f[8];
+    ushort8 v2 = r2 < M ? x[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    float f[8], f2[8];
     bf8_to_f8(v, f);
+    bf8_to_f8(v2, f2);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       f[j] = f[j] * sc[j] + sh[j];
-      if (relu) f[j] = fmaxf(f[j], 0.f);
+      f2[j] = f2[j] * sc[j] + sh[j];
+      if (relu) {
+        f[j] = fmaxf(f[j], 0.f);
+        f2[j] = fmaxf(f2[j], 0.f);
+      }
     }
     y[off] = f8_to_bf8(f);
+    if (r2 < M) y[off2] = f8_to_bf8(f2);
   }
 }
 
@@ -218,25 +239,35 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
     b[j] = k2[c];
     c3[j] = k3[c];
   }
+  long stride = (long)gridDim.x * rows_per_block;
   for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
-       row += (long)gridDim.x * rows_per_block) {
-    long off = row * C8 + cb;
+       row += 2 * stride) {
+    long r2 = row + stride;
+    long off = row * C8 + cb, off2 = r2 * C8 + cb;
     ushort8 vdy = dy[off], vx = x[off];
-    float fdy[8], fx[8];
+    ushort8 vdy2 = r2 < M ? dy[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    ushort8 vx2 = r2 < M ? x[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+    float fdy[8], fx[8], fdy2[8], fx2[8];
     bf8_to_f8(vdy, fdy);
     bf8_to_f8(vx, fx);
+    bf8_to_f8(vdy2, fdy2);
+    bf8_to_f8(vx2, fx2);
     if (relu) {
       ushort8 vy = y[off];
+      ushort8 vy2 = r2 < M ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
+      for (int j = 0; j < 8; ++j) {
         if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+        if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+      }
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float xhat = (fx[j] - mn[j]) * is[j];
-      fdy[j] = a[j] * fdy[j] - b[j] - c3[j] * xhat;
+      fdy[j] = a[j] * fdy[j] - b[j] - c3[j] * ((fx[j] - mn[j]) * is[j]);
+      fdy2[j] = a[j] * fdy2[j] - b[j] - c3[j] * ((fx2[j] - mn[j]) * is[j]);
     }
     dx[off] = f8_to_bf8(fdy);
+    if (r2 < M) dx[off2] = f8_to_bf8(fdy2);
   }
 }
 

@@ -234,28 +234,43 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
                          strm);
 }
 
+// float4 lanes + two split-accumulators: the scalar version issued one
+// dependent 4 B load per split per element and sat 93% wave-parked.
 template <bool OUT_BF16>
-__global__ void splitk_reduce_k(const float *__restrict__ partial, int splits,
-                                long len, void *__restrict__ out) {
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < len;
+__global__ void splitk_reduce_k(const float4v *__restrict__ partial,
+                                int splits, long len4, void *__restrict__ out) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < len4;
        i += (long)gridDim.x * blockDim.x) {
-    float a = 0;
-    for (int s = 0; s < splits; ++s) a += partial[(long)s * len + i];
-    if (OUT_BF16)
-      ((uint16_t *)out)[i] = f2bf(a);
-    else
-      ((float *)out)[i] = a;
+    float4v a = partial[i];
+    float4v b = {0.f, 0.f, 0.f, 0.f};
+    int s = 1;
+    for (; s + 1 < splits; s += 2) {
+      a += partial[(long)s * len4 + i];
+      b += partial[(long)(s + 1) * len4 + i];
+    }
+    if (s < splits) b += partial[(long)s * len4 + i];
+    a += b;
+    if (OUT_BF16) {
+      uint16_t *o = (uint16_t *)out + i * 4;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = f2bf(a[j]);
+    } else {
+      ((float4v *)out)[i] = a;
+    }
   }
 }
 
 extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
                                     void *out, int out_bf16, hipStream_t s) {
-  long blocks = (len + 255) / 256;
+  long len4 = len / 4; // len = Kout*RSC, both %8 ⇒ %4
+  long blocks = (len4 + 255) / 256;
   if (blocks > 2048) blocks = 2048;
   if (out_bf16)
-    splitk_reduce_k<true><<<(int)blocks, 256, 0, s>>>(partial, splits, len, out);
+    splitk_reduce_k<true><<<(int)blocks, 256, 0, s>>>((const float4v *)partial,
+                                                      splits, len4, out);
   else
-    splitk_reduce_k<false><<<(int)blocks, 256, 0, s>>>(partial, splits, len, out);
+    splitk_reduce_k<false><<<(int)blocks, 256, 0, s>>>((const float4v *)partial,
+                                                       splits, len4, out);
   return hipGetLastError();
 }
 

@@ -61,10 +61,15 @@ def init(backend: str | None = None, timeout_s: int = 300) -> None:
     os.environ.setdefault("MASTER_PORT", os.environ.get("MPIAMD_MASTER_PORT", "29500"))
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    kw = {}
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank())
+        if backend == "nccl":
+            # bind the device at init: eager RCCL communicator creation, and
+            # a bound PG is what makes collectives hipGraph-capturable
+            kw["device_id"] = torch.device(f"cuda:{local_rank()}")
     dist.init_process_group(backend=backend, init_method="env://",
-                            timeout=datetime.timedelta(seconds=timeout_s))
+                            timeout=datetime.timedelta(seconds=timeout_s), **kw)
     _initialized = True
 
 
