@@ -104,39 +104,25 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   int nk = min(kt_per_split, nk_total - t0);
   if (nk < 0) nk = 0;
 
-  // Distance-2 register pipeline: TWO in-flight register sets per operand.
-  // Tile t+2's global loads are issued at the START of step t (≈two compute
-  // bodies ≈ 1100 cyc of cover for the ~900-cyc HBM latency — distance-1
-  // left the loads ~300 cyc short and parked the waves; PMC showed
-  // SQ_WAIT_ANY 40-55% with MfmaUtil <15%). Even-parity tiles live in
-  // sa/sb, odd-parity in sa2/sb2; a set's registers are reused two steps
-  // after its LDS write.
+  // Register staging, issue-early / write-late (distance-1). A distance-2
+  // two-register-set variant was measured SLOWER on the gather stagers
+  // (address-compute duplication outweighed the extra latency cover) and
+  // neutral on plain ones — don't re-add it.
   sa.init(tid, row0);
   sb.init(tid, col0);
-  SA sa2 = sa;
-  SB sb2 = sb;
   if (nk > 0) {
     sa.load(tid, row0, t0 * BK);
     sb.load(tid, col0, t0 * BK);
     sa.write(tid, lds[0][0]);
     sb.write(tid, lds[0][1]);
   }
-  if (nk > 1) {
-    sa2.load(tid, row0, (t0 + 1) * BK);
-    sb2.load(tid, col0, (t0 + 1) * BK);
-  }
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     int buf = t & 1;
-    if (t + 2 < nk) { // issue tile t+2 into the set already drained to LDS
-      if (buf == 0) {
-        sa.load(tid, row0, (t0 + t + 2) * BK);
-        sb.load(tid, col0, (t0 + t + 2) * BK);
-      } else {
-        sa2.load(tid, row0, (t0 + t + 2) * BK);
-        sb2.load(tid, col0, (t0 + t + 2) * BK);
-      }
+    if (t + 1 < nk) { // issue-early: HBM latency hides under the MFMAs
+      sa.load(tid, row0, (t0 + t + 1) * BK);
+      sb.load(tid, col0, (t0 + t + 1) * BK);
     }
 #pragma unroll
     for (int kk = 0; kk < BK / 16; ++kk) { // 4 k-steps of 16
@@ -156,14 +142,9 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
     }
-    if (t + 1 < nk) { // write-late: tile t+1's set (loaded at step t-1)
-      if (buf == 0) {
-        sa2.write(tid, lds[buf ^ 1][0]);
-        sb2.write(tid, lds[buf ^ 1][1]);
-      } else {
-        sa.write(tid, lds[buf ^ 1][0]);
-        sb.write(tid, lds[buf ^ 1][1]);
-      }
+    if (t + 1 < nk) { // write-late
+      sa.write(tid, lds[buf ^ 1][0]);
+      sb.write(tid, lds[buf ^ 1][1]);
     }
     __syncthreads();
   }

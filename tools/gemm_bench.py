@@ -43,6 +43,34 @@ def main():
         del a, b
     torch.cuda.empty_cache()
 
+    # real ResNet conv shapes through the full fwd/dgrad/wgrad paths
+    convs = [
+        (64, 256, 56, 56, 64, 1, 1, 0, "b1 1x1 down (256->64 @56)"),
+        (64, 64, 56, 56, 64, 3, 1, 1, "b1 3x3 (64 @56)"),
+        (64, 512, 28, 28, 128, 1, 1, 0, "b2 1x1 down"),
+        (64, 128, 28, 28, 128, 3, 1, 1, "b2 3x3"),
+        (64, 256, 28, 28, 1024, 1, 1, 0, "b3 1x1 up (256->1024 @28... )"),
+        (64, 1024, 14, 14, 256, 1, 1, 0, "b3 1x1 down"),
+        (64, 256, 14, 14, 256, 3, 1, 1, "b3 3x3"),
+        (64, 512, 7, 7, 512, 3, 1, 1, "b4 3x3"),
+        (64, 256, 56, 56, 128, 1, 2, 0, "s2 1x1 (stage transition)"),
+    ]
+    for N, C, H, W, Kout, R, st, pad, name in convs:
+        HO = (H + 2 * pad - R) // st + 1
+        x = (torch.rand(N, C, H, W, device="cuda") * 2 - 1).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        w = ((torch.rand(Kout, C, R, R, device="cuda") * 2 - 1) * 0.1).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        dy = (torch.rand(N, Kout, HO, HO, device="cuda") * 2 - 1).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        gf = 2.0 * N * HO * HO * Kout * R * R * C / 1e12
+        tfw = gf / bench(lambda: ext.conv2d_fwd(x, w, st, pad))
+        tfd = gf / bench(lambda: ext.conv2d_dgrad(dy, w, H, W, st, pad))
+        tfg = gf / bench(lambda: ext.conv2d_wgrad(x, dy, R, R, st, pad))
+        print(f"conv     {name:28s} fwd {tfw:6.1f}  dgrad {tfd:6.1f}  wgrad {tfg:6.1f} TF")
+        del x, w, dy
+        torch.cuda.empty_cache()
+
 
 if __name__ == "__main__":
     main()
