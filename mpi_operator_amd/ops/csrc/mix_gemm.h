@@ -79,9 +79,28 @@ template <class L> struct TnStage {
 
 typedef __attribute__((ext_vector_type(16))) float float16v;
 
-template <class SA, class SB, bool C_F32>
+// Epilogue address map: row/col → element offset in the output. The
+// default is a plain row-major matrix; Stride2Writer scatters a parity
+// sub-image back into an NHWC tensor (stride-2 dgrad decomposition).
+struct LinearWriter {
+  long ldc;
+  DEV_INLINE long addr(int row, int col) const { return (long)row * ldc + col; }
+};
+
+struct Stride2Writer {
+  int W2, H2, ph, pw, W, H, C;
+  DEV_INLINE long addr(int row, int col) const {
+    int w_ = row % W2;
+    long t = row / W2;
+    int h_ = (int)(t % H2);
+    int n = (int)(t / H2);
+    return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C + col;
+  }
+};
+
+template <class SA, class SB, bool C_F32, class WR = LinearWriter>
 __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
-    SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, long ldc,
+    SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
     int tiles_n, int kt_per_split, long split_stride, int xcd_cpx) {
   int tile = blockIdx.x;
   if (xcd_cpx) // T1: contiguous tile chunk per XCD (L2 reuse of panels)
@@ -161,19 +180,21 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
       for (int r = 0; r < 16; ++r) {
         int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
         if (row >= M) continue;
+        long a = wrt.addr(row, col);
         if (C_F32)
-          ((float *)cptr)[(long)row * ldc + col] = acc[mi][ni][r];
+          ((float *)cptr)[a] = acc[mi][ni][r];
         else
-          ((uint16_t *)cptr)[(long)row * ldc + col] = f2bf(acc[mi][ni][r]);
+          ((uint16_t *)cptr)[a] = f2bf(acc[mi][ni][r]);
       }
     }
   }
 }
 
-template <class SA, class SB>
-static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
-                                  int N, int K, long ldc, bool c_f32,
-                                  hipStream_t s, int splits = 1) {
+template <class SA, class SB, class WR>
+static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
+                                     int M, int N, int K, const WR &wrt,
+                                     long ldc, bool c_f32, hipStream_t s,
+                                     int splits = 1) {
   int tiles_m = (M + BM - 1) / BM, tiles_n = (N + BN - 1) / BN;
   int nk = (K + BK - 1) / BK;
   if (splits > nk) splits = nk > 0 ? nk : 1;
@@ -186,12 +207,20 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
   dim3 grid(nwg, splits);
   if (c_f32)
-    mix_gemm_k<SA, SB, true><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride, cpx);
+    mix_gemm_k<SA, SB, true, WR><<<grid, NT_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
   else
-    mix_gemm_k<SA, SB, false><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, ldc, tiles_n, kts, split_stride, cpx);
+    mix_gemm_k<SA, SB, false, WR><<<grid, NT_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
   return hipGetLastError();
+}
+
+template <class SA, class SB>
+static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
+                                  int N, int K, long ldc, bool c_f32,
+                                  hipStream_t s, int splits = 1) {
+  return launch_mix_gemm_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc, c_f32,
+                            s, splits);
 }
 
 // Plain NT×NT entry (both operands k-contiguous row-major).
