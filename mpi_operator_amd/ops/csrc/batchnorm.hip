@@ -175,8 +175,11 @@ __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
   k3[c] = g_is * s1 * inv_m;
 }
 
-// apply scale/shift (+ReLU): fwd-train, fwd-eval share this
+// apply scale/shift (+optional residual add) (+ReLU): fwd-train, fwd-eval
+// and the bottleneck-join fusion (res != nullptr folds the skip connection
+// into this pass — one fewer full activation read+write per block).
 __global__ void bn_apply_k(const ushort8 *__restrict__ x,
+                           const ushort8 *__restrict__ res,
                            const float *__restrict__ scale,
                            const float *__restrict__ shift,
                            ushort8 *__restrict__ y, long M, int C8, int relu) {
@@ -197,13 +200,19 @@ __global__ void bn_apply_k(const ushort8 *__restrict__ x,
     long off = row * C8 + cb, off2 = r2 * C8 + cb;
     ushort8 v = x[off];
     ushort8 v2 = r2 < M ? x[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-    float f[8], f2[8];
+    float f[8], f2[8], g[8] = {0}, g2[8] = {0};
     bf8_to_f8(v, f);
     bf8_to_f8(v2, f2);
+    if (res) {
+      ushort8 w = res[off];
+      ushort8 w2 = r2 < M ? res[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+      bf8_to_f8(w, g);
+      bf8_to_f8(w2, g2);
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      f[j] = f[j] * sc[j] + sh[j];
-      f2[j] = f2[j] * sc[j] + sh[j];
+      f[j] = f[j] * sc[j] + sh[j] + g[j];
+      f2[j] = f2[j] * sc[j] + sh[j] + g2[j];
       if (relu) {
         f[j] = fmaxf(f[j], 0.f);
         f2[j] = fmaxf(f2[j], 0.f);
@@ -287,10 +296,10 @@ static int bn_apply_grid(long M, int C8) {
 }
 
 extern "C" hipError_t bn_fwd_train_launch(
-    const void *x, const float *gamma, const float *beta, float eps, int relu,
-    void *y, float *mean, float *invstd, float *scale, float *shift,
-    float *partial, float *running_mean, float *running_var, float momentum,
-    long M, int C, hipStream_t s) {
+    const void *x, const void *res, const float *gamma, const float *beta,
+    float eps, int relu, void *y, float *mean, float *invstd, float *scale,
+    float *shift, float *partial, float *running_mean, float *running_var,
+    float momentum, long M, int C, hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
@@ -304,7 +313,8 @@ extern "C" hipError_t bn_fwd_train_launch(
       shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)x, scale, shift, (ushort8 *)y, M, C8, relu);
+      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y, M,
+      C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -317,7 +327,7 @@ extern "C" hipError_t bn_fwd_eval_launch(const void *x, const float *scale,
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)x, scale, shift, (ushort8 *)y, M, C8, relu);
+      (const ushort8 *)x, nullptr, scale, shift, (ushort8 *)y, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

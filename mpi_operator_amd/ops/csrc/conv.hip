@@ -218,7 +218,7 @@ struct DgradS2Stage {
   const uint16_t *dy;
   int HO, WO, Q, W2, H2, K;
   long M;
-  int dh[2], dw[2], nth, ntw; // per-parity tap shifts (≤2 each for R≤3)
+  int dh[4], dw[4], nth, ntw; // per-parity tap shifts (≤4 each for R≤7)
   ushort8 r[4];
   long nbase_[4];
   int hb_[4], wb_[4];
@@ -264,7 +264,7 @@ struct DgradS2Stage {
 struct DgradWS2Tn {
   const uint16_t *w;
   int C, Q, K, RSC, S;
-  int rr[2], ss[2], nth, ntw;
+  int rr[4], ss[4], nth, ntw;
   DEV_INLINE ushort8 load(int k, int c0) const {
     if (k >= K || c0 >= C) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
     int q = k % Q, ti = k / Q;
@@ -277,15 +277,17 @@ static hipError_t conv_dgrad_s2(const void *dy, const void *w, void *dx,
                                 int N, int H, int W, int C, int Kout, int R,
                                 int S, int pad, int HO, int WO,
                                 hipStream_t strm) {
-  bool any_empty = false;
-  for (int ph = 0; ph < 2 && !any_empty; ++ph)
+  int n_live = 0;
+  for (int ph = 0; ph < 2; ++ph)
     for (int pw = 0; pw < 2; ++pw) {
       int nth = 0, ntw = 0;
       for (int r = (ph + pad) & 1; r < R; r += 2) nth++;
       for (int s = (pw + pad) & 1; s < S; s += 2) ntw++;
-      if (nth == 0 || ntw == 0) { any_empty = true; break; }
+      if (nth > 0 && ntw > 0) n_live++;
     }
-  if (any_empty) // e.g. 1x1 s2: three parities receive nothing
+  // n_live==1 (1x1 s2): the single parity's epilogue zero-fills its three
+  // sibling pixels (Stride2ZeroWriter) — no full-tensor memset pass.
+  if (n_live != 1 && n_live != 4) // exotic shapes: zero the dead parities
     (void)hipMemsetAsync(dx, 0, (long)N * H * W * C * 2, strm);
 
   for (int ph = 0; ph < 2; ++ph) {
@@ -315,9 +317,16 @@ static hipError_t conv_dgrad_s2(const void *dy, const void *w, void *dx,
       lb.w = (const uint16_t *)w;
       lb.C = C; lb.Q = Kout; lb.K = K; lb.RSC = R * S * C; lb.S = S;
       lb.nth = sa.nth; lb.ntw = sa.ntw;
-      Stride2Writer wrt{W2, H2, ph, pw, W, H, C};
-      hipError_t e = launch_mix_gemm_wr(sa, TnStage<DgradWS2Tn>{lb}, dx,
-                                        (int)M, C, K, wrt, C, false, strm);
+      hipError_t e;
+      if (n_live == 1) {
+        Stride2ZeroWriter wrt{W2, H2, ph, pw, W, H, C};
+        e = launch_mix_gemm_wr(sa, TnStage<DgradWS2Tn>{lb}, dx, (int)M, C, K,
+                               wrt, C, false, strm);
+      } else {
+        Stride2Writer wrt{W2, H2, ph, pw, W, H, C};
+        e = launch_mix_gemm_wr(sa, TnStage<DgradWS2Tn>{lb}, dx, (int)M, C, K,
+                               wrt, C, false, strm);
+      }
       if (e != hipSuccess) return e;
     }
   }

@@ -83,11 +83,26 @@ typedef __attribute__((ext_vector_type(16))) float float16v;
 // default is a plain row-major matrix; Stride2Writer scatters a parity
 // sub-image back into an NHWC tensor (stride-2 dgrad decomposition).
 struct LinearWriter {
+  static constexpr bool ACC = false;
+  static constexpr bool ZSIB = false;
   long ldc;
   DEV_INLINE long addr(int row, int col) const { return (long)row * ldc + col; }
+  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
+};
+
+// += into an existing bf16 tensor (bottleneck backward: conv1's dgrad
+// accumulates onto the skip-connection gradient — no separate add pass).
+struct LinearAccWriter {
+  static constexpr bool ACC = true;
+  static constexpr bool ZSIB = false;
+  long ldc;
+  DEV_INLINE long addr(int row, int col) const { return (long)row * ldc + col; }
+  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
 };
 
 struct Stride2Writer {
+  static constexpr bool ACC = false;
+  static constexpr bool ZSIB = false;
   int W2, H2, ph, pw, W, H, C;
   DEV_INLINE long addr(int row, int col) const {
     int w_ = row % W2;
@@ -95,6 +110,35 @@ struct Stride2Writer {
     int h_ = (int)(t % H2);
     int n = (int)(t / H2);
     return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C + col;
+  }
+  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
+};
+
+// 1x1 stride-2 dgrad: parity (pad%2, pad%2) is the ONLY contributor, so the
+// epilogue zeroes the three sibling pixels itself — no 100 MB memset pass.
+struct Stride2ZeroWriter {
+  static constexpr bool ACC = false;
+  static constexpr bool ZSIB = true;
+  int W2, H2, ph, pw, W, H, C;
+  DEV_INLINE long addr(int row, int col) const {
+    int w_ = row % W2;
+    long t = row / W2;
+    int h_ = (int)(t % H2);
+    int n = (int)(t / H2);
+    return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C + col;
+  }
+  DEV_INLINE void zero_siblings(uint16_t *p, int row, int col) const {
+    int w_ = row % W2;
+    long t = row / W2;
+    int h_ = (int)(t % H2);
+    int n = (int)(t / H2);
+    int hh = 2 * h_ + ph, ww = 2 * w_ + pw;
+    long base = ((long)n * H + hh) * W;
+    if (ww + 1 < W) p[(base + ww + 1) * C + col] = 0;
+    if (hh + 1 < H) {
+      p[(base + W + ww) * C + col] = 0;
+      if (ww + 1 < W) p[(base + W + ww + 1) * C + col] = 0;
+    }
   }
 };
 
@@ -181,10 +225,16 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
         int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
         if (row >= M) continue;
         long a = wrt.addr(row, col);
-        if (C_F32)
-          ((float *)cptr)[a] = acc[mi][ni][r];
-        else
-          ((uint16_t *)cptr)[a] = f2bf(acc[mi][ni][r]);
+        if (C_F32) {
+          float v = acc[mi][ni][r];
+          if (WR::ACC) v += ((float *)cptr)[a];
+          ((float *)cptr)[a] = v;
+        } else {
+          float v = acc[mi][ni][r];
+          if (WR::ACC) v += bf2f(((uint16_t *)cptr)[a]);
+          ((uint16_t *)cptr)[a] = f2bf(v);
+          if (WR::ZSIB) wrt.zero_siblings((uint16_t *)cptr, row, col);
+        }
       }
     }
   }
