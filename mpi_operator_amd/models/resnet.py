@@ -34,6 +34,11 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
+        if x.is_cuda and self.training and torch.is_grad_enabled():
+            # fused block: residual add folded into bn3's apply pass, join
+            # gradient summed inside conv1's dgrad epilogue (ops/fused_block)
+            from ..ops.fused_block import fused_bottleneck
+            return fused_bottleneck(x, self)
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))

@@ -56,10 +56,10 @@ struct SgdDesc {
 };
 hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
                            hipStream_t);
-hipError_t bn_fwd_train_launch(const void *, const float *, const float *,
-                               float, int, void *, float *, float *, float *,
-                               float *, float *, float *, float *, float, long,
-                               int, hipStream_t);
+hipError_t bn_fwd_train_launch(const void *, const void *, const float *,
+                               const float *, float, int, void *, float *,
+                               float *, float *, float *, float *, float *,
+                               float *, float, long, int, hipStream_t);
 hipError_t bn_fwd_eval_launch(const void *, const float *, const float *, int,
                               void *, long, int, hipStream_t);
 hipError_t bn_bwd_launch(const void *, const void *, const void *,
@@ -85,6 +85,8 @@ hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, hipStream_t);
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
                       int, int, int, int, int, int, int, hipStream_t);
+hipError_t conv_dgrad_1x1_acc(const void *, const void *, void *, long, int,
+                              int, hipStream_t);
 hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
                                int, int, int, int, int, int, int, int, int,
                                int, int, int, int, hipStream_t);
@@ -148,12 +150,26 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   return dw;
 }
 
+// dx_acc += dgrad(dy, w) for a 1x1 stride-1 conv (bottleneck conv1)
+static void conv2d_dgrad_acc(const Tensor &dy, const Tensor &w, Tensor &dx_acc) {
+  check_cl_bf16(dy, "dy");
+  check_cl_bf16(w, "w");
+  check_cl_bf16(dx_acc, "dx_acc");
+  const HIPDeviceGuard guard(dy.device());
+  TORCH_CHECK(w.size(2) == 1 && w.size(3) == 1, "acc dgrad is 1x1-only");
+  long M = (long)dy.size(0) * dy.size(2) * dy.size(3);
+  CHK(conv_dgrad_1x1_acc(dy.data_ptr(), w.data_ptr(), dx_acc.data_ptr(), M,
+                         (int)w.size(1), (int)dy.size(1), cur_stream()));
+}
+
 // ------------------------- batchnorm -------------------------
+// res (optional, may be undefined): residual tensor folded into the apply
+// pass — y = [relu](bn(x) + res), the bottleneck-join fusion.
 static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
                                         const Tensor &beta, double eps,
                                         bool relu, const Tensor &running_mean,
                                         const Tensor &running_var,
-                                        double momentum) {
+                                        double momentum, const Tensor &res) {
   check_cl_bf16(x, "x");
   const HIPDeviceGuard guard(x.device());
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
@@ -166,7 +182,12 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
   float *rm = running_mean.defined() && running_mean.numel() == C
                   ? running_mean.data_ptr<float>() : nullptr;
   float *rv = rm ? running_var.data_ptr<float>() : nullptr;
-  CHK(bn_fwd_train_launch(x.data_ptr(), gamma.data_ptr<float>(),
+  const void *resp = nullptr;
+  if (res.defined() && res.numel() > 0) {
+    check_cl_bf16(res, "res");
+    resp = res.data_ptr();
+  }
+  CHK(bn_fwd_train_launch(x.data_ptr(), resp, gamma.data_ptr<float>(),
                           beta.data_ptr<float>(), (float)eps, relu ? 1 : 0,
                           y.data_ptr(), mean.data_ptr<float>(),
                           invstd.data_ptr<float>(), scale.data_ptr<float>(),
@@ -388,6 +409,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("conv2d_dgrad_acc", &conv2d_dgrad_acc);
   m.def("bn_fwd_train", &bn_fwd_train);
   m.def("bn_fwd_eval", &bn_fwd_eval);
   m.def("bn_bwd", &bn_bwd);

@@ -197,6 +197,19 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
                          false, strm);
 }
 
+// 1x1 stride-1 dgrad that ACCUMULATES into dx (bottleneck backward: the
+// conv1 input-grad lands directly on the skip-connection gradient, so the
+// residual join needs no separate elementwise add pass).
+extern "C" hipError_t conv_dgrad_1x1_acc(const void *dy, const void *w,
+                                         void *dx_acc, long M, int C, int Kout,
+                                         hipStream_t strm) {
+  GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
+  TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
+  return launch_mix_gemm_wr(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                            dx_acc, (int)M, C, Kout, LinearAccWriter{(long)C},
+                            C, false, strm);
+}
+
 struct DgradWTn {
   const uint16_t *w;
   int C, Q, K, RSC;

@@ -72,7 +72,8 @@ class BNReLUFn(torch.autograd.Function):
             y, mean, invstd = hip_ext().bn_fwd_train(
                 x, gamma, beta, eps, relu,
                 running_mean if running_mean is not None else empty,
-                running_var if running_var is not None else empty, momentum)
+                running_var if running_var is not None else empty, momentum,
+                empty)
         else:
             y, mean, invstd = ref.bn_relu_fwd_train(x, gamma, beta, eps, relu)
             if running_mean is not None:
