@@ -82,62 +82,84 @@ typedef __attribute__((ext_vector_type(16))) float float16v;
 // Epilogue address map: row/col → element offset in the output. The
 // default is a plain row-major matrix; Stride2Writer scatters a parity
 // sub-image back into an NHWC tensor (stride-2 dgrad decomposition).
+// Writer concept: RowCtx row_ctx(row) hoists the per-row address work out
+// of the column loop; store_*(ptr, ctx, col, v) places one value.
 struct LinearWriter {
   static constexpr bool ACC = false;
-  static constexpr bool ZSIB = false;
   long ldc;
-  DEV_INLINE long addr(int row, int col) const { return (long)row * ldc + col; }
-  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] = v;
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    p[b + col] = f2bf(v);
+  }
 };
 
 // += into an existing bf16 tensor (bottleneck backward: conv1's dgrad
 // accumulates onto the skip-connection gradient — no separate add pass).
 struct LinearAccWriter {
   static constexpr bool ACC = true;
-  static constexpr bool ZSIB = false;
   long ldc;
-  DEV_INLINE long addr(int row, int col) const { return (long)row * ldc + col; }
-  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] += v;
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    p[b + col] = f2bf(v + bf2f(p[b + col]));
+  }
 };
 
 struct Stride2Writer {
   static constexpr bool ACC = false;
-  static constexpr bool ZSIB = false;
   int W2, H2, ph, pw, W, H, C;
-  DEV_INLINE long addr(int row, int col) const {
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const {
     int w_ = row % W2;
     long t = row / W2;
     int h_ = (int)(t % H2);
     int n = (int)(t / H2);
-    return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C + col;
+    return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C;
   }
-  DEV_INLINE void zero_siblings(uint16_t *, int, int) const {}
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] = v;
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    p[b + col] = f2bf(v);
+  }
 };
 
 // 1x1 stride-2 dgrad: parity (pad%2, pad%2) is the ONLY contributor, so the
 // epilogue zeroes the three sibling pixels itself — no 100 MB memset pass.
 struct Stride2ZeroWriter {
   static constexpr bool ACC = false;
-  static constexpr bool ZSIB = true;
   int W2, H2, ph, pw, W, H, C;
-  DEV_INLINE long addr(int row, int col) const {
-    int w_ = row % W2;
-    long t = row / W2;
-    int h_ = (int)(t % H2);
-    int n = (int)(t / H2);
-    return (((long)n * H + 2 * h_ + ph) * W + 2 * w_ + pw) * (long)C + col;
-  }
-  DEV_INLINE void zero_siblings(uint16_t *p, int row, int col) const {
+  struct RowCtx {
+    long base;
+    int sib; // bit0: w+1 valid, bit1: h+1 valid
+  };
+  DEV_INLINE RowCtx row_ctx(int row) const {
     int w_ = row % W2;
     long t = row / W2;
     int h_ = (int)(t % H2);
     int n = (int)(t / H2);
     int hh = 2 * h_ + ph, ww = 2 * w_ + pw;
-    long base = ((long)n * H + hh) * W;
-    if (ww + 1 < W) p[(base + ww + 1) * C + col] = 0;
-    if (hh + 1 < H) {
-      p[(base + W + ww) * C + col] = 0;
-      if (ww + 1 < W) p[(base + W + ww + 1) * C + col] = 0;
+    RowCtx c;
+    c.base = (((long)n * H + hh) * W + ww) * (long)C;
+    c.sib = (ww + 1 < W ? 1 : 0) | (hh + 1 < H ? 2 : 0);
+    return c;
+  }
+  DEV_INLINE void store_f32(float *p, RowCtx c, int col, float v) const {
+    p[c.base + col] = v; // (f32 split-K path never uses ZSIB)
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx c, int col, float v) const {
+    p[c.base + col] = f2bf(v);
+    if (c.sib & 1) p[c.base + C + col] = 0;
+    if (c.sib & 2) {
+      p[c.base + (long)W * C + col] = 0;
+      if (c.sib & 1) p[c.base + (long)W * C + C + col] = 0;
     }
   }
 };
@@ -213,28 +235,23 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   }
 
   // 32x32 C/D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  // Row-major loop order so each row's address context is computed once.
   cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
-      int col = col0 + wc * 64 + ni * 32 + (lane & 31);
-      if (col >= N) continue;
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      if (row >= M) continue;
+      typename WR::RowCtx rc = wrt.row_ctx(row);
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-        if (row >= M) continue;
-        long a = wrt.addr(row, col);
-        if (C_F32) {
-          float v = acc[mi][ni][r];
-          if (WR::ACC) v += ((float *)cptr)[a];
-          ((float *)cptr)[a] = v;
-        } else {
-          float v = acc[mi][ni][r];
-          if (WR::ACC) v += bf2f(((uint16_t *)cptr)[a]);
-          ((uint16_t *)cptr)[a] = f2bf(v);
-          if (WR::ZSIB) wrt.zero_siblings((uint16_t *)cptr, row, col);
-        }
+      for (int ni = 0; ni < 2; ++ni) {
+        int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+        if (col >= N) continue;
+        if (C_F32)
+          wrt.store_f32((float *)cptr, rc, col, acc[mi][ni][r]);
+        else
+          wrt.store_bf16((uint16_t *)cptr, rc, col, acc[mi][ni][r]);
       }
     }
   }

@@ -38,8 +38,10 @@ def test_fused_bottleneck_matches_unfused(stride, ds):
     import torch.nn as nn
 
     torch.manual_seed(0)
-    in_ch, width = 64, 32
+    width = 32
     out_ch = width * 4
+    # identity blocks (no downsample) require in_ch == out_ch, as in ResNet
+    in_ch = out_ch if not ds else 64
     downsample = None
     if ds:
         downsample = nn.Sequential(Conv2d(in_ch, out_ch, 1, stride=stride),
