@@ -231,7 +231,10 @@ struct DgradS2Stage {
   const uint16_t *dy;
   int HO, WO, Q, W2, H2, K;
   long M;
-  int dh[4], dw[4], nth, ntw; // per-parity tap shifts (≤4 each for R≤7)
+  // Tap shifts are AFFINE in the tap index (taps step by 2): dh_i = dh0-i,
+  // dw_j = dw0-j. (A runtime-indexed table here allocated the whole stager
+  // in scratch memory — 384 B/lane, every inner load a scratch round-trip.)
+  int dh0, dw0, nth, ntw;
   ushort8 r[4];
   long nbase_[4];
   int hb_[4], wb_[4];
@@ -254,7 +257,8 @@ struct DgradS2Stage {
   DEV_INLINE void load(int tid, int, int kb) {
     int k = kb + (tid & 7) * 8;
     int q = k % Q, ti = k / Q;
-    int dho = dh[ti / ntw], dwo = dw[ti % ntw];
+    int ih = ti / ntw, iw = ti - ih * ntw;
+    int dho = dh0 - ih, dwo = dw0 - iw;
     bool kok = k < K;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
@@ -277,12 +281,13 @@ struct DgradS2Stage {
 struct DgradWS2Tn {
   const uint16_t *w;
   int C, Q, K, RSC, S;
-  int rr[4], ss[4], nth, ntw;
+  int r0, s0, nth, ntw; // parity taps: r = r0 + 2·ih, s = s0 + 2·iw
   DEV_INLINE ushort8 load(int k, int c0) const {
     if (k >= K || c0 >= C) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
     int q = k % Q, ti = k / Q;
-    int r_ = rr[ti / ntw], s_ = ss[ti % ntw];
-    return *(const ushort8 *)(w + (long)q * RSC + (r_ * S + s_) * C + c0);
+    int ih = ti / ntw, iw = ti - ih * ntw;
+    return *(const ushort8 *)(w + (long)q * RSC +
+                              ((r0 + 2 * ih) * S + s0 + 2 * iw) * C + c0);
   }
 };
 
@@ -309,18 +314,13 @@ static hipError_t conv_dgrad_s2(const void *dy, const void *w, void *dx,
       int W2 = (W - pw + 1) / 2;
       DgradS2Stage sa{};
       DgradWS2Tn lb{};
-      sa.nth = 0;
-      for (int r = (ph + pad) & 1; r < R; r += 2) {
-        lb.rr[sa.nth] = r;
-        sa.dh[sa.nth] = (ph + pad - r) / 2;
-        sa.nth++;
-      }
-      sa.ntw = 0;
-      for (int s = (pw + pad) & 1; s < S; s += 2) {
-        lb.ss[sa.ntw] = s;
-        sa.dw[sa.ntw] = (pw + pad - s) / 2;
-        sa.ntw++;
-      }
+      int r_first = (ph + pad) & 1, s_first = (pw + pad) & 1;
+      sa.nth = (R - r_first + 1) / 2;
+      sa.ntw = (S - s_first + 1) / 2;
+      sa.dh0 = (ph + pad - r_first) / 2;
+      sa.dw0 = (pw + pad - s_first) / 2;
+      lb.r0 = r_first;
+      lb.s0 = s_first;
       if (sa.nth == 0 || sa.ntw == 0 || H2 == 0 || W2 == 0) continue;
       long M = (long)N * H2 * W2;
       int K = sa.nth * sa.ntw * Kout;
