@@ -42,6 +42,7 @@ Tensor empty_cl_bf16(int64_t n, int64_t c, int64_t h, int64_t w, const Tensor &l
 // ---- extern "C" launchers from the .hip TUs ----
 extern "C" {
 hipError_t add_relu_fwd(const void *, const void *, void *, long, hipStream_t);
+hipError_t add_bf16(const void *, const void *, void *, long, hipStream_t);
 hipError_t add_relu_bwd(const void *, const void *, void *, long, hipStream_t);
 hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
@@ -148,6 +149,15 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
                           Kout, (int)R, (int)S, (int)stride, (int)pad, HO, WO,
                           splits, 1, cur_stream()));
   return dw;
+}
+
+// c = a + b, bf16 16B-vectorized (the residual-join gradient sum)
+static Tensor add_bf16_b(const Tensor &a, const Tensor &b) {
+  const HIPDeviceGuard guard(a.device());
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % 8 == 0);
+  Tensor c = at::empty_like(a);
+  CHK(add_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(), a.numel(), cur_stream()));
+  return c;
 }
 
 // dx_acc += dgrad(dy, w) for a 1x1 stride-1 conv (bottleneck conv1)
@@ -410,6 +420,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_dgrad_acc", &conv2d_dgrad_acc);
+  m.def("add_bf16", &add_bf16_b);
   m.def("bn_fwd_train", &bn_fwd_train);
   m.def("bn_fwd_eval", &bn_fwd_eval);
   m.def("bn_bwd", &bn_bwd);

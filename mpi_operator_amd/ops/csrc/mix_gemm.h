@@ -164,7 +164,11 @@ struct Stride2ZeroWriter {
   }
 };
 
-template <class SA, class SB, bool C_F32, class WR = LinearWriter>
+// ONEBUF: single LDS buffer (36 KB vs 72 KB) with a second barrier per
+// k-step — 4 workgroups/CU instead of 2 (16 waves/CU): trades one barrier
+// for 2x the latency-hiding wave pool. A/B via MPIAMD_GEMM_ONEBUF=0/1.
+template <class SA, class SB, bool C_F32, class WR = LinearWriter,
+          bool ONEBUF = false>
 __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
     int tiles_n, int kt_per_split, long split_stride, int xcd_cpx) {
@@ -178,7 +182,7 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   int lane = tid & 63, wave = tid >> 6;
   int wr = wave >> 1, wc = wave & 1;
 
-  __shared__ ushort8 lds[2][2][BM * MXP]; // [buf][A|B][image]
+  __shared__ ushort8 lds[ONEBUF ? 1 : 2][2][BM * MXP]; // [buf][A|B][image]
 
   // 32x32x16 MFMA (higher ceiling than 16x16x32): each wave computes a
   // 64x64 quadrant as 2x2 fragments of 32x32, 16 fp32 accumulators each.
@@ -204,7 +208,7 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
-    int buf = t & 1;
+    int buf = ONEBUF ? 0 : (t & 1);
     if (t + 1 < nk) { // issue-early: HBM latency hides under the MFMAs
       sa.load(tid, row0, (t0 + t + 1) * BK);
       sb.load(tid, col0, (t0 + t + 1) * BK);
@@ -227,7 +231,13 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
     }
-    if (t + 1 < nk) { // write-late
+    if (ONEBUF) {
+      if (t + 1 < nk) { // reuse the single buffer: drain-readers barrier,
+        __syncthreads(); // then overwrite with tile t+1
+        sa.write(tid, lds[0][0]);
+        sb.write(tid, lds[0][1]);
+      }
+    } else if (t + 1 < nk) { // write-late into the other buffer
       sa.write(tid, lds[buf ^ 1][0]);
       sb.write(tid, lds[buf ^ 1][1]);
     }

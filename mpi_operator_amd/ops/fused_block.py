@@ -85,12 +85,15 @@ class BottleneckFn(torch.autograd.Function):
         if ctx.has_ds:
             dad, dgd, dbd = ext.bn_bwd(g, ad, ad, gd, md, vd, False)
             dwd = ext.conv2d_wgrad(x, dad, 1, 1, st, 0)
-            dxt = ext.conv2d_dgrad(dad, wd, x.shape[2], x.shape[3], st, 0)
+            skip = ext.conv2d_dgrad(dad, wd, x.shape[2], x.shape[3], st, 0)
         else:
             dwd = dgd = dbd = None
-            dxt = g  # owned fresh tensor — safe to accumulate into
-        # conv1 dgrad accumulates straight onto the skip-connection grad
-        ext.conv2d_dgrad_acc(dx1, w1, dxt)
+            skip = g
+        # residual-join gradient sum: skip grad + conv1's input grad (one
+        # vectorized kernel; an in-epilogue accumulate measured slower —
+        # the RMW on one tensor serializes the epilogue on aliasing)
+        dx0 = ext.conv2d_dgrad(dx1, w1, x.shape[2], x.shape[3], 1, 0)
+        dxt = ext.add_bf16(skip, dx0)
 
         f32 = torch.float32
         return (dxt, None, None, None,
