@@ -283,6 +283,19 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   // and enough tiles to matter.
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
   dim3 grid(nwg, splits);
+  static const bool onebuf = [] {
+    const char *e = getenv("MPIAMD_GEMM_ONEBUF");
+    return e && e[0] == '1';
+  }();
+  if (onebuf) {
+    if (c_f32)
+      mix_gemm_k<SA, SB, true, WR, true><<<grid, NT_THREADS, 0, s>>>(
+          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+    else
+      mix_gemm_k<SA, SB, false, WR, true><<<grid, NT_THREADS, 0, s>>>(
+          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+    return hipGetLastError();
+  }
   if (c_f32)
     mix_gemm_k<SA, SB, true, WR><<<grid, NT_THREADS, 0, s>>>(
         sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
