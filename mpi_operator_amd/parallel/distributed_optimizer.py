@@ -94,16 +94,19 @@ class DistributedOptimizer:
         for dt in list(by_dtype):
             flush(dt)
 
-        # allocate flats and wire p.grad views
+        # allocate flats + per-param views. Grads are NOT pre-wired: leaving
+        # p.grad = None lets autograd ASSIGN each produced grad (no per-param
+        # accumulate kernel — ~2 launches per layer per step in the wired
+        # scheme); the bucket-completion hook then gathers all grads into the
+        # flat with ONE fused multi-tensor copy and rebinds p.grad to the
+        # views so the optimizer (and any user code) sees the reduced values.
         for b in self.buckets:
             n = sum(p.numel() for p in b.params)
             p0 = b.params[0]
             b.flat = torch.zeros(n, dtype=p0.dtype, device=p0.device)
             off = 0
             for p in b.params:
-                v = _strided_view(b.flat, off, p)
-                p.grad = v
-                b.views.append(v)
+                b.views.append(_strided_view(b.flat, off, p))
                 off += p.numel()
 
     def _register_hooks(self):
@@ -120,7 +123,22 @@ class DistributedOptimizer:
         return hook
 
     # -- comm ---------------------------------------------------------------
+    def _gather(self, bucket: _Bucket):
+        """Fuse the bucket's assigned grads into the flat; rebind p.grad to
+        the flat views (one multi-tensor copy instead of per-param adds)."""
+        srcs, dsts = [], []
+        for p, v in zip(bucket.params, bucket.views):
+            if p.grad is None:
+                v.zero_()
+            elif p.grad is not v:
+                dsts.append(v)
+                srcs.append(p.grad)
+            p.grad = v
+        if dsts:
+            torch._foreach_copy_(dsts, srcs)
+
     def _launch(self, bucket: _Bucket):
+        self._gather(bucket)
         if dist.is_available() and dist.is_initialized():
             bucket.handle = dist.all_reduce(bucket.flat, async_op=True, group=self.group)
 
@@ -149,13 +167,12 @@ class DistributedOptimizer:
         return self.optimizer.step(closure)
 
     def zero_grad(self, set_to_none: bool = False):
-        # grads are persistent bucket views — zero in place, never detach
+        # detach grads so the next backward ASSIGNS them (no accumulate
+        # kernels); the flat is overwritten by _gather, never zeroed
         for b in self.buckets:
-            b.flat.zero_()
             b.ready = 0
-            for p, v in zip(b.params, b.views):
-                if p.grad is not v:
-                    p.grad = v  # re-attach if user code replaced it
+            for p in b.params:
+                p.grad = None
 
     @property
     def param_groups(self):
