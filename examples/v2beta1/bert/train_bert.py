@@ -4,9 +4,14 @@
 rendezvous; RCCL stays on xGMI across pods because both pods share
 /dev/kfd). Synthetic MLM+NSP data, bf16, sequences/sec reported."""
 import argparse
+import os
+import sys
 import time
 
 import torch
+
+# allow running straight from a source checkout (python examples/.../x.py)
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..", "..", "..")))
 
 from mpi_operator_amd import parallel as hvd
 from mpi_operator_amd.models.bert import bert_large, bert_base, to_mi355x_bert

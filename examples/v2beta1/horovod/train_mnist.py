@@ -10,8 +10,13 @@ or elastically (reference proposals/elastic-horovod.md path):
            --min-np 2 --max-np 8 -- python3 train_mnist.py
 """
 import argparse
+import os
+import sys
 
 import torch
+
+# allow running straight from a source checkout (python examples/.../x.py)
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..", "..", "..")))
 
 from mpi_operator_amd import parallel as hvd
 from mpi_operator_amd.models import SimpleCNN

@@ -136,10 +136,15 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   int Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
   int RSC = (int)R * S * C;
   int tiles = ((Kout + 127) / 128) * ((RSC + 127) / 128);
-  // fill the chip: tiles×splits ≈ 2 WGs per CU even for 1-2-tile outputs
-  // (stage-1 dw is [64][256]: 2 tiles — a splits cap of 64 left 3/4 of the
-  // CUs idle); slab stays ≤16 MB because tiny outputs have tiny Kout*RSC
-  int splits = std::min(std::max(1024 / tiles, 1), 256);
+  long M = (long)N * HO * WO;
+  int nk = (int)((M + 63) / 64);
+  // fill the chip (tiles×splits ≈ 512 blocks = 2 WG/CU) but keep ≥8 k-tiles
+  // per split so each block amortizes its pipeline prologue; a flat 1024
+  // cap overshot mid shapes (b2 1x1 dw: 256 splits of nk=3 ran at 94 TF
+  // vs 167 at 64 splits)
+  int splits = std::max(512 / tiles, 1);
+  splits = std::min(splits, std::max(nk / 8, 1));
+  splits = std::min(splits, 256);
   Tensor partial = at::empty({(long)splits, (long)Kout, (long)RSC},
                              x.options().dtype(at::kFloat));
   // dw bf16 (fp32-accumulated in the split-K slabs, rounded once at the
