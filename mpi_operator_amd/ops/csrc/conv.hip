@@ -19,6 +19,9 @@
 // into init(); per k-step only the (r,s,c) split of one k remains — the
 // innermost gather is 2 adds + 2 bounds checks + 1 load.
 struct ConvFwdStage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   const uint16_t *x;
   int H, W, C, HO, WO, S, stride, pad, K;
   long M;
@@ -41,7 +44,7 @@ struct ConvFwdStage {
       wb_[i] = wo * stride - pad;
     }
   }
-  DEV_INLINE void load(int tid, int, int kb) {
+  DEV_INLINE void load(int tid, int, int kb, ushort8 *) {
     int k = kb + (tid & 7) * 8;
     int c = k % C, rs = k / C;
     int s_ = rs % S, rr = rs / S;
@@ -66,6 +69,9 @@ struct ConvFwdStage {
 // STRIDE is a template parameter so the innermost %/÷ are shifts, not the
 // full integer division a runtime stride emits.
 template <int STRIDE> struct ConvDgradStage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   const uint16_t *dy;
   int H, W, Q /*Kout*/, HO, WO, S, pad, K;
   long M;
@@ -88,7 +94,7 @@ template <int STRIDE> struct ConvDgradStage {
       wb_[i] = w_ + pad;
     }
   }
-  DEV_INLINE void load(int tid, int, int kb) {
+  DEV_INLINE void load(int tid, int, int kb, ushort8 *) {
     int k = kb + (tid & 7) * 8;
     int q = k % Q, rs = k / Q;
     int s_ = rs % S, rr = rs / S;
@@ -116,6 +122,9 @@ template <int STRIDE> struct ConvDgradStage {
 // init(); per k-step only the even output-pixel m is decomposed (odd m is
 // derived by carry), so the inner gather is adds + bounds + load.
 struct XcolStage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   const uint16_t *x;
   int H, W, C, HO, WO, S, stride, pad, RSC;
   long M;
@@ -141,7 +150,7 @@ struct XcolStage {
       return *(const ushort8 *)(x + ((long)(n * H + h) * W + w) * C + coff_[it]);
     return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
   }
-  DEV_INLINE void load(int tid, int, int kb) {
+  DEV_INLINE void load(int tid, int, int kb, ushort8 *) {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       long m = kb + k0_[it]; // even
@@ -228,6 +237,9 @@ struct DgradWTn {
 
 // NT stager over sub-image rows: k = (tap, q) with q fastest.
 struct DgradS2Stage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   const uint16_t *dy;
   int HO, WO, Q, W2, H2, K;
   long M;
@@ -254,7 +266,7 @@ struct DgradS2Stage {
       wb_[i] = w_;
     }
   }
-  DEV_INLINE void load(int tid, int, int kb) {
+  DEV_INLINE void load(int tid, int, int kb, ushort8 *) {
     int k = kb + (tid & 7) * 8;
     int q = k % Q, ti = k / Q;
     int ih = ti / ntw, iw = ti - ih * ntw;

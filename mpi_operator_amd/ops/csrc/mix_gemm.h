@@ -31,10 +31,13 @@ constexpr int MXP = 9; // slots per LDS image row (pitch 144 B)
 // TN: loader.load(k, col0) -> 8 bf16 along the output dim (k-strided).
 
 template <class L> struct NtStage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   L l;
   ushort8 r[4];
   DEV_INLINE void init(int, int) {}
-  DEV_INLINE void load(int tid, int base, int kb) {
+  DEV_INLINE void load(int tid, int base, int kb, ushort8 *) {
     int s_row = tid >> 3, s_slot = tid & 7;
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -48,10 +51,13 @@ template <class L> struct NtStage {
 };
 
 template <class L> struct TnStage {
+  static constexpr int PITCH = MXP;
+  static constexpr bool SWZ = false;
+  static constexpr bool GLDS = false;
   L l;
   ushort8 r[4]; // [it][k-parity]: 2 pair-loads of 2 adjacent k each
   DEV_INLINE void init(int, int) {}
-  DEV_INLINE void load(int tid, int base, int kb) {
+  DEV_INLINE void load(int tid, int base, int kb, ushort8 *) {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int p = tid + it * 256;  // pair index in [0,512)
@@ -75,6 +81,40 @@ template <class L> struct TnStage {
             (uint32_t)va[j] | ((uint32_t)vb[j] << 16);
     }
   }
+};
+
+// Plain k-contiguous operand staged by global_load_lds (direct HBM→LDS DMA,
+// no staging registers, no ds_write pass — the measured +69% rung of the
+// guide's ladder at this tile). glds writes wave-uniform-base + lane*16, so
+// the LDS image is lane-linear (pitch 8, no pad); bank conflicts are dodged
+// by pre-swizzling the SOURCE k-octet with (row&7) and XOR-ing the same on
+// the read side.
+struct GldsNt {
+  static constexpr int PITCH = 8;
+  static constexpr bool SWZ = true;
+  static constexpr bool GLDS = true;
+  const uint16_t *p;
+  int rows;
+  long ld;
+  int kdim;
+  DEV_INLINE void init(int, int) {}
+  DEV_INLINE void load(int tid, int base, int kb, ushort8 *img) {
+    int s_row = tid >> 3, s_slot = tid & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = base + s_row + 32 * i;
+      int k = kb + (s_slot ^ (row & 7)) * 8; // pre-swizzled source octet
+      if (row < rows && k < kdim) {
+        auto *dst = (__attribute__((address_space(3))) ushort8 *)(img + i * 256 + tid);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void *)(p + (long)row * ld + k),
+            (__attribute__((address_space(3))) void *)dst, 16, 0, 0);
+      } else {
+        img[i * 256 + tid] = ushort8{0, 0, 0, 0, 0, 0, 0, 0}; // edge zero-fill
+      }
+    }
+  }
+  DEV_INLINE void write(int, ushort8 *) const {} // loads land in LDS directly
 };
 
 typedef __attribute__((ext_vector_type(16))) float float16v;
@@ -182,7 +222,17 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   int lane = tid & 63, wave = tid >> 6;
   int wr = wave >> 1, wc = wave & 1;
 
-  __shared__ ushort8 lds[ONEBUF ? 1 : 2][2][BM * MXP]; // [buf][A|B][image]
+  // ONE shared object (a second __shared__ array makes hipcc drain vmcnt(0)
+  // before every ds_read while a glds is in flight — guide §5 trap 4a)
+  constexpr int ASZ = BM * SA::PITCH, BSZ = BM * SB::PITCH;
+  constexpr int NBUF = ONEBUF ? 1 : 2;
+  __shared__ ushort8 lds[NBUF * (ASZ + BSZ)];
+  ushort8 *imgA[2], *imgB[2];
+#pragma unroll
+  for (int b = 0; b < 2; ++b) {
+    imgA[b] = lds + (b % NBUF) * (ASZ + BSZ);
+    imgB[b] = imgA[b] + ASZ;
+  }
 
   // 32x32x16 MFMA (higher ceiling than 16x16x32): each wave computes a
   // 64x64 quadrant as 2x2 fragments of 32x32, 16 fp32 accumulators each.
@@ -200,18 +250,18 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   sa.init(tid, row0);
   sb.init(tid, col0);
   if (nk > 0) {
-    sa.load(tid, row0, t0 * BK);
-    sb.load(tid, col0, t0 * BK);
-    sa.write(tid, lds[0][0]);
-    sb.write(tid, lds[0][1]);
+    sa.load(tid, row0, t0 * BK, imgA[0]);
+    sb.load(tid, col0, t0 * BK, imgB[0]);
+    sa.write(tid, imgA[0]);
+    sb.write(tid, imgB[0]);
   }
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     int buf = ONEBUF ? 0 : (t & 1);
-    if (t + 1 < nk) { // issue-early: HBM latency hides under the MFMAs
-      sa.load(tid, row0, (t0 + t + 1) * BK);
-      sb.load(tid, col0, (t0 + t + 1) * BK);
+    if (t + 1 < nk) { // issue-early: loads (or glds DMA) land under MFMAs
+      sa.load(tid, row0, (t0 + t + 1) * BK, imgA[buf ^ 1]);
+      sb.load(tid, col0, (t0 + t + 1) * BK, imgB[buf ^ 1]);
     }
 #pragma unroll
     for (int kk = 0; kk < BK / 16; ++kk) { // 4 k-steps of 16
@@ -219,11 +269,17 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
       // lane l: row = l&31, k = kk*16 + (l>>5)*8 .. +7 → slot kk*2+(l>>5)
       int slot = kk * 2 + (lane >> 5);
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
-        af[mi] = us8_to_bf8v(lds[buf][0][(wr * 64 + mi * 32 + (lane & 31)) * MXP + slot]);
+      for (int mi = 0; mi < 2; ++mi) {
+        int arow = wr * 64 + mi * 32 + (lane & 31);
+        int aslot = SA::SWZ ? (slot ^ (arow & 7)) : slot;
+        af[mi] = us8_to_bf8v(imgA[buf][arow * SA::PITCH + aslot]);
+      }
 #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
-        bf_[ni] = us8_to_bf8v(lds[buf][1][(wc * 64 + ni * 32 + (lane & 31)) * MXP + slot]);
+      for (int ni = 0; ni < 2; ++ni) {
+        int brow = wc * 64 + ni * 32 + (lane & 31);
+        int bslot = SB::SWZ ? (slot ^ (brow & 7)) : slot;
+        bf_[ni] = us8_to_bf8v(imgB[buf][brow * SB::PITCH + bslot]);
+      }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -234,12 +290,12 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     if (ONEBUF) {
       if (t + 1 < nk) { // reuse the single buffer: drain-readers barrier,
         __syncthreads(); // then overwrite with tile t+1
-        sa.write(tid, lds[0][0]);
-        sb.write(tid, lds[0][1]);
+        sa.write(tid, imgA[0]);
+        sb.write(tid, imgB[0]);
       }
     } else if (t + 1 < nk) { // write-late into the other buffer
-      sa.write(tid, lds[buf ^ 1][0]);
-      sb.write(tid, lds[buf ^ 1][1]);
+      sa.write(tid, imgA[buf ^ 1]);
+      sb.write(tid, imgB[buf ^ 1]);
     }
     __syncthreads();
   }
@@ -283,10 +339,13 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   // and enough tiles to matter.
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
   dim3 grid(nwg, splits);
-  static const bool onebuf = [] {
+  static const bool onebuf_env = [] {
     const char *e = getenv("MPIAMD_GEMM_ONEBUF");
     return e && e[0] == '1';
   }();
+  // glds targets the next buffer while the current one is being read — a
+  // single buffer would race
+  const bool onebuf = onebuf_env && !SA::GLDS && !SB::GLDS;
   if (onebuf) {
     if (c_f32)
       mix_gemm_k<SA, SB, true, WR, true><<<grid, NT_THREADS, 0, s>>>(
@@ -313,13 +372,15 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
                             s, splits);
 }
 
-// Plain NT×NT entry (both operands k-contiguous row-major).
+// Plain NT×NT entry (both operands k-contiguous row-major): both staged
+// via global_load_lds.
 template <class LA, class LB>
 static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
                                  int N, int K, long ldc, bool c_f32,
                                  hipStream_t s, int splits = 1) {
-  return launch_mix_gemm(NtStage<LA>{la}, NtStage<LB>{lb}, c, M, N, K, ldc,
-                         c_f32, s, splits);
+  GldsNt ga{la.p, la.rows, la.ld, la.kdim};
+  GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
+  return launch_mix_gemm(ga, gb, c, M, N, K, ldc, c_f32, s, splits);
 }
 
 // ---- TN loaders ----
