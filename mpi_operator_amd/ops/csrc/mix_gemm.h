@@ -227,12 +227,11 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   constexpr int ASZ = BM * SA::PITCH, BSZ = BM * SB::PITCH;
   constexpr int NBUF = ONEBUF ? 1 : 2;
   __shared__ ushort8 lds[NBUF * (ASZ + BSZ)];
-  ushort8 *imgA[2], *imgB[2];
-#pragma unroll
-  for (int b = 0; b < 2; ++b) {
-    imgA[b] = lds + (b % NBUF) * (ASZ + BSZ);
-    imgB[b] = imgA[b] + ASZ;
-  }
+  // NOTE: compute image bases arithmetically at each use — a runtime-indexed
+  // private array of LDS pointers de-optimizes every ds_read (measured -15%
+  // across ALL kernels when these were ushort8* imgA[2] arrays)
+#define MXG_A(b) (lds + ((b) % NBUF) * (ASZ + BSZ))
+#define MXG_B(b) (lds + ((b) % NBUF) * (ASZ + BSZ) + ASZ)
 
   // 32x32x16 MFMA (higher ceiling than 16x16x32): each wave computes a
   // 64x64 quadrant as 2x2 fragments of 32x32, 16 fp32 accumulators each.
@@ -250,18 +249,18 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
   sa.init(tid, row0);
   sb.init(tid, col0);
   if (nk > 0) {
-    sa.load(tid, row0, t0 * BK, imgA[0]);
-    sb.load(tid, col0, t0 * BK, imgB[0]);
-    sa.write(tid, imgA[0]);
-    sb.write(tid, imgB[0]);
+    sa.load(tid, row0, t0 * BK, MXG_A(0));
+    sb.load(tid, col0, t0 * BK, MXG_B(0));
+    sa.write(tid, MXG_A(0));
+    sb.write(tid, MXG_B(0));
   }
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     int buf = ONEBUF ? 0 : (t & 1);
     if (t + 1 < nk) { // issue-early: loads (or glds DMA) land under MFMAs
-      sa.load(tid, row0, (t0 + t + 1) * BK, imgA[buf ^ 1]);
-      sb.load(tid, col0, (t0 + t + 1) * BK, imgB[buf ^ 1]);
+      sa.load(tid, row0, (t0 + t + 1) * BK, MXG_A(buf ^ 1));
+      sb.load(tid, col0, (t0 + t + 1) * BK, MXG_B(buf ^ 1));
     }
 #pragma unroll
     for (int kk = 0; kk < BK / 16; ++kk) { // 4 k-steps of 16
@@ -272,13 +271,13 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
       for (int mi = 0; mi < 2; ++mi) {
         int arow = wr * 64 + mi * 32 + (lane & 31);
         int aslot = SA::SWZ ? (slot ^ (arow & 7)) : slot;
-        af[mi] = us8_to_bf8v(imgA[buf][arow * SA::PITCH + aslot]);
+        af[mi] = us8_to_bf8v(MXG_A(buf)[arow * SA::PITCH + aslot]);
       }
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
         int brow = wc * 64 + ni * 32 + (lane & 31);
         int bslot = SB::SWZ ? (slot ^ (brow & 7)) : slot;
-        bf_[ni] = us8_to_bf8v(imgB[buf][brow * SB::PITCH + bslot]);
+        bf_[ni] = us8_to_bf8v(MXG_B(buf)[brow * SB::PITCH + bslot]);
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
@@ -290,12 +289,12 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     if (ONEBUF) {
       if (t + 1 < nk) { // reuse the single buffer: drain-readers barrier,
         __syncthreads(); // then overwrite with tile t+1
-        sa.write(tid, imgA[0]);
-        sb.write(tid, imgB[0]);
+        sa.write(tid, MXG_A(0));
+        sb.write(tid, MXG_B(0));
       }
     } else if (t + 1 < nk) { // write-late into the other buffer
-      sa.write(tid, imgA[buf ^ 1]);
-      sb.write(tid, imgB[buf ^ 1]);
+      sa.write(tid, MXG_A(buf ^ 1));
+      sb.write(tid, MXG_B(buf ^ 1));
     }
     __syncthreads();
   }
@@ -322,6 +321,9 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     }
   }
 }
+
+#undef MXG_A
+#undef MXG_B
 
 template <class SA, class SB, class WR>
 static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
