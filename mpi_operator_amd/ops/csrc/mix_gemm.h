@@ -374,15 +374,23 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
                             s, splits);
 }
 
-// Plain NT×NT entry (both operands k-contiguous row-major): both staged
-// via global_load_lds.
+// Plain NT×NT entry (both operands k-contiguous row-major). glds staging
+// vs register staging selectable for same-box A/B (MPIAMD_GLDS=0 reverts).
 template <class LA, class LB>
 static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
                                  int N, int K, long ldc, bool c_f32,
                                  hipStream_t s, int splits = 1) {
-  GldsNt ga{la.p, la.rows, la.ld, la.kdim};
-  GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
-  return launch_mix_gemm(ga, gb, c, M, N, K, ldc, c_f32, s, splits);
+  static const bool use_glds = [] {
+    const char *e = getenv("MPIAMD_GLDS");
+    return !(e && e[0] == '0');
+  }();
+  if (use_glds) {
+    GldsNt ga{la.p, la.rows, la.ld, la.kdim};
+    GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
+    return launch_mix_gemm(ga, gb, c, M, N, K, ldc, c_f32, s, splits);
+  }
+  return launch_mix_gemm(NtStage<LA>{la}, NtStage<LB>{lb}, c, M, N, K, ldc,
+                         c_f32, s, splits);
 }
 
 // ---- TN loaders ----
