@@ -111,39 +111,42 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
         a1[j] += lds[1][(rl * C8 + cb) * 8 + j];
       }
     }
+    // slab publish with WRITE-THROUGH (sc1) stores: an agent release fence
+    // here (buffer_wbl2) flushed every XCD's whole dirty L2 per block and
+    // made the WHOLE STEP 3.4x slower — sc1 stores skip L1/L2 dirty state,
+    // and the sc1 reader needs no acquire (guide G16 publish forms)
     float *p0 = partial + (long)blockIdx.x * 2 * C + cb * 8;
     float *p1 = p0 + C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      p0[j] = a0[j];
-      p1[j] = a1[j];
+      __hip_atomic_store(&p0[j], a0[j], __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(&p1[j], a1[j], __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
     }
   }
-  // ---- last-arriver finalize (in-launch combine, guide G16 recipe) ----
+  // ---- last-arriver finalize (in-launch combine) ----
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   int *flag = (int *)&lds[0][0]; // reuse the ONE shared object
   if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     int old = __hip_atomic_fetch_add(cnt, 1, __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_AGENT);
     *flag = (old == (int)gridDim.x - 1) ? 1 : 0;
   }
   __syncthreads();
   if (*flag == 0) return;
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    *cnt = 0; // self-reset: next launch / graph replay starts clean
-  }
+  if (threadIdx.x == 0) *cnt = 0; // next launch / graph replay starts clean
   __syncthreads();
   int grid = gridDim.x;
   float inv_m = 1.f / (float)M;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
     float s0 = 0.f, s1 = 0.f;
     for (int g = 0; g < grid; ++g) {
-      s0 += partial[(long)g * 2 * C + c];
-      s1 += partial[(long)g * 2 * C + C + c];
+      s0 += __hip_atomic_load(&partial[(long)g * 2 * C + c], __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT);
+      s1 += __hip_atomic_load(&partial[(long)g * 2 * C + C + c],
+                              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
     if (WHAT == 0) {
       float mu = s0 * inv_m;
