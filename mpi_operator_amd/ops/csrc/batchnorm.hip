@@ -136,17 +136,21 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
   }
   __syncthreads();
   if (*flag == 0) return;
-  if (threadIdx.x == 0) *cnt = 0; // next launch / graph replay starts clean
+  if (threadIdx.x == 0) {
+    // ONE agent acquire for the whole read phase (drops this CU's L1; the
+    // sc1-published slab is L2-resident) — per-element atomic loads cannot
+    // be pipelined by the compiler and serialized ~1000 L2 round trips
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    *cnt = 0; // next launch / graph replay starts clean
+  }
   __syncthreads();
   int grid = gridDim.x;
   float inv_m = 1.f / (float)M;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
     float s0 = 0.f, s1 = 0.f;
     for (int g = 0; g < grid; ++g) {
-      s0 += __hip_atomic_load(&partial[(long)g * 2 * C + c], __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT);
-      s1 += __hip_atomic_load(&partial[(long)g * 2 * C + C + c],
-                              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      s0 += partial[(long)g * 2 * C + c];
+      s1 += partial[(long)g * 2 * C + C + c];
     }
     if (WHAT == 0) {
       float mu = s0 * inv_m;
