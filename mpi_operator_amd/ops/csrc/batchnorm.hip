@@ -11,33 +11,14 @@
 // cb = t % C8 and row-lane t / C8 — consecutive lanes read consecutive
 // 16 B groups (fully coalesced); accumulation stays in registers.
 // Requires C8 = C/8 <= 256 (C <= 2048 — every ResNet/BERT channel width).
-// The finalize runs inside the LAST-ARRIVING block of this kernel (plain
-// slab stores -> agent release fence -> relaxed ticket; the last block takes
-// an agent acquire and reduces the L2-hot slab) — one launch instead of
-// two per BN direction. Counter is a persistent device int, reset by the
-// last arriver so the next launch (or graph replay) starts from 0.
 template <int WHAT> // 0: fwd stats (sum, sumsq); 1: bwd stats (dy*m, dy*m*xhat)
 __global__ void bn_partials_k(const ushort8 *__restrict__ x,
                               const ushort8 *__restrict__ dy,
                               const ushort8 *__restrict__ y,
-                              const float *__restrict__ mean_in,
-                              const float *__restrict__ invstd_in,
+                              const float *__restrict__ mean,
+                              const float *__restrict__ invstd,
                               float *__restrict__ partial, // [grid][2][C]
-                              long M, int C8, int relu, int *__restrict__ cnt,
-                              const float *__restrict__ gamma,
-                              const float *__restrict__ beta, float eps,
-                              float *__restrict__ mean_out,
-                              float *__restrict__ invstd_out,
-                              float *__restrict__ scale,
-                              float *__restrict__ shift,
-                              float *__restrict__ running_mean,
-                              float *__restrict__ running_var, float momentum,
-                              float *__restrict__ dbeta,
-                              float *__restrict__ dgamma,
-                              float *__restrict__ k1, float *__restrict__ k2,
-                              float *__restrict__ k3) {
-  const float *mean = mean_in;
-  const float *invstd = invstd_in;
+                              long M, int C8, int relu) {
   int C = C8 * 8;
   int cb = threadIdx.x % C8;
   int row_lane = threadIdx.x / C8;
@@ -111,71 +92,87 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
         a1[j] += lds[1][(rl * C8 + cb) * 8 + j];
       }
     }
-    // slab publish with WRITE-THROUGH (sc1) stores: an agent release fence
-    // here (buffer_wbl2) flushed every XCD's whole dirty L2 per block and
-    // made the WHOLE STEP 3.4x slower — sc1 stores skip L1/L2 dirty state,
-    // and the sc1 reader needs no acquire (guide G16 publish forms)
     float *p0 = partial + (long)blockIdx.x * 2 * C + cb * 8;
     float *p1 = p0 + C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      __hip_atomic_store(&p0[j], a0[j], __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      __hip_atomic_store(&p1[j], a1[j], __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
+      p0[j] = a0[j];
+      p1[j] = a1[j];
     }
   }
-  // ---- last-arriver finalize (in-launch combine) ----
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  int *flag = (int *)&lds[0][0]; // reuse the ONE shared object
-  if (threadIdx.x == 0) {
-    int old = __hip_atomic_fetch_add(cnt, 1, __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-    *flag = (old == (int)gridDim.x - 1) ? 1 : 0;
+}
+
+// finalize: 8 lanes cooperate per channel (grid entries split across lanes,
+// shfl-reduced) — the serial-per-channel version was one-wave latency-bound
+// at 124 µs/call and 31% of the whole step.
+DEV_INLINE void lane8_sums(const float *__restrict__ partial, int grid, int C,
+                           int c, int lane8, float &s0, float &s1) {
+  s0 = 0.f;
+  s1 = 0.f;
+  for (int g = lane8; g < grid; g += 32) {
+    s0 += partial[(long)g * 2 * C + c];
+    s1 += partial[(long)g * 2 * C + C + c];
   }
-  __syncthreads();
-  if (*flag == 0) return;
-  if (threadIdx.x == 0) {
-    // ONE agent acquire for the whole read phase (drops this CU's L1; the
-    // sc1-published slab is L2-resident) — per-element atomic loads cannot
-    // be pipelined by the compiler and serialized ~1000 L2 round trips
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    *cnt = 0; // next launch / graph replay starts clean
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) {
+    s0 += __shfl_down(s0, off, 32);
+    s1 += __shfl_down(s1, off, 32);
   }
-  __syncthreads();
-  int grid = gridDim.x;
-  float inv_m = 1.f / (float)M;
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float s0 = 0.f, s1 = 0.f;
-    for (int g = 0; g < grid; ++g) {
-      s0 += partial[(long)g * 2 * C + c];
-      s1 += partial[(long)g * 2 * C + C + c];
-    }
-    if (WHAT == 0) {
-      float mu = s0 * inv_m;
-      float var = fmaxf(s1 * inv_m - mu * mu, 0.f);
-      float is = rsqrtf(var + eps);
-      mean_out[c] = mu;
-      invstd_out[c] = is;
-      float sc = gamma[c] * is;
-      scale[c] = sc;
-      shift[c] = beta[c] - mu * sc;
-      if (running_mean) {
-        float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
-        running_mean[c] = running_mean[c] * (1.f - momentum) + mu * momentum;
-        running_var[c] =
-            running_var[c] * (1.f - momentum) + var * unbias * momentum;
-      }
-    } else {
-      dbeta[c] = s0;
-      dgamma[c] = s1;
-      float g_is = gamma[c] * invstd[c];
-      k1[c] = g_is;
-      k2[c] = g_is * s0 * inv_m;
-      k3[c] = g_is * s1 * inv_m;
-    }
+}
+
+// fwd: mean/invstd + scale/shift + fused running-stats update (saves ~5 eager
+// tensor ops per BN layer per step on the torch side).
+__global__ void bn_finalize_fwd_k(const float *__restrict__ partial, int grid,
+                                  int C, const float *__restrict__ gamma,
+                                  const float *__restrict__ beta, float inv_m,
+                                  float eps, float *__restrict__ mean,
+                                  float *__restrict__ invstd,
+                                  float *__restrict__ scale,
+                                  float *__restrict__ shift,
+                                  float *__restrict__ running_mean,
+                                  float *__restrict__ running_var,
+                                  float momentum, float unbias) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = t / 32, lane8 = t % 32;
+  if (c >= C) return;
+  float s, sq;
+  lane8_sums(partial, grid, C, c, lane8, s, sq);
+  if (lane8 != 0) return;
+  float mu = s * inv_m;
+  float var = fmaxf(sq * inv_m - mu * mu, 0.f);
+  float is = rsqrtf(var + eps);
+  mean[c] = mu;
+  invstd[c] = is;
+  float sc = gamma[c] * is;
+  scale[c] = sc;
+  shift[c] = beta[c] - mu * sc;
+  if (running_mean) {
+    running_mean[c] = running_mean[c] * (1.f - momentum) + mu * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + var * unbias * momentum;
   }
+}
+
+// bwd: dbeta/dgamma + the three per-channel dx coefficients
+__global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
+                                  int C, const float *__restrict__ gamma,
+                                  const float *__restrict__ invstd, float inv_m,
+                                  float *__restrict__ dbeta,
+                                  float *__restrict__ dgamma,
+                                  float *__restrict__ k1, // gamma*invstd
+                                  float *__restrict__ k2, // k1*dbeta/m
+                                  float *__restrict__ k3) { // k1*dgamma/m (×xhat in apply)
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = t / 32, lane8 = t % 32;
+  if (c >= C) return;
+  float s0, s1;
+  lane8_sums(partial, grid, C, c, lane8, s0, s1);
+  if (lane8 != 0) return;
+  dbeta[c] = s0;
+  dgamma[c] = s1;
+  float g_is = gamma[c] * invstd[c];
+  k1[c] = g_is;
+  k2[c] = g_is * s0 * inv_m;
+  k3[c] = g_is * s1 * inv_m;
 }
 
 // apply scale/shift (+optional residual add) (+ReLU): fwd-train, fwd-eval
@@ -301,16 +298,19 @@ static int bn_apply_grid(long M, int C8) {
 extern "C" hipError_t bn_fwd_train_launch(
     const void *x, const void *res, const float *gamma, const float *beta,
     float eps, int relu, void *y, float *mean, float *invstd, float *scale,
-    float *shift, float *partial, int *cnt, float *running_mean,
-    float *running_var, float momentum, long M, int C, hipStream_t s) {
+    float *shift, float *partial, float *running_mean, float *running_var,
+    float momentum, long M, int C, hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
-  bn_partials_k<0><<<grid, 256, 0, s>>>(
-      (const ushort8 *)x, nullptr, nullptr, nullptr, nullptr, partial, M, C8,
-      0, cnt, gamma, beta, eps, mean, invstd, scale, shift, running_mean,
-      running_var, momentum, nullptr, nullptr, nullptr, nullptr, nullptr);
+  bn_partials_k<0><<<grid, 256, 0, s>>>((const ushort8 *)x, nullptr, nullptr,
+                                        nullptr, nullptr, partial, M, C8, 0);
+  HIP_KERNEL_CHECK();
+  float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  bn_finalize_fwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+      partial, grid, C, gamma, beta, 1.f / (float)M, eps, mean, invstd, scale,
+      shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
       (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y, M,
@@ -337,16 +337,18 @@ extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
                                     const float *mean, const float *invstd,
                                     int relu, void *dx, float *dgamma,
                                     float *dbeta, float *k1, float *k2,
-                                    float *k3, float *partial, int *cnt,
-                                    long M, int C, hipStream_t s) {
+                                    float *k3, float *partial, long M, int C,
+                                    hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
-  bn_partials_k<1><<<grid, 256, 0, s>>>(
-      (const ushort8 *)x, (const ushort8 *)dy, (const ushort8 *)y, mean,
-      invstd, partial, M, C8, relu, cnt, gamma, nullptr, 0.f, nullptr, nullptr,
-      nullptr, nullptr, nullptr, nullptr, 0.f, dbeta, dgamma, k1, k2, k3);
+  bn_partials_k<1><<<grid, 256, 0, s>>>((const ushort8 *)x, (const ushort8 *)dy,
+                                        (const ushort8 *)y, mean, invstd,
+                                        partial, M, C8, relu);
+  HIP_KERNEL_CHECK();
+  bn_finalize_bwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+      partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
   HIP_KERNEL_CHECK();
   bn_bwd_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
       (const ushort8 *)dy, (const ushort8 *)x, (const ushort8 *)y, mean,

@@ -7,8 +7,6 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
-#include <map>
-#include <mutex>
 #include <vector>
 
 namespace {
@@ -61,15 +59,14 @@ hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
                            hipStream_t);
 hipError_t bn_fwd_train_launch(const void *, const void *, const float *,
                                const float *, float, int, void *, float *,
-                               float *, float *, float *, float *, int *,
-                               float *, float *, float, long, int,
-                               hipStream_t);
+                               float *, float *, float *, float *, float *,
+                               float *, float, long, int, hipStream_t);
 hipError_t bn_fwd_eval_launch(const void *, const float *, const float *, int,
                               void *, long, int, hipStream_t);
 hipError_t bn_bwd_launch(const void *, const void *, const void *,
                          const float *, const float *, const float *, int,
                          void *, float *, float *, float *, float *, float *,
-                         float *, int *, long, int, hipStream_t);
+                         float *, long, int, hipStream_t);
 hipError_t maxpool_fwd_launch(const void *, void *, uint8_t *, int, int, int,
                               int, int, int, int, int, int, hipStream_t);
 hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
@@ -184,20 +181,6 @@ static void conv2d_dgrad_acc(const Tensor &dy, const Tensor &w, Tensor &dx_acc) 
 }
 
 // ------------------------- batchnorm -------------------------
-// Persistent per-device ticket counter for the BN in-launch finalize
-// (allocated zeroed once — the kernel's last arriver resets it, so it is
-// reusable across launches and hipGraph replays; same-stream ordering makes
-// one counter per device sufficient).
-static int *bn_counter(const Tensor &like) {
-  static std::mutex mu;
-  static std::map<int, Tensor> cache;
-  std::lock_guard<std::mutex> g(mu);
-  int dev = like.device().index();
-  auto it = cache.find(dev);
-  if (it == cache.end())
-    it = cache.emplace(dev, at::zeros({1}, like.options().dtype(at::kInt))).first;
-  return it->second.data_ptr<int>();
-}
 // res (optional, may be undefined): residual tensor folded into the apply
 // pass — y = [relu](bn(x) + res), the bottleneck-join fusion.
 static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
@@ -227,8 +210,7 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
                           y.data_ptr(), mean.data_ptr<float>(),
                           invstd.data_ptr<float>(), scale.data_ptr<float>(),
                           shift.data_ptr<float>(), partial.data_ptr<float>(),
-                          bn_counter(x), rm, rv, (float)momentum, M, C,
-                          cur_stream()));
+                          rm, rv, (float)momentum, M, C, cur_stream()));
   return {y, mean, invstd};
 }
 
@@ -263,8 +245,8 @@ static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
                     invstd.data_ptr<float>(), relu ? 1 : 0, dx.data_ptr(),
                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                     k1.data_ptr<float>(), k2.data_ptr<float>(),
-                    k3.data_ptr<float>(), partial.data_ptr<float>(),
-                    bn_counter(x), M, C, cur_stream()));
+                    k3.data_ptr<float>(), partial.data_ptr<float>(), M, C,
+                    cur_stream()));
   return {dx, dgamma, dbeta};
 }
 
