@@ -81,7 +81,6 @@ hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
-hipError_t transpose2d_bf16(const void *, void *, int, int, long, hipStream_t);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, hipStream_t);
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,

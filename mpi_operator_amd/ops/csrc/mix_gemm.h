@@ -412,17 +412,6 @@ struct TnRowMajor {
   }
 };
 
-// Implicit-im2col column gather for wgrad: k = output pixel m = (n,ho,wo),
-// col = (r,s,c) with c fastest; returns x[n][h][w][c0..c0+7] or zeros where
-// the patch leaves the image (requires C%8==0 so an 8-col run stays inside
-// one (r,s) cell — guaranteed by the NHWC8 input padding).
-struct TnXcol {
-  const uint16_t *x;
-  int H, W, C, HO, WO, S, stride, pad;
-  long M;
-  int RSC;
-  DEV_INLINE ushort8 load(int m, int rsc0) const {
-    if (m >= M || rsc0 >= RSC) return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
     int c = rsc0 % C, rs = rsc0 / C;
     int s_ = rs % S, r = rs / S;
     int wo = m % WO;
