@@ -411,16 +411,3 @@ struct TnRowMajor {
     return v;
   }
 };
-
-    int c = rsc0 % C, rs = rsc0 / C;
-    int s_ = rs % S, r = rs / S;
-    int wo = m % WO;
-    long t = m / WO;
-    int ho = t % HO;
-    int n = t / HO;
-    int h = ho * stride + r - pad, w = wo * stride + s_ - pad;
-    if ((unsigned)h >= (unsigned)H || (unsigned)w >= (unsigned)W)
-      return ushort8{0, 0, 0, 0, 0, 0, 0, 0};
-    return *(const ushort8 *)(x + ((long)(n * H + h) * W + w) * C + c);
-  }
-};
