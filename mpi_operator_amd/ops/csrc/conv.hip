@@ -20,8 +20,8 @@
 // innermost gather is 2 adds + 2 bounds checks + 1 load.
 struct ConvFwdStage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int) { return slot; }
   const uint16_t *x;
   int H, W, C, HO, WO, S, stride, pad, K;
   long M;
@@ -70,8 +70,8 @@ struct ConvFwdStage {
 // full integer division a runtime stride emits.
 template <int STRIDE> struct ConvDgradStage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int) { return slot; }
   const uint16_t *dy;
   int H, W, Q /*Kout*/, HO, WO, S, pad, K;
   long M;
@@ -123,8 +123,10 @@ template <int STRIDE> struct ConvDgradStage {
 // derived by carry), so the inner gather is adds + bounds + load.
 struct XcolStage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int row) {
+    return slot ^ ((row >> 3) & 7); // matches the TN write's slot-XOR
+  }
   const uint16_t *x;
   int H, W, C, HO, WO, S, stride, pad, RSC;
   long M;
@@ -135,8 +137,8 @@ struct XcolStage {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int p = tid + it * 256;
-      k0_[it] = (p & 31) * 2;
-      int rsc0 = base + (p >> 5) * 8;
+      k0_[it] = (p >> 4) * 2;
+      int rsc0 = base + (p & 15) * 8; // 16 lanes: 16 contiguous col octets
       cok_[it] = rsc0 < RSC;
       int c = rsc0 % C, rs = rsc0 / C;
       ss_[it] = rs % S;
@@ -176,13 +178,16 @@ struct XcolStage {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int p = tid + it * 256;
-      int k0 = (p & 31) * 2;
-      int col0 = (p >> 5) * 8;
+      int k0 = (p >> 4) * 2;
+      int col0 = (p & 15) * 8;
+      int slot = k0 >> 3, within = (k0 & 7) >> 1;
       const ushort8 &va = r[it * 2], &vb = r[it * 2 + 1];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        img32[(col0 + j) * (MXP * 4) + (k0 >> 1)] =
+      for (int j = 0; j < 8; ++j) {
+        int row = col0 + j;
+        img32[row * (MXP * 4) + rslot(slot, row) * 4 + within] =
             (uint32_t)va[j] | ((uint32_t)vb[j] << 16);
+      }
     }
   }
 };
@@ -238,8 +243,8 @@ struct DgradWTn {
 // NT stager over sub-image rows: k = (tap, q) with q fastest.
 struct DgradS2Stage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int) { return slot; }
   const uint16_t *dy;
   int HO, WO, Q, W2, H2, K;
   long M;

@@ -32,8 +32,8 @@ constexpr int MXP = 9; // slots per LDS image row (pitch 144 B)
 
 template <class L> struct NtStage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int) { return slot; }
   L l;
   ushort8 r[4];
   DEV_INLINE void init(int, int) {}
@@ -50,10 +50,18 @@ template <class L> struct NtStage {
   }
 };
 
+// TN pair assignment: 16 consecutive lanes cover 16 consecutive column
+// octets of ONE k-pair, so each wave instruction reads 4 CONTIGUOUS 256 B
+// spans instead of 64 scattered 16 B lines (ablation: staging loads were
+// 62% of the TN kernel). The transposed LDS write then lands 16 lanes on
+// one slot column — the (row>>3)&7 slot-XOR spreads them across banks
+// (matching XOR on the read side via rslot()).
 template <class L> struct TnStage {
   static constexpr int PITCH = MXP;
-  static constexpr bool SWZ = false;
   static constexpr bool GLDS = false;
+  static DEV_INLINE int rslot(int slot, int row) {
+    return slot ^ ((row >> 3) & 7);
+  }
   L l;
   ushort8 r[4]; // [it][k-parity]: 2 pair-loads of 2 adjacent k each
   DEV_INLINE void init(int, int) {}
@@ -61,8 +69,8 @@ template <class L> struct TnStage {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int p = tid + it * 256;  // pair index in [0,512)
-      int k0 = (p & 31) * 2;   // even k within the 64-deep tile
-      int col0 = (p >> 5) * 8; // 8 output-dim columns
+      int k0 = (p >> 4) * 2;   // even k within the 64-deep tile
+      int col0 = (p & 15) * 8; // 8 output-dim columns, contiguous per wave
       r[it * 2] = l.load(kb + k0, base + col0);
       r[it * 2 + 1] = l.load(kb + k0 + 1, base + col0);
     }
@@ -72,13 +80,16 @@ template <class L> struct TnStage {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int p = tid + it * 256;
-      int k0 = (p & 31) * 2;
-      int col0 = (p >> 5) * 8;
+      int k0 = (p >> 4) * 2;
+      int col0 = (p & 15) * 8;
+      int slot = k0 >> 3, within = (k0 & 7) >> 1;
       const ushort8 &va = r[it * 2], &vb = r[it * 2 + 1];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        img32[(col0 + j) * (MXP * 4) + (k0 >> 1)] =
+      for (int j = 0; j < 8; ++j) {
+        int row = col0 + j;
+        img32[row * (MXP * 4) + rslot(slot, row) * 4 + within] =
             (uint32_t)va[j] | ((uint32_t)vb[j] << 16);
+      }
     }
   }
 };
@@ -91,8 +102,8 @@ template <class L> struct TnStage {
 // the read side.
 struct GldsNt {
   static constexpr int PITCH = 8;
-  static constexpr bool SWZ = true;
   static constexpr bool GLDS = true;
+  static DEV_INLINE int rslot(int slot, int row) { return slot ^ (row & 7); }
   const uint16_t *p;
   int rows;
   long ld;
@@ -270,14 +281,12 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
         int arow = wr * 64 + mi * 32 + (lane & 31);
-        int aslot = SA::SWZ ? (slot ^ (arow & 7)) : slot;
-        af[mi] = us8_to_bf8v(MXG_A(buf)[arow * SA::PITCH + aslot]);
+        af[mi] = us8_to_bf8v(MXG_A(buf)[arow * SA::PITCH + SA::rslot(slot, arow)]);
       }
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
         int brow = wc * 64 + ni * 32 + (lane & 31);
-        int bslot = SB::SWZ ? (slot ^ (brow & 7)) : slot;
-        bf_[ni] = us8_to_bf8v(MXG_B(buf)[brow * SB::PITCH + bslot]);
+        bf_[ni] = us8_to_bf8v(MXG_B(buf)[brow * SB::PITCH + SB::rslot(slot, brow)]);
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
