@@ -337,7 +337,8 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   Tensor dw = at::empty({N, K}, f32);
   CHK(gemm_tn_tn(dyc.data_ptr(), xc.data_ptr(), dw.data_ptr(), N, K, M, N, K,
                  K, 1, cur_stream()));
-  Tensor db = at::empty({N}, f32);
+  // zeros, not empty: the fast colsum path accumulates atomically
+  Tensor db = at::zeros({N}, f32);
   CHK(colsum_bf16(dyc.data_ptr(), db.data_ptr<float>(), M, N, cur_stream()));
   return {dx, dw, db};
 }

@@ -120,6 +120,176 @@ static float run(const uint16_t *a, const uint16_t *b, float *c, int M, int N,
   return ms / iters;
 }
 
+// ---- experiment: 128(M) x 256(N) tile, 512 threads (8 waves as 2x4) ----
+// Halves B-panel re-reads for wide outputs (ablation: TN staging loads are
+// ~60% of the 128x128 kernel and operand-BW bound).
+__global__ __launch_bounds__(512) void wide_tn_k(TnRowMajor la, TnRowMajor lb,
+                                                 float *cptr, int M, int N,
+                                                 int K, int tiles_n, int kt,
+                                                 long sstride) {
+  constexpr int ASZ = 128 * MXP, BSZ = 256 * MXP;
+  __shared__ ushort8 lds[2 * (ASZ + BSZ)];
+  int tile = blockIdx.x, split = blockIdx.y;
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * 128, col0 = tn * 256;
+  int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 2, wc = wave & 3;
+  float16v acc[2][2] = {};
+  int nk_total = (K + BK - 1) / BK;
+  int t0 = split * kt;
+  int nk = min(kt, nk_total - t0);
+  if (nk < 0) nk = 0;
+  ushort8 ra[2], rb[4];
+  auto loadA = [&](int kb) {
+    int k0 = (tid >> 4) * 2, c0 = (tid & 15) * 8; // 512 pairs, 1/thread
+    ra[0] = la.load(kb + k0, row0 + c0);
+    ra[1] = la.load(kb + k0 + 1, row0 + c0);
+  };
+  auto loadB = [&](int kb) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 512;
+      int k0 = (p >> 5) * 2, c0 = (p & 31) * 8;
+      rb[it * 2] = lb.load(kb + k0, col0 + c0);
+      rb[it * 2 + 1] = lb.load(kb + k0 + 1, col0 + c0);
+    }
+  };
+  auto writeA = [&](ushort8 *img) {
+    uint32_t *im = (uint32_t *)img;
+    int k0 = (tid >> 4) * 2, c0 = (tid & 15) * 8;
+    int slot = k0 >> 3, within = (k0 & 7) >> 1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = c0 + j;
+      im[row * (MXP * 4) + (slot ^ ((row >> 3) & 7)) * 4 + within] =
+          (uint32_t)ra[0][j] | ((uint32_t)ra[1][j] << 16);
+    }
+  };
+  auto writeB = [&](ushort8 *img) {
+    uint32_t *im = (uint32_t *)img;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 512;
+      int k0 = (p >> 5) * 2, c0 = (p & 31) * 8;
+      int slot = k0 >> 3, within = (k0 & 7) >> 1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int row = c0 + j;
+        im[row * (MXP * 4) + (slot ^ ((row >> 3) & 7)) * 4 + within] =
+            (uint32_t)rb[it * 2][j] | ((uint32_t)rb[it * 2 + 1][j] << 16);
+      }
+    }
+  };
+#define WIMG_A(b) (lds + (b) * (ASZ + BSZ))
+#define WIMG_B(b) (lds + (b) * (ASZ + BSZ) + ASZ)
+  if (nk > 0) {
+    loadA(t0 * BK);
+    loadB(t0 * BK);
+    writeA(WIMG_A(0));
+    writeB(WIMG_B(0));
+  }
+  __syncthreads();
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+    if (t + 1 < nk) {
+      loadA((t0 + t + 1) * BK);
+      loadB((t0 + t + 1) * BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 16; ++kk) {
+      bf16x8 af[2], bf_[2];
+      int slot = kk * 2 + (lane >> 5);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int arow = wr * 64 + mi * 32 + (lane & 31);
+        af[mi] = us8_to_bf8v(WIMG_A(buf)[arow * MXP + (slot ^ ((arow >> 3) & 7))]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int brow = wc * 64 + ni * 32 + (lane & 31);
+        bf_[ni] = us8_to_bf8v(WIMG_B(buf)[brow * MXP + (slot ^ ((brow >> 3) & 7))]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < nk) {
+      writeA(WIMG_A(buf ^ 1));
+      writeB(WIMG_B(buf ^ 1));
+    }
+    __syncthreads();
+  }
+  cptr += split * sstride;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        if (row >= M) continue;
+        cptr[(long)row * N + col] = acc[mi][ni][r];
+      }
+    }
+}
+
+static float run_wide(const uint16_t *a, const uint16_t *b, float *c, int M,
+                      int N, long Kpix, int splits, int iters) {
+  TnRowMajor la{a, (long)M, (int)Kpix, M};
+  TnRowMajor lb{b, (long)N, (int)Kpix, N};
+  int tiles_m = (M + 127) / 128, tiles_n = (N + 255) / 256;
+  int nk = (int)((Kpix + 63) / 64);
+  if (splits > nk) splits = nk;
+  int kt = (nk + splits - 1) / splits;
+  dim3 grid(tiles_m * tiles_n, splits);
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  wide_tn_k<<<grid, 512>>>(la, lb, c, M, N, (int)Kpix, tiles_n, kt, (long)M * N);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    wide_tn_k<<<grid, 512>>>(la, lb, c, M, N, (int)Kpix, tiles_n, kt, (long)M * N);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  return ms / iters;
+}
+
+// correctness check helper: fp64 host reference on a small shape
+static void check_wide() {
+  int M = 128, N = 512;
+  long Kp = 256;
+  uint16_t *a, *b;
+  float *c;
+  hipMallocManaged(&a, Kp * M * 2);
+  hipMallocManaged(&b, Kp * N * 2);
+  hipMallocManaged(&c, (long)M * N * 4);
+  auto f2b = [](float f) { unsigned u; __builtin_memcpy(&u, &f, 4); return (uint16_t)(u >> 16); };
+  auto b2f = [](uint16_t h) { unsigned u = (unsigned)h << 16; float f; __builtin_memcpy(&f, &u, 4); return f; };
+  srand(7);
+  for (long i = 0; i < Kp * M; ++i) a[i] = f2b((rand() % 2000 - 1000) / 997.f);
+  for (long i = 0; i < Kp * N; ++i) b[i] = f2b((rand() % 2000 - 1000) / 997.f);
+  run_wide(a, b, c, M, N, Kp, 1, 1);
+  hipDeviceSynchronize();
+  double maxerr = 0;
+  for (int i = 0; i < M; i += 7)
+    for (int j = 0; j < N; j += 13) {
+      double ref = 0;
+      for (long k = 0; k < Kp; ++k) ref += (double)b2f(a[k * M + i]) * b2f(b[k * N + j]);
+      double err = fabs(c[(long)i * N + j] - ref) / (fabs(ref) + 1.0);
+      if (err > maxerr) maxerr = err;
+    }
+  printf("wide check maxrelerr %.4g %s\n", maxerr, maxerr < 0.02 ? "OK" : "FAIL");
+  hipFree(a); hipFree(b); hipFree(c);
+}
+
 int main() {
   // b2-1x1 wgrad shape: dw[128][512] over M=50176 pixels
   int M = 128, N = 512;
@@ -144,5 +314,25 @@ int main() {
   printf("no-mfma   %.1f us  (mfma+lds-read cost %.1f us)\n", nomfma * 1e3, (full - nomfma) * 1e3);
   printf("no-epilog %.1f us  (epilogue cost %.1f us)\n", noepi * 1e3, (full - noepi) * 1e3);
   printf("lds-only  %.1f us  (pure mfma+ds_read floor)\n", nolw * 1e3);
+  check_wide();
+  float wide = run_wide(a, b, c, M, N, Kpix, splits, iters);
+  printf("wide128x256 %.1f us  %.1f TF (vs full %.1f)\n", wide * 1e3,
+         gf / (wide / 1e3), gf / (full / 1e3));
+  // a conv3x3-wgrad-ish wider shape: dw[256][1152] over 50176 pixels
+  {
+    int M2 = 256, N2 = 1152;
+    uint16_t *a2, *b2; float *c2;
+    hipMalloc(&a2, Kpix * M2 * 2);
+    hipMalloc(&b2, Kpix * N2 * 2);
+    hipMalloc(&c2, (long)64 * M2 * N2 * 4);
+    hipMemset(a2, 0x3c, Kpix * M2 * 2);
+    hipMemset(b2, 0x3c, Kpix * N2 * 2);
+    double gf2 = 2.0 * M2 * N2 * Kpix / 1e12;
+    float f2 = run<0>(a2, b2, c2, M2, N2, Kpix, 48, iters);
+    float w2 = run_wide(a2, b2, c2, M2, N2, Kpix, 48, iters);
+    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f us %.1f TF\n",
+           f2 * 1e3, gf2 / (f2 / 1e3), w2 * 1e3, gf2 / (w2 / 1e3));
+  }
   return 0;
 }
+
