@@ -290,6 +290,122 @@ static void check_wide() {
   hipFree(a); hipFree(b); hipFree(c);
 }
 
+
+// distance-2 register pipeline variant (two in-flight register sets):
+// re-test of the earlier cross-box "regression" under same-box A/B.
+__global__ __launch_bounds__(NT_THREADS) void dist2_k(
+    TnStage<TnRowMajor> sa, TnStage<TnRowMajor> sb, float *cptr, int M, int N,
+    int K, int tiles_n, int kt, long sstride) {
+  int tile = blockIdx.x, split = blockIdx.y;
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * BM, col0 = tn * BN;
+  int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 1, wc = wave & 1;
+  constexpr int ASZ = BM * MXP;
+  __shared__ ushort8 lds[2 * 2 * ASZ];
+  float16v acc[2][2] = {};
+  int nk_total = (K + BK - 1) / BK;
+  int t0 = split * kt;
+  int nk = min(kt, nk_total - t0);
+  if (nk < 0) nk = 0;
+  TnStage<TnRowMajor> sa2 = sa, sb2 = sb;
+  if (nk > 0) {
+    sa.load(tid, row0, t0 * BK, nullptr);
+    sb.load(tid, col0, t0 * BK, nullptr);
+    sa.write(tid, lds);
+    sb.write(tid, lds + ASZ);
+  }
+  if (nk > 1) {
+    sa2.load(tid, row0, (t0 + 1) * BK, nullptr);
+    sb2.load(tid, col0, (t0 + 1) * BK, nullptr);
+  }
+  __syncthreads();
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+    if (t + 2 < nk) {
+      if (buf == 0) {
+        sa.load(tid, row0, (t0 + t + 2) * BK, nullptr);
+        sb.load(tid, col0, (t0 + t + 2) * BK, nullptr);
+      } else {
+        sa2.load(tid, row0, (t0 + t + 2) * BK, nullptr);
+        sb2.load(tid, col0, (t0 + t + 2) * BK, nullptr);
+      }
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 16; ++kk) {
+      bf16x8 af[2], bf_[2];
+      int slot = kk * 2 + (lane >> 5);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int arow = wr * 64 + mi * 32 + (lane & 31);
+        af[mi] = us8_to_bf8v(lds[buf * 2 * ASZ + arow * MXP + TnStage<TnRowMajor>::rslot(slot, arow)]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int brow = wc * 64 + ni * 32 + (lane & 31);
+        bf_[ni] = us8_to_bf8v(lds[buf * 2 * ASZ + ASZ + brow * MXP + TnStage<TnRowMajor>::rslot(slot, brow)]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < nk) {
+      if (buf == 0) {
+        sa2.write(tid, lds + 2 * ASZ);
+        sb2.write(tid, lds + 3 * ASZ);
+      } else {
+        sa.write(tid, lds);
+        sb.write(tid, lds + ASZ);
+      }
+    }
+    __syncthreads();
+  }
+  cptr += split * sstride;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        if (row >= M) continue;
+        cptr[(long)row * N + col] = acc[mi][ni][r];
+      }
+    }
+}
+
+static float run_d2(const uint16_t *a, const uint16_t *b, float *c, int M,
+                    int N, long Kpix, int splits, int iters) {
+  TnRowMajor la{a, (long)M, (int)Kpix, M};
+  TnRowMajor lb{b, (long)N, (int)Kpix, N};
+  int tiles_m = (M + 127) / 128, tiles_n = (N + 127) / 128;
+  int nk = (int)((Kpix + 63) / 64);
+  if (splits > nk) splits = nk;
+  int kt = (nk + splits - 1) / splits;
+  dim3 grid(tiles_m * tiles_n, splits);
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  dist2_k<<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
+                                c, M, N, (int)Kpix, tiles_n, kt, (long)M * N);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    dist2_k<<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la},
+                                  TnStage<TnRowMajor>{lb}, c, M, N, (int)Kpix,
+                                  tiles_n, kt, (long)M * N);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  return ms / iters;
+}
+
 int main() {
   // b2-1x1 wgrad shape: dw[128][512] over M=50176 pixels
   int M = 128, N = 512;
@@ -318,6 +434,12 @@ int main() {
   float wide = run_wide(a, b, c, M, N, Kpix, splits, iters);
   printf("wide128x256 %.1f us  %.1f TF (vs full %.1f)\n", wide * 1e3,
          gf / (wide / 1e3), gf / (full / 1e3));
+  float d2 = run_d2(a, b, c, M, N, Kpix, splits, iters);
+  printf("dist2     %.1f us  %.1f TF\n", d2 * 1e3, gf / (d2 / 1e3));
+  // interleaved repeat for noise bounds
+  float fullb = run<0>(a, b, c, M, N, Kpix, splits, iters);
+  float d2b = run_d2(a, b, c, M, N, Kpix, splits, iters);
+  printf("repeat: full %.1f  dist2 %.1f us\n", fullb * 1e3, d2b * 1e3);
   // a conv3x3-wgrad-ish wider shape: dw[256][1152] over 50176 pixels
   {
     int M2 = 256, N2 = 1152;
@@ -330,8 +452,10 @@ int main() {
     double gf2 = 2.0 * M2 * N2 * Kpix / 1e12;
     float f2 = run<0>(a2, b2, c2, M2, N2, Kpix, 48, iters);
     float w2 = run_wide(a2, b2, c2, M2, N2, Kpix, 48, iters);
-    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f us %.1f TF\n",
-           f2 * 1e3, gf2 / (f2 / 1e3), w2 * 1e3, gf2 / (w2 / 1e3));
+    float d22 = run_d2(a2, b2, c2, M2, N2, Kpix, 48, iters);
+    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f us %.1f | dist2 %.1f us %.1f TF\n",
+           f2 * 1e3, gf2 / (f2 / 1e3), w2 * 1e3, gf2 / (w2 / 1e3),
+           d22 * 1e3, gf2 / (d22 / 1e3));
   }
   return 0;
 }
