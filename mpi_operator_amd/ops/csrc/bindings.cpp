@@ -144,6 +144,9 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   int splits = std::max(512 / tiles, 1);
   splits = std::min(splits, std::max(nk / 8, 1));
   splits = std::min(splits, 256);
+  // multiple of 8 so the split-major grid keeps a panel's N-tile sharers
+  // on one XCD ((split + splits*tile) % 8 must not walk with tile)
+  if (splits >= 8) splits &= ~7;
   Tensor partial = at::empty({(long)splits, (long)Kout, (long)RSC},
                              x.options().dtype(at::kFloat));
   // dw bf16 (fp32-accumulated in the split-K slabs, rounded once at the

@@ -223,10 +223,15 @@ template <class SA, class SB, bool C_F32, class WR = LinearWriter,
 __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
     int tiles_n, int kt_per_split, long split_stride, int xcd_cpx) {
-  int tile = blockIdx.x;
+  // SPLIT-MAJOR grid (x = split, y = tile): the N-tiles sharing an A panel
+  // are y-adjacent, which lands them on ONE XCD's L2 when gridDim.x%8==0.
+  // Tile-major dispatch spread a panel's sharers over 4+ XCDs and re-read
+  // the panel from HBM every time (ablation: staging ran at full HBM BW;
+  // this swap alone measured +16-28% on split-K wgrad shapes).
+  int tile = blockIdx.y;
   if (xcd_cpx) // T1: contiguous tile chunk per XCD (L2 reuse of panels)
     tile = (tile & 7) * xcd_cpx + (tile >> 3);
-  int split = blockIdx.y;
+  int split = blockIdx.x;
   int tm = tile / tiles_n, tn = tile % tiles_n;
   int row0 = tm * BM, col0 = tn * BN;
   int tid = threadIdx.x;
@@ -348,8 +353,12 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   // T1 XCD swizzle: give each XCD a contiguous chunk of tiles so neighbor
   // tiles (sharing operand panels) hit the same per-XCD L2. Needs nwg%8==0
   // and enough tiles to matter.
-  int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
-  dim3 grid(nwg, splits);
+  // T1 swizzle only when one XCD still sees many tiles AND splits don't
+  // already interleave the XCD mapping (splits-major x dominates (linear%8)
+  // unless splits%8==0, in which case tile adjacency passes through)
+  int cpx = (nwg % 8 == 0 && nwg >= 32 && (splits == 1 || splits % 8 == 0))
+                ? nwg / 8 : 0;
+  dim3 grid(splits, nwg);
   static const bool onebuf_env = [] {
     const char *e = getenv("MPIAMD_GEMM_ONEBUF");
     return e && e[0] == '1';

@@ -12,12 +12,15 @@
 
 // variant bits: 1 = skip global loads, 2 = skip MFMA, 4 = skip epilogue,
 // 8 = skip LDS writes (with 1: pure-LDS-read+MFMA loop)
-template <int SKIP>
+template <int SKIP, bool SWAPG = false>
 __global__ __launch_bounds__(NT_THREADS) void ablate_k(
     TnStage<TnRowMajor> sa, TnStage<TnRowMajor> sb, float *cptr, int M, int N,
     int K, int tiles_n, int kt, long sstride) {
-  int tile = blockIdx.x;
-  int split = blockIdx.y;
+  // SWAPG: split-major grid.x — co-locates a tile-row's N-tile sharers on
+  // ONE XCD's L2 (x-major dispatch put them on 4 different XCDs and the
+  // A-panel re-reads all went to HBM)
+  int tile = SWAPG ? blockIdx.y : blockIdx.x;
+  int split = SWAPG ? blockIdx.x : blockIdx.y;
   int tm = tile / tiles_n, tn = tile % tiles_n;
   int row0 = tm * BM, col0 = tn * BN;
   int tid = threadIdx.x;
@@ -90,7 +93,7 @@ __global__ __launch_bounds__(NT_THREADS) void ablate_k(
     }
 }
 
-template <int SKIP>
+template <int SKIP, bool SWAPG = false>
 static float run(const uint16_t *a, const uint16_t *b, float *c, int M, int N,
                  long Kpix, int splits, int iters) {
   TnRowMajor la{a, (long)M, (int)Kpix, M};
@@ -100,17 +103,18 @@ static float run(const uint16_t *a, const uint16_t *b, float *c, int M, int N,
   if (splits > nk) splits = nk;
   int kt = (nk + splits - 1) / splits;
   dim3 grid(tiles_m * tiles_n, splits);
+  if (SWAPG) grid = dim3(splits, tiles_m * tiles_n);
   hipEvent_t e0, e1;
   hipEventCreate(&e0);
   hipEventCreate(&e1);
   // warmup
-  ablate_k<SKIP><<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la},
+  ablate_k<SKIP, SWAPG><<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la},
                                        TnStage<TnRowMajor>{lb}, c, M, N,
                                        (int)Kpix, tiles_n, kt, (long)M * N);
   hipDeviceSynchronize();
   hipEventRecord(e0);
   for (int i = 0; i < iters; ++i)
-    ablate_k<SKIP><<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la},
+    ablate_k<SKIP, SWAPG><<<grid, NT_THREADS>>>(TnStage<TnRowMajor>{la},
                                          TnStage<TnRowMajor>{lb}, c, M, N,
                                          (int)Kpix, tiles_n, kt, (long)M * N);
   hipEventRecord(e1);
@@ -436,10 +440,12 @@ int main() {
          gf / (wide / 1e3), gf / (full / 1e3));
   float d2 = run_d2(a, b, c, M, N, Kpix, splits, iters);
   printf("dist2     %.1f us  %.1f TF\n", d2 * 1e3, gf / (d2 / 1e3));
+  float sw = run<0, true>(a, b, c, M, N, Kpix, splits, iters);
+  printf("swapgrid  %.1f us  %.1f TF\n", sw * 1e3, gf / (sw / 1e3));
   // interleaved repeat for noise bounds
   float fullb = run<0>(a, b, c, M, N, Kpix, splits, iters);
-  float d2b = run_d2(a, b, c, M, N, Kpix, splits, iters);
-  printf("repeat: full %.1f  dist2 %.1f us\n", fullb * 1e3, d2b * 1e3);
+  float swb = run<0, true>(a, b, c, M, N, Kpix, splits, iters);
+  printf("repeat: full %.1f  swap %.1f us\n", fullb * 1e3, swb * 1e3);
   // a conv3x3-wgrad-ish wider shape: dw[256][1152] over 50176 pixels
   {
     int M2 = 256, N2 = 1152;
@@ -453,9 +459,10 @@ int main() {
     float f2 = run<0>(a2, b2, c2, M2, N2, Kpix, 48, iters);
     float w2 = run_wide(a2, b2, c2, M2, N2, Kpix, 48, iters);
     float d22 = run_d2(a2, b2, c2, M2, N2, Kpix, 48, iters);
-    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f us %.1f | dist2 %.1f us %.1f TF\n",
+    float sw2 = run<0, true>(a2, b2, c2, M2, N2, Kpix, 48, iters);
+    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f %.1f | dist2 %.1f %.1f | swap %.1f us %.1f TF\n",
            f2 * 1e3, gf2 / (f2 / 1e3), w2 * 1e3, gf2 / (w2 / 1e3),
-           d22 * 1e3, gf2 / (d22 / 1e3));
+           d22 * 1e3, gf2 / (d22 / 1e3), sw2 * 1e3, gf2 / (sw2 / 1e3));
   }
   return 0;
 }
