@@ -222,16 +222,16 @@ template <class SA, class SB, bool C_F32, class WR = LinearWriter,
           bool ONEBUF = false>
 __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
     SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
-    int tiles_n, int kt_per_split, long split_stride, int xcd_cpx) {
-  // SPLIT-MAJOR grid (x = split, y = tile): the N-tiles sharing an A panel
-  // are y-adjacent, which lands them on ONE XCD's L2 when gridDim.x%8==0.
-  // Tile-major dispatch spread a panel's sharers over 4+ XCDs and re-read
-  // the panel from HBM every time (ablation: staging ran at full HBM BW;
-  // this swap alone measured +16-28% on split-K wgrad shapes).
-  int tile = blockIdx.y;
+    int tiles_n, int kt_per_split, long split_stride, int xcd_cpx, int swap) {
+  // SPLIT-MAJOR grid (x = split, y = tile) when swap: the N-tiles sharing
+  // an A panel are y-adjacent, which lands them on ONE XCD's L2 when
+  // gridDim.x%8==0. Tile-major dispatch spread a panel's sharers over 4+
+  // XCDs and the panel re-reads all went to HBM (ablation: staging at full
+  // HBM BW; +16-28% on split-K wgrad shapes in the standalone A/B).
+  int tile = swap ? blockIdx.y : blockIdx.x;
+  int split = swap ? blockIdx.x : blockIdx.y;
   if (xcd_cpx) // T1: contiguous tile chunk per XCD (L2 reuse of panels)
     tile = (tile & 7) * xcd_cpx + (tile >> 3);
-  int split = blockIdx.x;
   int tm = tile / tiles_n, tn = tile % tiles_n;
   int row0 = tm * BM, col0 = tn * BN;
   int tid = threadIdx.x;
@@ -353,12 +353,13 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   // T1 XCD swizzle: give each XCD a contiguous chunk of tiles so neighbor
   // tiles (sharing operand panels) hit the same per-XCD L2. Needs nwg%8==0
   // and enough tiles to matter.
-  // T1 swizzle only when one XCD still sees many tiles AND splits don't
-  // already interleave the XCD mapping (splits-major x dominates (linear%8)
-  // unless splits%8==0, in which case tile adjacency passes through)
-  int cpx = (nwg % 8 == 0 && nwg >= 32 && (splits == 1 || splits % 8 == 0))
-                ? nwg / 8 : 0;
-  dim3 grid(splits, nwg);
+  static const int swap_env = [] {
+    const char *e = getenv("MPIAMD_SPLITMAJOR");
+    return e && e[0] == '0' ? 0 : 1;
+  }();
+  int swap = (splits > 1 && splits % 8 == 0) ? swap_env : 0;
+  int cpx = (nwg % 8 == 0 && nwg >= 32 && (splits == 1 || swap)) ? nwg / 8 : 0;
+  dim3 grid = swap ? dim3(splits, nwg) : dim3(nwg, splits);
   static const bool onebuf_env = [] {
     const char *e = getenv("MPIAMD_GEMM_ONEBUF");
     return e && e[0] == '1';
@@ -369,18 +370,18 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   if (onebuf) {
     if (c_f32)
       mix_gemm_k<SA, SB, true, WR, true><<<grid, NT_THREADS, 0, s>>>(
-          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, swap);
     else
       mix_gemm_k<SA, SB, false, WR, true><<<grid, NT_THREADS, 0, s>>>(
-          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+          sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, swap);
     return hipGetLastError();
   }
   if (c_f32)
     mix_gemm_k<SA, SB, true, WR><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, swap);
   else
     mix_gemm_k<SA, SB, false, WR><<<grid, NT_THREADS, 0, s>>>(
-        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx);
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, swap);
   return hipGetLastError();
 }
 
