@@ -353,9 +353,12 @@ static hipError_t launch_mix_gemm_wr(const SA &sa, const SB &sb, void *c,
   // T1 XCD swizzle: give each XCD a contiguous chunk of tiles so neighbor
   // tiles (sharing operand panels) hit the same per-XCD L2. Needs nwg%8==0
   // and enough tiles to matter.
+  // default OFF: the standalone-harness win (+16-28% on synthetic TnTn
+  // wgrad shapes) did not transfer to the full model (-0.4% same-box on the
+  // bench) — real wgrads gather B via XcolStage and use smaller splits
   static const int swap_env = [] {
     const char *e = getenv("MPIAMD_SPLITMAJOR");
-    return e && e[0] == '0' ? 0 : 1;
+    return e && e[0] == '1' ? 1 : 0;
   }();
   int swap = (splits > 1 && splits % 8 == 0) ? swap_env : 0;
   int cpx = (nwg % 8 == 0 && nwg >= 32 && (splits == 1 || swap)) ? nwg / 8 : 0;
