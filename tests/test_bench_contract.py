@@ -1,0 +1,40 @@
+"""The driver contract on bench.py: runs on CPU (tiny shapes), prints ONE
+JSON line with the required fields, and supports the --impl torch A/B."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_bench(*args):
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run([sys.executable, os.path.join(REPO, "bench.py"),
+                        "--steps", "2", "--warmup", "1", *args],
+                       capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    return json.loads(lines[0])
+
+
+def test_bench_json_contract():
+    out = run_bench()
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in out, key
+    assert out["metric"] == "images/sec"
+    assert out["n_gpus"] == 1
+    assert out["steps"] == 2
+    assert out["scaling"] == "weak"
+    assert out["data"] == "synthetic"
+    assert out["config"]["model"] == "resnet101"
+    assert out["value"] > 0
+
+
+def test_bench_torch_impl_ab_path():
+    out = run_bench("--impl", "torch", "--model", "resnet50")
+    assert out["config"]["impl"] == "torch"
+    assert out["value"] > 0
