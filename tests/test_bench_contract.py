@@ -38,3 +38,21 @@ def test_bench_torch_impl_ab_path():
     out = run_bench("--impl", "torch", "--model", "resnet50")
     assert out["config"]["impl"] == "torch"
     assert out["value"] > 0
+
+
+def test_bench_under_torchrun_two_ranks():
+    """The driver's exact N>1 launch shape: torchrun, one JSON from rank 0,
+    whole-job aggregate value, allreduce over gloo."""
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout  # rank 0 only
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
