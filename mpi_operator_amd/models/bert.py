@@ -23,6 +23,24 @@ import torch.nn.functional as F
 from ..ops import functional as Fx
 
 
+class LayerNorm(nn.Module):
+    """LayerNorm dispatching to the fused HIP kernel on GPU bf16 inputs
+    (fp32 statistics either way); params stay fp32-computed via .float()."""
+
+    def __init__(self, n: int, eps: float = 1e-12):
+        super().__init__()
+        self.n, self.eps = n, eps
+        self.weight = nn.Parameter(torch.ones(n))
+        self.bias = nn.Parameter(torch.zeros(n))
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            return Fx.layer_norm(x, self.weight.float(), self.bias.float(),
+                                 self.eps)
+        return F.layer_norm(x.float(), (self.n,), self.weight.float(),
+                            self.bias.float(), self.eps).to(x.dtype)
+
+
 @dataclass
 class BertConfig:
     vocab_size: int = 30522
@@ -90,10 +108,10 @@ class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.attn = SelfAttention(cfg)
-        self.ln1 = nn.LayerNorm(cfg.hidden, eps=cfg.eps)
+        self.ln1 = LayerNorm(cfg.hidden, eps=cfg.eps)
         self.fc1 = BertLinear(cfg.hidden, cfg.intermediate)
         self.fc2 = BertLinear(cfg.intermediate, cfg.hidden)
-        self.ln2 = nn.LayerNorm(cfg.hidden, eps=cfg.eps)
+        self.ln2 = LayerNorm(cfg.hidden, eps=cfg.eps)
 
     def forward(self, x, attn_mask=None):
         x = self.ln1(x + self.attn(x, attn_mask))
@@ -107,7 +125,7 @@ class BertEmbeddings(nn.Module):
         self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
         self.pos = nn.Embedding(cfg.max_seq, cfg.hidden)
         self.typ = nn.Embedding(cfg.type_vocab, cfg.hidden)
-        self.ln = nn.LayerNorm(cfg.hidden, eps=cfg.eps)
+        self.ln = LayerNorm(cfg.hidden, eps=cfg.eps)
         for e in (self.tok, self.pos, self.typ):
             nn.init.normal_(e.weight, std=0.02)
 
@@ -145,7 +163,7 @@ class BertForPreTraining(nn.Module):
         self.cfg = cfg
         self.bert = BertModel(cfg)
         self.mlm_transform = BertLinear(cfg.hidden, cfg.hidden)
-        self.mlm_ln = nn.LayerNorm(cfg.hidden, eps=cfg.eps)
+        self.mlm_ln = LayerNorm(cfg.hidden, eps=cfg.eps)
         # decoder tied to token embeddings (standard BERT weight tying)
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.nsp = BertLinear(cfg.hidden, 2)

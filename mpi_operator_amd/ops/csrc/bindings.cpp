@@ -91,6 +91,11 @@ hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
                                int, int, int, int, int, int, int, int, int,
                                int, int, int, int, hipStream_t);
 hipError_t mfma_probe(const void *, const void *, float *, hipStream_t);
+hipError_t ln_fwd(const void *, const float *, const float *, void *, float *,
+                  float *, long, int, float, hipStream_t);
+hipError_t ln_bwd(const void *, const void *, const float *, const float *,
+                  const float *, void *, float *, float *, long, int, int *,
+                  hipStream_t);
 }
 
 // ------------------------- conv -------------------------
@@ -300,6 +305,44 @@ static Tensor gap_bwd_b(const Tensor &dy, int64_t H, int64_t W) {
   return dx;
 }
 
+// ------------------------- layernorm -------------------------
+static std::vector<Tensor> layernorm_fwd(const Tensor &x, const Tensor &gamma,
+                                         const Tensor &beta, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  const HIPDeviceGuard guard(x.device());
+  Tensor xc = x.contiguous();
+  long M = xc.numel() / xc.size(-1);
+  int N = xc.size(-1);
+  auto f32 = x.options().dtype(at::kFloat);
+  Tensor y = at::empty_like(xc);
+  Tensor mean = at::empty({M}, f32), rstd = at::empty({M}, f32);
+  CHK(ln_fwd(xc.data_ptr(), gamma.data_ptr<float>(), beta.data_ptr<float>(),
+             y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), M,
+             N, (float)eps, cur_stream()));
+  return {y, mean, rstd};
+}
+
+static std::vector<Tensor> layernorm_bwd(const Tensor &dy, const Tensor &x,
+                                         const Tensor &gamma,
+                                         const Tensor &mean,
+                                         const Tensor &rstd) {
+  const HIPDeviceGuard guard(x.device());
+  Tensor dyc = dy.contiguous(), xc = x.contiguous();
+  long M = xc.numel() / xc.size(-1);
+  int N = xc.size(-1);
+  auto f32 = x.options().dtype(at::kFloat);
+  long grid_max = (M + 3) / 4;
+  if (grid_max > 1024) grid_max = 1024;
+  Tensor partial = at::empty({grid_max, 2L * N}, f32);
+  Tensor dgb = at::empty({2L * N}, f32);
+  Tensor dx = at::empty_like(xc);
+  CHK(ln_bwd(dyc.data_ptr(), xc.data_ptr(), gamma.data_ptr<float>(),
+             mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
+             partial.data_ptr<float>(), dgb.data_ptr<float>(), M, N, nullptr,
+             cur_stream()));
+  return {dx, dgb.narrow(0, 0, N), dgb.narrow(0, N, N)};
+}
+
 // ------------------------- linear -------------------------
 static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
@@ -439,6 +482,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("gap_fwd", &gap_fwd_b);
   m.def("gap_bwd", &gap_bwd_b);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);

@@ -191,6 +191,28 @@ def linear(x, w, b):
     return LinearFn.apply(x, w, b)
 
 
+class LayerNormFn(torch.autograd.Function):
+    """Fused LayerNorm over the last dim (bf16 activations, fp32 params) —
+    one wave per row, shfl row reduction; bwd reduces dgamma/dbeta through
+    per-block fp32 slabs (ops/csrc/layernorm.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps: float):
+        y, mean, rstd = hip_ext().layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dg, db = hip_ext().layernorm_bwd(dy, x, weight, mean, rstd)
+        return dx, dg, db, None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-12):
+    return LayerNormFn.apply(x, weight, bias, eps)
+
+
 class SoftmaxCrossEntropyFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, target):
