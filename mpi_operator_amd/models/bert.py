@@ -23,9 +23,16 @@ import torch.nn.functional as F
 from ..ops import functional as Fx
 
 
+import os as _os
+
+_FUSED_LN = _os.environ.get("MPIAMD_FUSED_LN") == "1"
+
+
 class LayerNorm(nn.Module):
-    """LayerNorm dispatching to the fused HIP kernel on GPU bf16 inputs
-    (fp32 statistics either way); params stay fp32-computed via .float()."""
+    """LayerNorm with fp32 statistics. The hand-written HIP kernel
+    (ops/csrc/layernorm.hip, numerics-tested) is behind MPIAMD_FUSED_LN=1;
+    torch's native LN measured faster end-to-end on BERT-Large (same-box:
+    fused 51.3 ms/step vs native 42.8), so native is the default."""
 
     def __init__(self, n: int, eps: float = 1e-12):
         super().__init__()
@@ -34,7 +41,7 @@ class LayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(n))
 
     def forward(self, x):
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _FUSED_LN and x.is_cuda and x.dtype == torch.bfloat16:
             return Fx.layer_norm(x, self.weight.float(), self.bias.float(),
                                  self.eps)
         return F.layer_norm(x.float(), (self.n,), self.weight.float(),
