@@ -410,6 +410,173 @@ static float run_d2(const uint16_t *a, const uint16_t *b, float *c, int M,
   return ms / iters;
 }
 
+// ---- experiment: BK=128, single LDS buffer (2 barriers/step) ----
+// Twice the compute window per staged tile: loads get ~1024 MFMA cycles of
+// cover instead of ~512, and half the barrier crossings per FLOP.
+constexpr int MXP2 = 17; // 16 slots + 1 pad (bank spread for b128 reads)
+__global__ __launch_bounds__(NT_THREADS) void bk128_k(
+    TnRowMajor la, TnRowMajor lb, float *cptr, int M, int N, int K,
+    int tiles_n, int kt, long sstride) {
+  constexpr int ASZ = BM * MXP2;
+  __shared__ ushort8 lds[2 * ASZ];
+  int tile = blockIdx.x, split = blockIdx.y;
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * BM, col0 = tn * BN;
+  int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 1, wc = wave & 1;
+  float16v acc[2][2] = {};
+  int nk_total = (K + 127) / 128;
+  int t0 = split * kt;
+  int nk = min(kt, nk_total - t0);
+  if (nk < 0) nk = 0;
+  ushort8 ra[4], rb[4], ra2[4], rb2[4];
+  // pair mapping per k-half: 512 pairs (64 k x 16 col-octets), two halves
+  // in separate register sets so all 16 loads issue before the MFMAs
+  auto loadI = [&](TnRowMajor &l, int base, int kb, ushort8 *r) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;   // [0,512): k-pairs 64 x col-octets 16? no:
+      int k0 = (p >> 4) * 2;    // [0,64)? need 128 k: use 2 loads per pair
+      int c0 = (p & 15) * 8;
+      r[it * 2] = l.load(kb + k0, base + c0);
+      r[it * 2 + 1] = l.load(kb + k0 + 1, base + c0);
+    }
+  };
+  auto writeI = [&](ushort8 *img, const ushort8 *r, int khalf) {
+    uint32_t *im = (uint32_t *)img;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int p = tid + it * 256;
+      int k0 = (p >> 4) * 2 + khalf * 64;
+      int c0 = (p & 15) * 8;
+      int slot = k0 >> 3, within = (k0 & 7) >> 1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int row = c0 + j;
+        im[row * (MXP2 * 4) + (slot ^ ((row >> 3) & 15)) * 4 + within] =
+            (uint32_t)r[it * 2][j] | ((uint32_t)r[it * 2 + 1][j] << 16);
+      }
+    }
+  };
+  // prologue: stage tile 0 (both k-halves)
+  if (nk > 0) {
+    loadI(la, row0, t0 * 128, ra);
+    loadI(lb, col0, t0 * 128, rb);
+    loadI(la, row0, t0 * 128 + 64, ra2);
+    loadI(lb, col0, t0 * 128 + 64, rb2);
+    writeI(lds, ra, 0);
+    writeI(lds + ASZ, rb, 0);
+    writeI(lds, ra2, 1);
+    writeI(lds + ASZ, rb2, 1);
+  }
+  __syncthreads();
+  for (int t = 0; t < nk; ++t) {
+    // issue BOTH k-halves of the next tile before the MFMAs (16 loads in
+    // flight with ~1024 MFMA cycles of cover)
+    if (t + 1 < nk) {
+      loadI(la, row0, (t0 + t + 1) * 128, ra);
+      loadI(lb, col0, (t0 + t + 1) * 128, rb);
+      loadI(la, row0, (t0 + t + 1) * 128 + 64, ra2);
+      loadI(lb, col0, (t0 + t + 1) * 128 + 64, rb2);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) { // 8 k-steps of 16 = 128
+      bf16x8 af[2], bf_[2];
+      int slot = kk * 2 + (lane >> 5);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int arow = wr * 64 + mi * 32 + (lane & 31);
+        af[mi] = us8_to_bf8v(lds[arow * MXP2 + (slot ^ ((arow >> 3) & 15))]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int brow = wc * 64 + ni * 32 + (lane & 31);
+        bf_[ni] = us8_to_bf8v(lds[ASZ + brow * MXP2 + (slot ^ ((brow >> 3) & 15))]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < nk) {
+      __syncthreads(); // readers done with the single buffer
+      writeI(lds, ra, 0);
+      writeI(lds + ASZ, rb, 0);
+      writeI(lds, ra2, 1);
+      writeI(lds + ASZ, rb2, 1);
+    }
+    __syncthreads();
+  }
+  cptr += split * sstride;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        if (row >= M) continue;
+        cptr[(long)row * N + col] = acc[mi][ni][r];
+      }
+    }
+}
+
+static float run_bk128(const uint16_t *a, const uint16_t *b, float *c, int M,
+                       int N, long Kpix, int splits, int iters, bool check) {
+  TnRowMajor la{a, (long)M, (int)Kpix, M};
+  TnRowMajor lb{b, (long)N, (int)Kpix, N};
+  int tiles_m = (M + 127) / 128, tiles_n = (N + 127) / 128;
+  int nk = (int)((Kpix + 127) / 128);
+  if (splits > nk) splits = nk;
+  int kt = (nk + splits - 1) / splits;
+  dim3 grid(tiles_m * tiles_n, splits);
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  bk128_k<<<grid, NT_THREADS>>>(la, lb, c, M, N, (int)Kpix, tiles_n, kt, (long)M * N);
+  hipDeviceSynchronize();
+  if (check) return 0.f;
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    bk128_k<<<grid, NT_THREADS>>>(la, lb, c, M, N, (int)Kpix, tiles_n, kt, (long)M * N);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  return ms / iters;
+}
+
+static void check_bk128() {
+  int M = 128, N = 256;
+  long Kp = 384;
+  uint16_t *a, *b;
+  float *c;
+  hipMallocManaged(&a, Kp * M * 2);
+  hipMallocManaged(&b, Kp * N * 2);
+  hipMallocManaged(&c, (long)M * N * 4);
+  auto f2b = [](float f) { unsigned u; __builtin_memcpy(&u, &f, 4); return (uint16_t)(u >> 16); };
+  auto b2f = [](uint16_t h) { unsigned u = (unsigned)h << 16; float f; __builtin_memcpy(&f, &u, 4); return f; };
+  srand(11);
+  for (long i = 0; i < Kp * M; ++i) a[i] = f2b((rand() % 2000 - 1000) / 997.f);
+  for (long i = 0; i < Kp * N; ++i) b[i] = f2b((rand() % 2000 - 1000) / 997.f);
+  run_bk128(a, b, c, M, N, Kp, 1, 1, true);
+  hipDeviceSynchronize();
+  double maxerr = 0;
+  for (int i = 0; i < M; i += 5)
+    for (int j = 0; j < N; j += 11) {
+      double ref = 0;
+      for (long k = 0; k < Kp; ++k) ref += (double)b2f(a[k * M + i]) * b2f(b[k * N + j]);
+      double err = fabs(c[(long)i * N + j] - ref) / (fabs(ref) + 1.0);
+      if (err > maxerr) maxerr = err;
+    }
+  printf("bk128 check maxrelerr %.4g %s\n", maxerr, maxerr < 0.02 ? "OK" : "FAIL");
+  hipFree(a); hipFree(b); hipFree(c);
+}
+
 int main() {
   // b2-1x1 wgrad shape: dw[128][512] over M=50176 pixels
   int M = 128, N = 512;
@@ -446,6 +613,9 @@ int main() {
   float fullb = run<0>(a, b, c, M, N, Kpix, splits, iters);
   float swb = run<0, true>(a, b, c, M, N, Kpix, splits, iters);
   printf("repeat: full %.1f  swap %.1f us\n", fullb * 1e3, swb * 1e3);
+  check_bk128();
+  float bk = run_bk128(a, b, c, M, N, Kpix, splits, iters, false);
+  printf("bk128    %.1f us  %.1f TF\n", bk * 1e3, gf / (bk / 1e3));
   // a conv3x3-wgrad-ish wider shape: dw[256][1152] over 50176 pixels
   {
     int M2 = 256, N2 = 1152;
@@ -460,10 +630,13 @@ int main() {
     float w2 = run_wide(a2, b2, c2, M2, N2, Kpix, 48, iters);
     float d22 = run_d2(a2, b2, c2, M2, N2, Kpix, 48, iters);
     float sw2 = run<0, true>(a2, b2, c2, M2, N2, Kpix, 48, iters);
-    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f %.1f | dist2 %.1f %.1f | swap %.1f us %.1f TF\n",
+    float bk2 = run_bk128(a2, b2, c2, M2, N2, Kpix, 48, iters, false);
+    printf("3x3ish 128x128 %.1f us %.1f TF | wide %.1f %.1f | dist2 %.1f %.1f | swap %.1f %.1f | bk128 %.1f us %.1f TF\n",
            f2 * 1e3, gf2 / (f2 / 1e3), w2 * 1e3, gf2 / (w2 / 1e3),
-           d22 * 1e3, gf2 / (d22 / 1e3), sw2 * 1e3, gf2 / (sw2 / 1e3));
+           d22 * 1e3, gf2 / (d22 / 1e3), sw2 * 1e3, gf2 / (sw2 / 1e3),
+           bk2 * 1e3, gf2 / (bk2 / 1e3));
   }
   return 0;
 }
+
 
