@@ -161,8 +161,11 @@ class MPIJobController:
                     self._get_or_create_pod_group(job)
                 workers = self._get_or_create_workers(job)
             if launcher is None:
+                # WaitForWorkersReady gates on the DESIRED replica count:
+                # comparing against the created list would let a suspended
+                # job (workers == []) create its launcher early
                 if t.launcher_creation_policy(job) == c.LAUNCHER_CREATION_AT_STARTUP \
-                        or self._count_ready(workers) == len(workers):
+                        or self._count_ready(workers) == t.worker_replicas(job):
                     try:
                         launcher = self.client.jobs.create(
                             namespace,

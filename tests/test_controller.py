@@ -325,3 +325,16 @@ def test_invalid_job_emits_warning_no_requeue():
     evs = client.events.list("default")
     assert any(e["reason"] == c.VALIDATION_ERROR for e in evs)
     assert client.actions_of("create", "pods") == []
+
+
+def test_wait_for_workers_ready_suspended_does_not_create_launcher():
+    """Suspended + WaitForWorkersReady: no workers exist, so the launcher
+    must NOT be created (the gate counts desired replicas, not the empty
+    created list)."""
+    from mpi_operator_amd.controller.client.base import NotFound
+    client, ctrl = make_controller()
+    seed_and_sync(client, ctrl, make_job(
+        launcherCreationPolicy="WaitForWorkersReady",
+        runPolicy={"suspend": True}))
+    with pytest.raises(NotFound):
+        client.jobs.get("default", "test-launcher")
