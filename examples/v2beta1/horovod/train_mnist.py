@@ -23,8 +23,10 @@ from mpi_operator_amd.models import SimpleCNN
 from mpi_operator_amd.parallel import elastic
 
 
-def synthetic_mnist(batch, device):
+def synthetic_mnist(batch, device, bf16):
     x = torch.randn(batch, 1, 28, 28, device=device)
+    if bf16:
+        x = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
     y = torch.randint(0, 10, (batch,), device=device)
     return x, y
 
@@ -37,9 +39,17 @@ def main():
     args = ap.parse_args()
 
     hvd.init()
-    device = f"cuda:{hvd.local_rank()}" if torch.cuda.is_available() else "cpu"
+    use_cuda = torch.cuda.is_available()
+    device = f"cuda:{hvd.local_rank()}" if use_cuda else "cpu"
 
-    model = SimpleCNN(in_ch=1, num_classes=10).to(device)
+    model = SimpleCNN(in_ch=1, num_classes=10)
+    if use_cuda:
+        # MI355X layout/dtype contract: bf16 conv/linear weights,
+        # channels-last memory, fp32 BN stats
+        from mpi_operator_amd.models import to_mi355x
+        model = to_mi355x(model, device)
+    else:
+        model = model.to(device)
     # lr scaled by world size, as in the reference example (lr * hvd.size())
     opt = torch.optim.Adam(model.parameters(), lr=args.lr * hvd.size())
     opt = hvd.DistributedOptimizer(opt, model.named_parameters())
@@ -54,7 +64,7 @@ def main():
     model.train()
     # StopAtStepHook equivalent: total steps divided across the world
     for step in range(start_step, args.steps // hvd.size()):
-        x, y = synthetic_mnist(args.batch, device)
+        x, y = synthetic_mnist(args.batch, device, use_cuda)
         opt.zero_grad()
         loss = loss_fn(model(x), y)
         loss.backward()

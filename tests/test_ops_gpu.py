@@ -240,3 +240,21 @@ def test_resnet50_train_step_gpu():
         losses.append(float(loss))
     assert all(l == l for l in losses), losses  # no NaN
     assert losses[-1] < losses[0], losses  # memorizing a fixed batch
+
+
+def test_simple_cnn_train_step_gpu():
+    """The mnist example model on the GPU contract (bf16 channels-last):
+    one fwd+bwd+step, loss finite."""
+    from mpi_operator_amd import models
+    from mpi_operator_amd.optim import FusedSGD
+    torch.manual_seed(5)
+    m = models.to_mi355x(models.SimpleCNN(in_ch=1, num_classes=10), "cuda")
+    m.train()
+    opt = FusedSGD(m.parameters(), lr=0.01)
+    x = torch.randn(8, 1, 28, 28, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (8,), device="cuda")
+    loss = m.loss(m(x), y)
+    loss.backward()
+    opt.step()
+    assert float(loss) == float(loss)
