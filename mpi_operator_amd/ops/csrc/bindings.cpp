@@ -13,6 +13,19 @@ namespace {
 
 using at::Tensor;
 
+// Late-stage convs produce few 128x128 output tiles (7x7 spatial ~100
+// blocks on 256 CUs): split the reduce dim so fwd/dgrad fill the chip,
+// paying one fp32 slab + reduce pass.
+int conv_splits(long M, int Ncols, int K) {
+  long tiles = ((M + 127) / 128) * ((Ncols + 127) / 128);
+  int nk = (K + 63) / 64;
+  if (tiles >= 256 || nk < 16) return 1;
+  long s = 512 / tiles;
+  if (s > nk / 8) s = nk / 8;
+  if (s > 16) s = 16;
+  return s < 1 ? 1 : (int)s;
+}
+
 #define CHK(call)                                                              \
   do {                                                                         \
     hipError_t e_ = (call);                                                    \
@@ -82,9 +95,11 @@ hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
-                    int, int, int, int, int, int, int, hipStream_t);
+                    int, int, int, int, int, int, int, int, float *,
+                    hipStream_t);
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
-                      int, int, int, int, int, int, int, hipStream_t);
+                      int, int, int, int, int, int, int, int, float *,
+                      hipStream_t);
 hipError_t conv_dgrad_1x1_acc(const void *, const void *, void *, long, int,
                               int, hipStream_t);
 hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
@@ -111,8 +126,17 @@ static Tensor conv2d_fwd(const Tensor &x, const Tensor &w, int64_t stride,
   int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
   int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
   Tensor y = empty_cl_bf16(N, Kout, HO, WO, x);
+  long M = (long)N * HO * WO;
+  int splits = conv_splits(M, Kout, R * S * C);
+  Tensor partial;
+  float *pp = nullptr;
+  if (splits > 1) {
+    partial = at::empty({(long)splits, M, (long)Kout},
+                        x.options().dtype(at::kFloat));
+    pp = partial.data_ptr<float>();
+  }
   CHK(conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W, C, Kout, R,
-               S, (int)stride, (int)pad, HO, WO, cur_stream()));
+               S, (int)stride, (int)pad, HO, WO, splits, pp, cur_stream()));
   return y;
 }
 
@@ -125,9 +149,18 @@ static Tensor conv2d_dgrad(const Tensor &dy, const Tensor &w, int64_t H,
   int C = w.size(1), R = w.size(2), S = w.size(3);
   TORCH_CHECK(Kout % 8 == 0, "dgrad requires out-channels %8==0");
   Tensor dx = empty_cl_bf16(N, C, H, W, dy);
+  long M = (long)N * H * W;
+  int splits = stride == 1 ? conv_splits(M, C, R * S * Kout) : 1;
+  Tensor partial;
+  float *pp = nullptr;
+  if (splits > 1) {
+    partial = at::empty({(long)splits, M, (long)C},
+                        dy.options().dtype(at::kFloat));
+    pp = partial.data_ptr<float>();
+  }
   CHK(conv_dgrad(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N, (int)H,
-                 (int)W, C, Kout, R, S, (int)stride, (int)pad, HO, WO,
-                 cur_stream()));
+                 (int)W, C, Kout, R, S, (int)stride, (int)pad, HO, WO, splits,
+                 pp, cur_stream()));
   return dx;
 }
 

@@ -14,6 +14,9 @@
 #include "mfma_tile.h"
 #include "mix_gemm.h"
 
+extern "C" hipError_t splitk_reduce(const float *, int, long, void *, int,
+                                    hipStream_t);
+
 // Custom NT stager for conv fwd: the 4 staged rows (output pixels) are
 // FIXED for the whole kernel, so their (n, ho, wo) decomposition is hoisted
 // into init(); per k-step only the (r,s,c) split of one k remains — the
@@ -368,24 +371,32 @@ static hipError_t conv_dgrad_s2(const void *dy, const void *w, void *dx,
 extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
                                  int N, int H, int W, int C, int Kout, int R,
                                  int S, int stride, int pad, int HO, int WO,
+                                 int splits, float *partial,
                                  hipStream_t strm) {
   long M = (long)N * H * W;
   int K = R * S * Kout;
-  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+  if (stride != 1)
+    return conv_dgrad_s2(dy, w, dx, N, H, W, C, Kout, R, S, pad, HO, WO, strm);
+  int nk = (K + BK - 1) / BK;
+  if (splits > nk) splits = nk > 0 ? nk : 1;
+  void *out = splits > 1 ? (void *)partial : dx;
+  bool f32 = splits > 1;
+  hipError_t e;
+  if (R == 1 && S == 1 && pad == 0) {
     // 1x1 dgrad: dx[M][C] = dy[M][Q] · w[Q][C] (w TN-staged, no gather)
     GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
     TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
-    return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
-                           dx, (int)M, C, Kout, C, false, strm);
-  }
-  // TN B: k=(r,s,q) with q fastest; element (c, k) = w[q][(r*S+s)*C + c]
-  DgradWTn lb{(const uint16_t *)w, C, Kout, K, R * S * C};
-  if (stride == 1) {
+    e = launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, out,
+                        (int)M, C, Kout, C, f32, strm, splits);
+  } else {
+    // TN B: k=(r,s,q) with q fastest; element (c, k) = w[q][(r*S+s)*C + c]
+    DgradWTn lb{(const uint16_t *)w, C, Kout, K, R * S * C};
     ConvDgradStage<1> sa{(const uint16_t *)dy, H, W, Kout, HO, WO, S, pad, K, M};
-    return launch_mix_gemm(sa, TnStage<DgradWTn>{lb}, dx, (int)M, C, K, C,
-                           false, strm);
+    e = launch_mix_gemm(sa, TnStage<DgradWTn>{lb}, out, (int)M, C, K, C, f32,
+                        strm, splits);
   }
-  return conv_dgrad_s2(dy, w, dx, N, H, W, C, Kout, R, S, pad, HO, WO, strm);
+  if (e != hipSuccess || splits <= 1) return e;
+  return splitk_reduce(partial, splits, M * C, dx, 1, strm);
 }
 
 // float4 lanes + two split-accumulators: the scalar version issued one
