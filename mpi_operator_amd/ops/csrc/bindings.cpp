@@ -17,6 +17,11 @@ using at::Tensor;
 // blocks on 256 CUs): split the reduce dim so fwd/dgrad fill the chip,
 // paying one fp32 slab + reduce pass.
 int conv_splits(long M, int Ncols, int K) {
+  static const int force = [] {
+    const char *e = getenv("MPIAMD_CONV_SPLITS");
+    return e ? atoi(e) : -1;
+  }();
+  if (force >= 1) return force;
   long tiles = ((M + 127) / 128) * ((Ncols + 127) / 128);
   int nk = (K + 63) / 64;
   if (tiles >= 256 || nk < 16) return 1;

@@ -195,23 +195,34 @@ struct XcolStage {
   }
 };
 
+// Late-stage convs have few 128x128 output tiles (7x7 spatial: ~100 blocks
+// on 256 CUs): when `splits` > 1 the GEMM runs split-K into the caller's
+// fp32 slab and splitk_reduce emits bf16 — same scheme as wgrad.
 extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
                                int H, int W, int C, int Kout, int R, int S,
-                               int stride, int pad, int HO, int WO,
-                               hipStream_t strm) {
+                               int stride, int pad, int HO, int WO, int splits,
+                               float *partial, hipStream_t strm) {
   long M = (long)N * HO * WO;
   int K = R * S * C;
+  int nk = (K + BK - 1) / BK;
+  if (splits > nk) splits = nk > 0 ? nk : 1;
+  void *out = splits > 1 ? (void *)partial : y;
+  bool f32 = splits > 1;
+  hipError_t e;
   if (R == 1 && S == 1 && stride == 1 && pad == 0) {
     // 1x1 conv IS a GEMM on the NHWC image viewed [M][C] — skip the
     // gather arithmetic entirely (≈47% of ResNet bottleneck FLOPs)
     GemmLoader la{(const uint16_t *)x, (int)M, (long)C, C};
     GemmLoader lb{(const uint16_t *)w, Kout, (long)C, C};
-    return launch_nt_gemm(la, lb, y, (int)M, Kout, C, Kout, false, strm);
+    e = launch_nt_gemm(la, lb, out, (int)M, Kout, C, Kout, f32, strm, splits);
+  } else {
+    ConvFwdStage sa{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, K, M};
+    GemmLoader lb{(const uint16_t *)w, Kout, (long)K, K};
+    e = launch_mix_gemm(sa, NtStage<GemmLoader>{lb}, out, (int)M, Kout, K,
+                        Kout, f32, strm, splits);
   }
-  ConvFwdStage sa{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, K, M};
-  GemmLoader lb{(const uint16_t *)w, Kout, (long)K, K};
-  return launch_mix_gemm(sa, NtStage<GemmLoader>{lb}, y, (int)M, Kout, K, Kout,
-                         false, strm);
+  if (e != hipSuccess || splits <= 1) return e;
+  return splitk_reduce(partial, splits, M * Kout, y, 1, strm);
 }
 
 // 1x1 stride-1 dgrad that ACCUMULATES into dx (bottleneck backward: the
