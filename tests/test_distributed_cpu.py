@@ -106,3 +106,72 @@ def test_distributed_optimizer_allreduce():
 @pytest.mark.timeout(180)
 def test_broadcast_and_allreduce():
     _run_workers(_worker_bcast_object, port=PORT + 1)
+
+def _worker_checkpoint(rank, world, port, q):
+    try:
+        _dist_env(rank, world, port)
+        import tempfile
+        from mpi_operator_amd import parallel as hvd
+        from mpi_operator_amd.parallel import checkpoint as ckpt
+
+        hvd.init(backend="gloo")
+        tmpdir = hvd.broadcast_object(
+            tempfile.mkdtemp() if rank == 0 else None)
+        path = os.path.join(tmpdir, "ck.pt")
+
+        torch.manual_seed(5 + rank)  # deliberately DIFFERENT init per rank
+        m = torch.nn.Linear(16, 4)
+        opt = torch.optim.SGD(m.parameters(), lr=0.1, momentum=0.9)
+        m(torch.randn(2, 16)).sum().backward()
+        opt.step()
+
+        # rank-0 state becomes the checkpoint
+        ckpt.save_checkpoint(path, m, opt, epoch=3, step=77)
+        assert os.path.exists(path)  # barrier in save => visible on all ranks
+
+        # fresh model with different weights on every rank; restore must
+        # converge everyone onto rank-0's saved state
+        torch.manual_seed(90 + rank)
+        m2 = torch.nn.Linear(16, 4)
+        opt2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9)
+        extra = ckpt.load_checkpoint(path, m2, opt2)
+        assert extra == {"epoch": 3, "step": 77}
+
+        import torch.distributed as dist
+        flat = torch.cat([p.data.flatten() for p in m2.parameters()])
+        ref = flat.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(flat, ref, atol=0), "params differ across ranks"
+        # momentum buffers restored too
+        n_buf = sum(1 for s in opt2.state.values() if "momentum_buffer" in s)
+        assert n_buf == len(list(m2.parameters()))
+
+        # missing file => fresh start, returns {}
+        assert ckpt.load_checkpoint(os.path.join(tmpdir, "none.pt"), m2) == {}
+        hvd.shutdown()
+        if rank == 0:
+            import shutil
+            shutil.rmtree(tmpdir, ignore_errors=True)
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        import traceback
+        q.put((rank, f"FAIL {e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(180)
+def test_checkpoint_rank0_save_broadcast_restore():
+    _run_workers(_worker_checkpoint, port=PORT + 2)
+
+
+def test_checkpoint_single_process(tmp_path):
+    """World-1 path (no process group): plain save/load roundtrip."""
+    from mpi_operator_amd.parallel import checkpoint as ckpt
+    m = torch.nn.Linear(8, 2)
+    opt = torch.optim.SGD(m.parameters(), lr=0.01)
+    p = str(tmp_path / "ck.pt")
+    ckpt.save_checkpoint(p, m, opt, epoch=1)
+    m2 = torch.nn.Linear(8, 2)
+    extra = ckpt.load_checkpoint(p, m2, torch.optim.SGD(m2.parameters(), lr=0.01))
+    assert extra == {"epoch": 1}
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(a, b)
