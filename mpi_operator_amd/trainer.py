@@ -59,6 +59,17 @@ def train_step(model, dopt, x, y):
     return loss
 
 
+def _all_ranks_ok(local_ok: bool, use_cuda: bool) -> bool:
+    """World-level AND of a local success flag (no-op in a world of 1)."""
+    if hvd.size() <= 1:
+        return local_ok
+    import torch.distributed as dist
+    flag = torch.tensor([1.0 if local_ok else 0.0],
+                        device="cuda" if use_cuda else "cpu")
+    dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+    return flag.item() > 0
+
+
 def timed_steps(model, dopt, data_iter, steps: int, warmup: int, device,
                 graph: bool | None = None) -> dict:
     """Run warmup+steps; barrier+sync bracketed timing of exactly `steps`.
@@ -81,20 +92,27 @@ def timed_steps(model, dopt, data_iter, steps: int, warmup: int, device,
     g = loss_static = None
     if graph and use_cuda:
         x, y = next(it)  # synthetic data: same tensors every step
+        gr = err = None
         try:
             torch.cuda.synchronize()
             gr = torch.cuda.CUDAGraph()
             with torch.cuda.graph(gr):
                 loss_static = train_step(model, dopt, x, y)
+        except Exception as e:  # capture unsupported (e.g. some comm paths)
+            gr, err = None, e
+        # Replays contain the bucket allreduces, so graph-vs-eager must be
+        # a WORLD-level decision: if any rank failed to capture, all ranks
+        # run eager (a split would mismatch collectives and deadlock).
+        if _all_ranks_ok(gr is not None, use_cuda):
             gr.replay()  # one verification replay (counts as warmup)
             torch.cuda.synchronize()
-            assert float(loss_static) == float(loss_static), "NaN after replay"
-            g = gr
-        except Exception as e:  # capture unsupported (e.g. some comm paths)
-            if hvd.rank() == 0:
-                print(f"# hipGraph capture unavailable ({e}); running eager",
-                      flush=True)
-            g = None
+            if _all_ranks_ok(float(loss_static) == float(loss_static), use_cuda):
+                g = gr
+            else:
+                err = err or RuntimeError("non-finite loss after replay")
+        if g is None and hvd.rank() == 0:
+            print(f"# hipGraph capture unavailable ({err}); running eager",
+                  flush=True)
 
     hvd.barrier()
     if use_cuda:
