@@ -148,7 +148,9 @@ class DistributedOptimizer:
         return 1
 
     def synchronize(self):
-        """Wait for all in-flight bucket allreduces; average."""
+        """Wait for all in-flight bucket allreduces; average. Idempotent:
+        the Horovod-shaped ``synchronize(); ...; step()`` pattern must not
+        average twice, so the div is tied to an actual handle wait."""
         w = self._world()
         for b in self.buckets:
             if b.ready != len(b.params) and b.ready > 0:
@@ -157,8 +159,8 @@ class DistributedOptimizer:
             if b.handle is not None:
                 b.handle.wait()
                 b.handle = None
-            if self.average and w > 1:
-                b.flat.div_(w)
+                if self.average and w > 1:
+                    b.flat.div_(w)
             b.ready = 0
 
     # -- optimizer protocol -------------------------------------------------

@@ -54,6 +54,13 @@ def _worker_allreduce(rank, world, port, q):
             avg /= world
             assert torch.allclose(p.grad, avg, atol=1e-6), (p.grad - avg).abs().max()
 
+        # synchronize() must be idempotent: step() re-enters it (the Horovod
+        # synchronize-then-step pattern) and must NOT average a second time
+        snap = [p.grad.clone() for p in m.parameters()]
+        dopt.synchronize()
+        for p, s in zip(m.parameters(), snap):
+            assert torch.equal(p.grad, s), "double-averaged grads"
+
         dopt.step()
         # params identical across ranks after the averaged step
         flat = torch.cat([p.data.flatten() for p in m.parameters()])
