@@ -49,7 +49,7 @@ def _torch_flags():
 def _stale(obj: str, src: str) -> bool:
     if not os.path.exists(obj):
         return True
-    dep = [src] + [os.path.join(CSRC, h) for h in ("common.h", "mfma_tile.h", "mix_gemm.h")]
+    dep = [src] + [os.path.join(CSRC, h) for h in ("common.h", "mfma_tile.h", "mix_gemm.h", "gemm256.h")]
     om = os.path.getmtime(obj)
     return any(os.path.getmtime(d) > om for d in dep if os.path.exists(d))
 

@@ -396,12 +396,28 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
                             s, splits);
 }
 
+#include "gemm256.h"
+
 // Plain NT×NT entry (both operands k-contiguous row-major). glds staging
 // vs register staging selectable for same-box A/B (MPIAMD_GLDS=0 reverts).
+// Full-tile shapes take the 256²-tile counted-vmcnt pipeline (gemm256.h);
+// MPIAMD_GEMM256=0 reverts those to the mix_gemm path.
 template <class LA, class LB>
 static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
                                  int N, int K, long ldc, bool c_f32,
                                  hipStream_t s, int splits = 1) {
+  static const bool use_256 = [] {
+    const char *e = getenv("MPIAMD_GEMM256");
+    return !(e && e[0] == '0');
+  }();
+  // occupancy floor: 256² tiles run 1 block/CU (96 KB LDS), so a grid
+  // under ~half the 256 CUs loses more to idle CUs than the pipeline wins
+  // (measured: M=12544,N=512 → 98 wgs: 338 vs 506 TF on the 128² path;
+  // M=50176,N=256 → 196 wgs: 505 vs 417 — the crossover is between)
+  if (use_256 && splits <= 1 && M % 256 == 0 && N % 256 == 0 && K % 32 == 0 &&
+      K > 0 && la.kdim == K && lb.kdim == K && ldc == N &&
+      (long)(M / 256) * (N / 256) >= 128)
+    return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s);
   static const bool use_glds = [] {
     const char *e = getenv("MPIAMD_GLDS");
     return !(e && e[0] == '0');

@@ -28,6 +28,9 @@ def _gemm_case(M, N, K, c_f32, seed=0):
     (64, 1000, 2048),   # classifier head shape (edge N)
     (200, 72, 136),     # every dim ragged (M,N edge; K%8==0 only)
     (512, 384, 576),    # conv-like K=9*64
+    (512, 256, 64),     # nt256 route: full tiles, 2-deep K pipeline
+    (768, 512, 96),     # nt256 route: odd K-tile count
+    (2048, 1024, 128),  # nt256 route: XCD-swizzled grid (nwg=32)
 ])
 @pytest.mark.parametrize("c_f32", [False, True])
 def test_gemm_nt_matches_matmul(shape, c_f32):
