@@ -22,7 +22,7 @@ manifest:         ## regenerate deploy/v2beta1/mpi-operator.yaml
 	hack/generate-manifest.sh
 
 wheel:            ## build the installable wheel (amdrun image ingredient)
-	$(PYTHON) -m pip wheel --no-deps -w dist .
+	$(PYTHON) -m pip wheel --no-deps --no-build-isolation -w dist .
 
 clean:
 	rm -rf build dist *.egg-info mpi_operator_amd/ops/*.so mpi_operator_amd/ops/csrc/*.o
