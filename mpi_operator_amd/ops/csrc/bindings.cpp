@@ -65,7 +65,8 @@ hipError_t add_relu_bwd(const void *, const void *, void *, long, hipStream_t);
 hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t bias_add(void *, const float *, long, int, hipStream_t);
-hipError_t colsum_bf16(const void *, float *, long, int, hipStream_t);
+hipError_t colsum_bf16(const void *, float *, float *, long, int, hipStream_t);
+int colsum_chunks(long, int);
 struct SgdDesc {
   const void *grad;
   float *master;
@@ -421,9 +422,11 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   Tensor dw = at::empty({N, K}, f32);
   CHK(gemm_tn_tn(dyc.data_ptr(), xc.data_ptr(), dw.data_ptr(), N, K, M, N, K,
                  K, 1, cur_stream()));
-  // zeros, not empty: the fast colsum path accumulates atomically
-  Tensor db = at::zeros({N}, f32);
-  CHK(colsum_bf16(dyc.data_ptr(), db.data_ptr<float>(), M, N, cur_stream()));
+  Tensor db = at::zeros({N}, f32); // zeros: ragged-N fallback writes directly
+  int chunks = colsum_chunks(M, N);
+  Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
+  CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),
+                  db.data_ptr<float>(), M, N, cur_stream()));
   return {dx, dw, db};
 }
 

@@ -264,10 +264,14 @@ extern "C" hipError_t bias_add(void *y, const float *b, long M, int N,
 
 // bias-grad column sum. colsum8_k: each thread owns one 8-column octet
 // (coalesced 16 B loads), blocks tile (row-chunks × col-octet groups) and
-// atomicAdd partials into a zeroed db — the one-thread-per-column serial
-// version ran 16 blocks and was 48% of a BERT-Large step.
+// write per-block partial rows into a [gridDim.x][N] fp32 slab that
+// splitk_reduce folds into db — the atomicAdd ending this replaces put
+// gridDim.x-way RMW contention on every db word (40 µs for a 33 MB read
+// that should take ~6; the atomics were the tail), and the one-thread-per
+// -column serial version before THAT ran 16 blocks and was 48% of a
+// BERT-Large step.
 __global__ void colsum8_k(const ushort8 *__restrict__ dy,
-                          float *__restrict__ db, long M, int C8) {
+                          float *__restrict__ partial, long M, int C8) {
   int cb = blockIdx.y * 32 + (threadIdx.x & 31);
   int rl = threadIdx.x >> 5; // 8 row lanes per block
   float a[8] = {0};
@@ -288,8 +292,9 @@ __global__ void colsum8_k(const ushort8 *__restrict__ dy,
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         a[j] += lds[((r << 5) | (threadIdx.x & 31)) * 8 + j];
+    float *row = partial + (long)blockIdx.x * C8 * 8;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&db[cb * 8 + j], a[j]);
+    for (int j = 0; j < 8; ++j) row[cb * 8 + j] = a[j];
   }
 }
 
@@ -304,20 +309,33 @@ __global__ void colsum_k(const uint16_t *__restrict__ dy, float *__restrict__ db
   }
 }
 
-// db must be ZEROED by the caller for the N%8==0 path (atomic accumulate)
-extern "C" hipError_t colsum_bf16(const void *dy, float *db, long M, int N,
-                                  hipStream_t s) {
+// row-chunk count of the fast path — the caller sizes the fp32 partial
+// slab as [colsum_chunks(M,N)][N]
+extern "C" int colsum_chunks(long M, int N) {
+  if (N % 8 != 0) return 0; // ragged fallback needs no workspace
+  int gy = (N / 8 + 31) / 32;
+  long gx = 1024 / gy;
+  long maxgx = (M + 7) / 8;
+  if (gx > maxgx) gx = maxgx;
+  if (gx < 1) gx = 1;
+  return (int)gx;
+}
+
+extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
+                                    void *out, int out_bf16,
+                                    hipStream_t s); // conv.hip
+
+extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
+                                  long M, int N, hipStream_t s) {
   if (N % 8 == 0) {
     int C8 = N / 8;
     int gy = (C8 + 31) / 32;
-    long gx = 1024 / gy;
-    long maxgx = (M + 7) / 8;
-    if (gx > maxgx) gx = maxgx;
-    if (gx < 1) gx = 1;
-    colsum8_k<<<dim3((int)gx, gy), 256, 0, s>>>((const ushort8 *)dy, db, M, C8);
-  } else {
-    colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N);
+    int gx = colsum_chunks(M, N);
+    colsum8_k<<<dim3(gx, gy), 256, 0, s>>>((const ushort8 *)dy, partial, M, C8);
+    HIP_KERNEL_CHECK();
+    return splitk_reduce(partial, gx, (long)N, db, 0, s);
   }
+  colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
