@@ -205,22 +205,45 @@ __global__ void sgd_step_k(SgdArgs args, float lr, float mu, float wd,
       } else
         for (int j = 0; j < m; ++j) g[j] = gp[j];
     }
-    for (int j = 0; j < m; ++j) {
-      mst[j] = D.master[i + j];
-      mm[j] = D.mom[i + j];
-      float gj = g[j] + wd * mst[j];
-      mm[j] = mu * mm[j] + gj;
-      float step = nesterov ? gj + mu * mm[j] : mm[j];
-      mst[j] -= lr * step;
-      D.master[i + j] = mst[j];
-      D.mom[i + j] = mm[j];
-    }
-    if (D.out) {
-      uint16_t *op = D.out + i;
-      if (full) {
-        *(ushort8 *)op = f8_to_bf8(mst);
-      } else
-        for (int j = 0; j < m; ++j) op[j] = f2bf(mst[j]);
+    // full octets take explicit 16 B vector loads/stores for master and
+    // momentum — the runtime-bound j<m loop alone left them as per-element
+    // dword traffic (measured 1.5 TB/s on BERT-Large's optimizer step)
+    if (full) {
+      float4v ma = *(const float4v *)(D.master + i);
+      float4v mb = *(const float4v *)(D.master + i + 4);
+      float4v va = *(const float4v *)(D.mom + i);
+      float4v vb = *(const float4v *)(D.mom + i + 4);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        mst[j] = j < 4 ? ma[j] : mb[j - 4];
+        mm[j] = j < 4 ? va[j] : vb[j - 4];
+        float gj = g[j] + wd * mst[j];
+        mm[j] = mu * mm[j] + gj;
+        float step = nesterov ? gj + mu * mm[j] : mm[j];
+        mst[j] -= lr * step;
+      }
+      ma = float4v{mst[0], mst[1], mst[2], mst[3]};
+      mb = float4v{mst[4], mst[5], mst[6], mst[7]};
+      va = float4v{mm[0], mm[1], mm[2], mm[3]};
+      vb = float4v{mm[4], mm[5], mm[6], mm[7]};
+      *(float4v *)(D.master + i) = ma;
+      *(float4v *)(D.master + i + 4) = mb;
+      *(float4v *)(D.mom + i) = va;
+      *(float4v *)(D.mom + i + 4) = vb;
+      if (D.out) *(ushort8 *)(D.out + i) = f8_to_bf8(mst);
+    } else {
+      for (int j = 0; j < m; ++j) {
+        mst[j] = D.master[i + j];
+        mm[j] = D.mom[i + j];
+        float gj = g[j] + wd * mst[j];
+        mm[j] = mu * mm[j] + gj;
+        float step = nesterov ? gj + mu * mm[j] : mm[j];
+        mst[j] -= lr * step;
+        D.master[i + j] = mst[j];
+        D.mom[i + j] = mm[j];
+      }
+      if (D.out)
+        for (int j = 0; j < m; ++j) D.out[i + j] = f2bf(mst[j]);
     }
   }
 }
