@@ -100,6 +100,9 @@ hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
+int gemm_tn_tn_splits(int, int, int);
+hipError_t gemm_tn_tn_sk(const void *, const void *, float *, void *, int,
+                         int, int, long, long, long, int, hipStream_t);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, int, float *,
                     hipStream_t);
@@ -418,10 +421,16 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   Tensor dx = at::empty({M, K}, xc.options());
   CHK(gemm_nt_tn(dyp.data_ptr(), wc.data_ptr(), dx.data_ptr(), M, K, N, Np, K,
                  K, 0, cur_stream()));
-  // dw = dy^T @ x: both operands k-strided (k = batch row m) → fp32
+  // dw = dy^T @ x: both operands k-strided (k = batch row m) → fp32.
+  // split-K when the [N][K] tile grid underfills the chip (K_reduce = M).
   Tensor dw = at::empty({N, K}, f32);
-  CHK(gemm_tn_tn(dyc.data_ptr(), xc.data_ptr(), dw.data_ptr(), N, K, M, N, K,
-                 K, 1, cur_stream()));
+  int dw_splits = gemm_tn_tn_splits(N, K, M);
+  Tensor dwp = dw_splits > 1
+                   ? at::empty({dw_splits, (long)N * K}, f32)
+                   : dw;
+  CHK(gemm_tn_tn_sk(dyc.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
+                    dw.data_ptr(), N, K, M, N, K, K, dw_splits,
+                    cur_stream()));
   Tensor db = at::zeros({N}, f32); // zeros: ragged-N fallback writes directly
   int chunks = colsum_chunks(M, N);
   Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;

@@ -4,6 +4,9 @@
 #include "mfma_tile.h"
 #include "mix_gemm.h"
 
+extern "C" hipError_t splitk_reduce(const float *, int, long, void *, int,
+                                    hipStream_t); // conv.hip
+
 extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
                               int N, int K, long lda, long ldb, long ldc,
                               int c_f32, hipStream_t s) {
@@ -31,6 +34,37 @@ extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   return launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb}, c,
                          M, N, K, ldc, c_f32 != 0, s);
+}
+
+// split-K form of gemm_tn_tn for weight-gradient shapes whose tile grid
+// underfills the chip (BERT attn-out dw = 64 workgroups on 256 CUs with a
+// K of batch*seq): fp32 slabs in `partial` [splits][M*ldc] reduced into a
+// dense fp32 C. Caller sizes `partial` with `splits` from gemm_tn_tn_splits.
+extern "C" int gemm_tn_tn_splits(int M, int N, int K) {
+  long tiles = ((M + 127) / 128) * ((N + 127) / 128);
+  int nk = (K + 63) / 64;
+  if (tiles >= 256 || nk < 16) return 1;
+  long s = 512 / tiles;
+  if (s > nk / 8) s = nk / 8;
+  if (s > 64) s = 64;
+  if (s >= 8) s &= ~7; // split-major-capable granularity
+  return s < 1 ? 1 : (int)s;
+}
+
+extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
+                                    float *partial, void *c, int M, int N,
+                                    int K, long lda, long ldb, long ldc,
+                                    int splits, hipStream_t s) {
+  TnRowMajor la{(const uint16_t *)a, lda, K, M};
+  TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
+  if (splits <= 1)
+    return launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
+                           c, M, N, K, ldc, true, s);
+  hipError_t e = launch_mix_gemm(TnStage<TnRowMajor>{la},
+                                 TnStage<TnRowMajor>{lb}, partial, M, N, K,
+                                 ldc, true, s, splits);
+  if (e != hipSuccess) return e;
+  return splitk_reduce(partial, splits, (long)M * ldc, c, 0, s);
 }
 
 // bf16 2-D transpose: out[j][i] = in[i][j], output leading dim ldo >= R

@@ -22,7 +22,9 @@ def test_bert_linear_gpu_matches_fp32():
     assert relerr(y, yr) < 0.02
 
 
-@pytest.mark.parametrize("M,N,K", [(512, 1000, 2048), (96, 24, 40), (256, 3072, 768)])
+@pytest.mark.parametrize("M,N,K", [(512, 1000, 2048), (96, 24, 40), (256, 3072, 768),
+                                   (8192, 256, 512),  # split-K dw (16 slabs)
+                                   (4096, 1024, 1024)])  # attn-out shape (8 slabs)
 def test_linear_bwd_ragged_shapes(M, N, K):
     """dx/dw via the TN-staged kernels on shapes incl. non-%128, non-%8."""
     from mpi_operator_amd.ops import hip_ext
