@@ -108,7 +108,7 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
     }
   }
   // fold this block's per-thread partials into its slab rows via LDS
-  __shared__ float lds[256 * 8];
+  __shared__ float lds[1024 * 8]; // sized for the 16-wave backward block
   float *slab_g = partial + (long)blockIdx.x * 2 * N;
   float *slab_b = slab_g + N;
 #pragma unroll
@@ -175,10 +175,17 @@ extern "C" hipError_t ln_bwd(const void *dy, const void *x, const float *gamma,
                              float *partial, float *dgamma_dbeta, long M,
                              int N, int *grid_out, hipStream_t s) {
   if (N % 8 || N > 2048) return hipErrorInvalidValue;
-  int grid = ln_grid(M, 4);
+  // ≤32 slab rows (16-wave blocks keep the dx pass at ~512 waves): the
+  // dgamma/dbeta splitk_reduce over [grid][2N] runs 2 blocks at this len —
+  // with the old 1024-slab grid it serialized 512-deep per element and was
+  // 23% of a fused-LN BERT step (profiled; fused LN lost to torch because
+  // of THIS kernel, not the LN passes themselves)
+  int grid = ln_grid(M, 16);
+  if (grid > 32) grid = 32;
   if (grid_out) *grid_out = grid;
-  ln_bwd_k<<<grid, 256, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x, gamma,
-                                mean, rstd, (ushort8 *)dx, partial, M, N / 8);
+  ln_bwd_k<<<grid, 1024, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x,
+                                 gamma, mean, rstd, (ushort8 *)dx, partial, M,
+                                 N / 8);
   HIP_KERNEL_CHECK();
   // dgamma_dbeta[0..N) = dgamma, [N..2N) = dbeta
   return splitk_reduce(partial, grid, 2L * N, dgamma_dbeta, 0, s);
