@@ -32,7 +32,8 @@ class LayerNorm(nn.Module):
     """LayerNorm with fp32 statistics. The hand-written HIP kernel
     (ops/csrc/layernorm.hip, numerics-tested) is behind MPIAMD_FUSED_LN=1;
     torch's native LN measured faster end-to-end on BERT-Large (same-box:
-    fused 51.3 ms/step vs native 42.8), so native is the default."""
+    fused 41.4 ms/step vs native 39.2 after the slab-reduce fix — the gap
+    is in ln_bwd's dx pass), so native is the default."""
 
     def __init__(self, n: int, eps: float = 1e-12):
         super().__init__()
