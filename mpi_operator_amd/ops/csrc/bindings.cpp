@@ -103,6 +103,9 @@ hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
 int gemm_tn_tn_splits(int, int, int);
 hipError_t gemm_tn_tn_sk(const void *, const void *, float *, void *, int,
                          int, int, long, long, long, int, hipStream_t);
+hipError_t gemm_nt_tn_sk(const void *, const void *, float *, void *, int,
+                         int, int, long, long, long, int, hipStream_t);
+int gemm_nt_tn_splits(int, int, int);
 hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
                     int, int, int, int, int, int, int, int, float *,
                     hipStream_t);
@@ -419,8 +422,12 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
     dyp.narrow(1, 0, N).copy_(dyc);
   }
   Tensor dx = at::empty({M, K}, xc.options());
-  CHK(gemm_nt_tn(dyp.data_ptr(), wc.data_ptr(), dx.data_ptr(), M, K, N, Np, K,
-                 K, 0, cur_stream()));
+  int dx_splits = gemm_nt_tn_splits(M, K, N);
+  Tensor dxp = dx_splits > 1 ? at::empty({dx_splits, (long)M * K}, f32) : dx;
+  CHK(gemm_nt_tn_sk(dyp.data_ptr(), wc.data_ptr(),
+                    dx_splits > 1 ? dxp.data_ptr<float>() : nullptr,
+                    dx.data_ptr(), M, K, N, Np, K, K, dx_splits,
+                    cur_stream()));
   // dw = dy^T @ x: both operands k-strided (k = batch row m) → fp32.
   // split-K when the [N][K] tile grid underfills the chip (K_reduce = M).
   Tensor dw = at::empty({N, K}, f32);

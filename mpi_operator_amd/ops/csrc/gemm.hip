@@ -25,6 +25,37 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
                          M, N, K, ldc, c_f32 != 0, s);
 }
 
+// split-K form (small-batch dx underfills the chip, e.g. BERT bs=8:
+// dx[1024][4096] = 64 workgroups with K_reduce up to 4096): fp32 slabs in
+// `partial` reduced into bf16 C. Same splits heuristic as the dw path.
+// dx split-K measured NET-NEGATIVE on BERT bs=8 same-box (314/322 vs
+// 321/330 with all linear split-K off, while dw-only splits win at bs=32):
+// the bf16 dx is cheap to produce once, and the fp32 slab pass costs more
+// than the fill gain at these sizes — default OFF, MPIAMD_DX_SK=1 enables.
+extern "C" int gemm_nt_tn_splits(int M, int N, int K) {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_DX_SK");
+    return e && e[0] == '1';
+  }();
+  return on ? gemm_tn_tn_splits(M, N, K) : 1;
+}
+
+extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
+                                    float *partial, void *c, int M, int N,
+                                    int K, long lda, long ldb, long ldc,
+                                    int splits, hipStream_t s) {
+  GemmLoader la{(const uint16_t *)a, M, lda, K};
+  TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
+  if (splits <= 1)
+    return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                           c, M, N, K, ldc, false, s);
+  hipError_t e =
+      launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                      partial, M, N, K, ldc, true, s, splits);
+  if (e != hipSuccess) return e;
+  return splitk_reduce(partial, splits, (long)M * ldc, c, 1, s);
+}
+
 // C[M][N] = Σ_k A(k-strided rows, M cols) · B(k-strided rows, N cols)
 // — linear dw (A = dy [Mbatch][N_out] viewed k-strided, B = x likewise)
 extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
@@ -41,9 +72,16 @@ extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
 // K of batch*seq): fp32 slabs in `partial` [splits][M*ldc] reduced into a
 // dense fp32 C. Caller sizes `partial` with `splits` from gemm_tn_tn_splits.
 extern "C" int gemm_tn_tn_splits(int M, int N, int K) {
+  static const bool off = [] {
+    const char *e = getenv("MPIAMD_LINEAR_SK");
+    return e && e[0] == '0';
+  }();
   long tiles = ((M + 127) / 128) * ((N + 127) / 128);
   int nk = (K + 63) / 64;
-  if (tiles >= 256 || nk < 16) return 1;
+  // nk>=32: at bs=8-class shapes (nk=16) the slab+reduce overhead beats
+  // the fill gain (same-box 312/290 vs 325 seq/s); the bs=32 attn-out dw
+  // (nk=64) is the stable +5% winner this path exists for
+  if (off || tiles >= 256 || nk < 32) return 1;
   long s = 512 / tiles;
   if (s > nk / 8) s = nk / 8;
   if (s > 64) s = 64;
