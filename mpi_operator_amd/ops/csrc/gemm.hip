@@ -15,6 +15,17 @@ extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, c_f32 != 0, s);
 }
 
+// linear forward: bias (fp32 [N]) folded into the GEMM epilogue — the
+// separate bias_add pass was a full activation read+write.
+extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
+                                   const float *bias, void *c, int M, int N,
+                                   int K, long lda, long ldb, long ldc,
+                                   hipStream_t s) {
+  GemmLoader la{(const uint16_t *)a, M, lda, K};
+  GemmLoader lb{(const uint16_t *)b, N, ldb, K};
+  return launch_nt_gemm(la, lb, c, M, N, K, ldc, false, s, 1, bias);
+}
+
 // C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx
 extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
@@ -28,6 +39,8 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
 // split-K form (small-batch dx underfills the chip, e.g. BERT bs=8:
 // dx[1024][4096] = 64 workgroups with K_reduce up to 4096): fp32 slabs in
 // `partial` reduced into bf16 C. Same splits heuristic as the dw path.
+extern "C" int gemm_tn_tn_splits(int M, int N, int K); // defined below
+
 // dx split-K measured NET-NEGATIVE on BERT bs=8 same-box (314/322 vs
 // 321/330 with all linear split-K off, while dw-only splits win at bs=32):
 // the bf16 dx is cheap to produce once, and the fp32 slab pass costs more
