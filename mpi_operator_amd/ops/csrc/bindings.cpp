@@ -96,6 +96,8 @@ hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, const float *, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
                    long, long, int, hipStream_t);
+hipError_t gemm_nt_bias(const void *, const void *, const float *, void *,
+                        int, int, int, long, long, long, hipStream_t);
 hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
@@ -396,13 +398,12 @@ static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   TORCH_CHECK(K % 8 == 0, "linear requires in_features %8==0");
   Tensor y = at::empty({M, N}, xc.options());
-  CHK(gemm_nt(xc.data_ptr(), wc.data_ptr(), y.data_ptr(), M, N, K, K, K, N, 0,
-              cur_stream()));
-  // bias kernel accumulates in fp32; accept bf16 bias (BERT) or fp32
-  // (classifier keeps an fp32 bias for exactness)
+  // bias folded into the GEMM epilogue (fp32; accept bf16 bias (BERT) or
+  // fp32 — the classifier keeps an fp32 bias for exactness)
   Tensor bf = b.scalar_type() == at::kFloat ? b.contiguous()
                                             : b.to(at::kFloat).contiguous();
-  CHK(bias_add(y.data_ptr(), bf.data_ptr<float>(), M, N, cur_stream()));
+  CHK(gemm_nt_bias(xc.data_ptr(), wc.data_ptr(), bf.data_ptr<float>(),
+                   y.data_ptr(), M, N, K, K, K, N, cur_stream()));
   return y;
 }
 

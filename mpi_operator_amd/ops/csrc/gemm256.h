@@ -27,11 +27,11 @@
 // included from mix_gemm.h AFTER GemmLoader/LinearWriter/f2bf/us8_to_bf8v
 // are declared; do not include standalone.
 
-template <bool C_F32>
+template <bool C_F32, bool BIAS>
 __global__ __launch_bounds__(512) void nt256_gemm_k(
     const uint16_t *__restrict__ a, long lda, const uint16_t *__restrict__ b,
     long ldb, void *__restrict__ cptr, int M, int N, int K, long ldc,
-    int tiles_n, int xcd_cpx) {
+    int tiles_n, int xcd_cpx, const float *__restrict__ bias) {
   int tile = blockIdx.x;
   if (xcd_cpx) // T1: contiguous tile chunk per XCD
     tile = (tile & 7) * xcd_cpx + (tile >> 3);
@@ -125,10 +125,11 @@ __global__ __launch_bounds__(512) void nt256_gemm_k(
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
         int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+        float v = BIAS ? acc[mi][ni][r] + bias[col] : acc[mi][ni][r];
         if (C_F32)
-          ((float *)cptr)[base + col] = acc[mi][ni][r];
+          ((float *)cptr)[base + col] = v;
         else
-          ((uint16_t *)cptr)[base + col] = f2bf(acc[mi][ni][r]);
+          ((uint16_t *)cptr)[base + col] = f2bf(v);
       }
     }
   }
@@ -139,15 +140,26 @@ __global__ __launch_bounds__(512) void nt256_gemm_k(
 template <class LA, class LB>
 static hipError_t launch_nt256(const LA &la, const LB &lb, void *c, int M,
                                int N, int K, long ldc, bool c_f32,
-                               hipStream_t s) {
+                               hipStream_t s, const float *bias = nullptr) {
   int tiles_m = M / 256, tiles_n = N / 256;
   int nwg = tiles_m * tiles_n;
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
-  if (c_f32)
-    nt256_gemm_k<true><<<nwg, 512, 0, s>>>(la.p, la.ld, lb.p, lb.ld, c, M, N,
-                                           K, ldc, tiles_n, cpx);
-  else
-    nt256_gemm_k<false><<<nwg, 512, 0, s>>>(la.p, la.ld, lb.p, lb.ld, c, M, N,
-                                            K, ldc, tiles_n, cpx);
+  // compile-time bias split: a runtime `bias ? bias[col] : 0` select in
+  // the epilogue measured 20.3 -> 37.6 us mean on the ResNet 1x1 shapes
+  if (c_f32) {
+    if (bias)
+      nt256_gemm_k<true, true><<<nwg, 512, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+    else
+      nt256_gemm_k<true, false><<<nwg, 512, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+  } else {
+    if (bias)
+      nt256_gemm_k<false, true><<<nwg, 512, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+    else
+      nt256_gemm_k<false, false><<<nwg, 512, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+  }
   return hipGetLastError();
 }

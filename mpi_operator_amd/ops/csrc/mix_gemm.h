@@ -148,6 +148,23 @@ struct LinearWriter {
   }
 };
 
+// linear forward with the bias folded into the epilogue (fp32 bias, the
+// separate bias_add pass was a full activation read+write — 2.4% of a
+// BERT-Large step).
+struct BiasWriter {
+  static constexpr bool ACC = false;
+  long ldc;
+  const float *bias;
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] = v + bias[col];
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    p[b + col] = f2bf(v + bias[col]);
+  }
+};
+
 // += into an existing bf16 tensor (bottleneck backward: conv1's dgrad
 // accumulates onto the skip-connection gradient — no separate add pass).
 struct LinearAccWriter {
@@ -405,7 +422,8 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
 template <class LA, class LB>
 static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
                                  int N, int K, long ldc, bool c_f32,
-                                 hipStream_t s, int splits = 1) {
+                                 hipStream_t s, int splits = 1,
+                                 const float *bias = nullptr) {
   static const bool use_256 = [] {
     const char *e = getenv("MPIAMD_GEMM256");
     return !(e && e[0] == '0');
@@ -417,7 +435,7 @@ static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
   if (use_256 && splits <= 1 && M % 256 == 0 && N % 256 == 0 && K % 32 == 0 &&
       K > 0 && la.kdim == K && lb.kdim == K && ldc == N &&
       (long)(M / 256) * (N / 256) >= 128)
-    return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s);
+    return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
   static const bool use_glds = [] {
     const char *e = getenv("MPIAMD_GLDS");
     return !(e && e[0] == '0');
@@ -425,8 +443,14 @@ static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
   if (use_glds) {
     GldsNt ga{la.p, la.rows, la.ld, la.kdim};
     GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
+    if (bias)
+      return launch_mix_gemm_wr(ga, gb, c, M, N, K, BiasWriter{ldc, bias},
+                                ldc, c_f32, s, splits);
     return launch_mix_gemm(ga, gb, c, M, N, K, ldc, c_f32, s, splits);
   }
+  if (bias)
+    return launch_mix_gemm_wr(NtStage<LA>{la}, NtStage<LB>{lb}, c, M, N, K,
+                              BiasWriter{ldc, bias}, ldc, c_f32, s, splits);
   return launch_mix_gemm(NtStage<LA>{la}, NtStage<LB>{lb}, c, M, N, K, ldc,
                          c_f32, s, splits);
 }
