@@ -28,8 +28,10 @@ BASELINE_TOTAL_IMAGES_PER_SEC = 308.27  # reference README.md:211-213 (2 GPUs)
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    # >=100 graph-replayed steps: a multi-second timed region that coarse
+    # SMI samplers can't miss (the 20-step default produced a 0.38 s window)
+    ap.add_argument("--steps", type=int, default=100)
+    ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--model", default="resnet101", choices=["resnet101", "resnet50"])
     ap.add_argument("--batch", type=int, default=64, help="per-GPU batch")
     ap.add_argument("--image", type=int, default=224)
@@ -45,6 +47,13 @@ def main():
 
     hvd.init()
     use_cuda = torch.cuda.is_available()
+    # --gpus N>1 REQUIRES an N-rank process group (the driver's torchrun
+    # contract). A single process claiming N GPUs would multiply
+    # global_batch by N while computing on one GPU — refuse to inflate.
+    if use_cuda and args.gpus > 1 and hvd.size() != args.gpus:
+        raise SystemExit(
+            f"bench.py --gpus {args.gpus} requires world_size={args.gpus} "
+            f"(got {hvd.size()}); launch via torch.distributed.run")
     n_gpus = hvd.size() if hvd.size() > 1 else args.gpus
 
     if args.impl == "torch":

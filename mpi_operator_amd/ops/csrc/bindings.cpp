@@ -162,6 +162,10 @@ static Tensor conv2d_dgrad(const Tensor &dy, const Tensor &w, int64_t H,
   int N = dy.size(0), Kout = dy.size(1), HO = dy.size(2), WO = dy.size(3);
   int C = w.size(1), R = w.size(2), S = w.size(3);
   TORCH_CHECK(Kout % 8 == 0, "dgrad requires out-channels %8==0");
+  // conv_dgrad routes every stride != 1 to the parity-2 decomposition
+  // (conv.hip conv_dgrad_s2), which is stride-2-only by construction.
+  TORCH_CHECK(stride == 1 || stride == 2,
+              "conv2d_dgrad supports stride 1 or 2 only, got ", stride);
   Tensor dx = empty_cl_bf16(N, C, H, W, dy);
   long M = (long)N * H * W;
   int splits = stride == 1 ? conv_splits(M, C, R * S * Kout) : 1;

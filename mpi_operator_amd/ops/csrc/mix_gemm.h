@@ -414,6 +414,7 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
 }
 
 #include "gemm256.h"
+#include "pipe256.h"
 
 // Plain NT×NT entry (both operands k-contiguous row-major). glds staging
 // vs register staging selectable for same-box A/B (MPIAMD_GLDS=0 reverts).
@@ -432,10 +433,19 @@ static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
   // under ~half the 256 CUs loses more to idle CUs than the pipeline wins
   // (measured: M=12544,N=512 → 98 wgs: 338 vs 506 TF on the 128² path;
   // M=50176,N=256 → 196 wgs: 505 vs 417 — the crossover is between)
+  // 4-phase deep pipeline (pipe256.h) for K%64 full-tile shapes;
+  // MPIAMD_PIPE=0 reverts to the round-1 BK=32 3-buffer kernel (gemm256.h)
+  static const bool use_pipe = [] {
+    const char *e = getenv("MPIAMD_PIPE");
+    return !(e && e[0] == '0');
+  }();
   if (use_256 && splits <= 1 && M % 256 == 0 && N % 256 == 0 && K % 32 == 0 &&
       K > 0 && la.kdim == K && lb.kdim == K && ldc == N &&
-      (long)(M / 256) * (N / 256) >= 128)
+      (long)(M / 256) * (N / 256) >= 128) {
+    if (use_pipe && K % 64 == 0)
+      return launch_pipe256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
     return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
+  }
   static const bool use_glds = [] {
     const char *e = getenv("MPIAMD_GLDS");
     return !(e && e[0] == '0');

@@ -58,12 +58,17 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
         return {}
     if rank() == 0:
         model.load_state_dict(state["model"])
-        if optimizer is not None and state["optimizer"] is not None:
-            optimizer.load_state_dict(state["optimizer"])
     broadcast_parameters(model)
     if optimizer is not None:
-        sd = optimizer.state_dict() if rank() == 0 else None
+        # Broadcast the CPU-resident dict straight from the file (rank 0
+        # loaded it with map_location="cpu").  Broadcasting the live
+        # optimizer's state_dict() instead would pickle rank 0's CUDA
+        # tensors, and every receiving rank would unpickle the full fp32
+        # master + momentum state onto cuda:0 — a multi-GB spike on GPU 0
+        # during an 8-rank restore.  load_state_dict re-homes CPU tensors
+        # to each rank's own device.
+        sd = state["optimizer"] if rank() == 0 else None
         sd = broadcast_object(sd)
-        if rank() != 0 and sd is not None:
+        if sd is not None:
             optimizer.load_state_dict(sd)
     return broadcast_object(state["extra"] if rank() == 0 else None) or {}

@@ -168,13 +168,21 @@ class DistributedOptimizer:
         self.synchronize()
         return self.optimizer.step(closure)
 
-    def zero_grad(self, set_to_none: bool = False):
-        # detach grads so the next backward ASSIGNS them (no accumulate
-        # kernels); the flat is overwritten by _gather, never zeroed
+    def zero_grad(self, set_to_none: bool = True):
+        """Default (set_to_none=True): detach grads so the next backward
+        ASSIGNS them (no accumulate kernels); the flat is overwritten by
+        _gather, never zeroed. set_to_none=False keeps torch.optim
+        semantics for gradient-accumulation callers: grads stay bound to
+        the zeroed bucket views, so the next backward ACCUMULATES."""
         for b in self.buckets:
             b.ready = 0
-            for p in b.params:
-                p.grad = None
+            if set_to_none:
+                for p in b.params:
+                    p.grad = None
+            else:
+                b.flat.zero_()
+                for p, v in zip(b.params, b.views):
+                    p.grad = v
 
     @property
     def param_groups(self):

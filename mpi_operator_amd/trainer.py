@@ -106,7 +106,8 @@ def timed_steps(model, dopt, data_iter, steps: int, warmup: int, device,
         if _all_ranks_ok(gr is not None, use_cuda):
             gr.replay()  # one verification replay (counts as warmup)
             torch.cuda.synchronize()
-            if _all_ranks_ok(float(loss_static) == float(loss_static), use_cuda):
+            lv = float(loss_static.detach())
+            if _all_ranks_ok(lv == lv, use_cuda):
                 g = gr
             else:
                 err = err or RuntimeError("non-finite loss after replay")
@@ -138,5 +139,5 @@ def timed_steps(model, dopt, data_iter, steps: int, warmup: int, device,
         import torch.distributed as dist
         t = t.to("cuda") if use_cuda else t
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-    return {"elapsed": float(t.item()), "loss": float(loss.item()), "steps": steps,
-            "graph": g is not None}
+    return {"elapsed": float(t.item()), "loss": float(loss.detach().item()),
+            "steps": steps, "graph": g is not None}
