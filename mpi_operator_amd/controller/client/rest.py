@@ -109,6 +109,15 @@ class RestResourceClient(ResourceClient):
         out = self._req("GET", _path(self.gvr, namespace), params=params)
         return out.get("items", [])
 
+    def list_with_rv(self, namespace, label_selector=None):
+        """List returning (items, resourceVersion) — the resume point for a
+        watch (client-go ListWatch semantics)."""
+        params = {}
+        if label_selector:
+            params["labelSelector"] = ",".join(f"{k}={v}" for k, v in label_selector.items())
+        out = self._req("GET", _path(self.gvr, namespace), params=params)
+        return out.get("items", []), out.get("metadata", {}).get("resourceVersion")
+
     def create(self, namespace, obj):
         return self._req("POST", _path(self.gvr, namespace), json=obj)
 
@@ -136,8 +145,14 @@ class RestResourceClient(ResourceClient):
 
 
 class RestKubeClient(KubeClient):
-    def __init__(self, config: RestConfig | None = None, user_agent: str = "mpi-operator"):
+    def __init__(self, config: RestConfig | None = None, user_agent: str = "mpi-operator",
+                 qps: float = 5.0, burst: int = 10):
+        """qps/burst: client-side token-bucket throttle, the
+        --kube-api-qps/--kube-api-burst contract (reference options.go:87-88,
+        client-go flowcontrol defaults 5/10)."""
         import requests
+
+        from ..ratelimit import TokenBucket
 
         self.config = config or RestConfig.auto()
         self.session = requests.Session()
@@ -148,8 +163,12 @@ class RestKubeClient(KubeClient):
             self.session.cert = self.config.cert
         self.session.verify = self.config.verify
         self._clients: dict[GVR, RestResourceClient] = {}
+        self._throttle = TokenBucket(qps, burst)
 
     def raw_request(self, method, path, **kw):
+        # watches hold a streaming connection open — don't charge the bucket
+        if not kw.get("stream"):
+            self._throttle.wait()
         resp = self.session.request(method, self.config.host + path, **kw)
         if resp.status_code == 404:
             raise NotFound(path)

@@ -41,7 +41,15 @@ def parse_args(argv=None):
     p.add_argument("--health-port", type=int, default=8080)
     p.add_argument("--gang-scheduling", default="",
                    help="'' = off, 'volcano', or a scheduler-plugins scheduler name")
-    p.add_argument("--lock-namespace", default="default")
+    p.add_argument("--lock-namespace", default="mpi-operator")
+    p.add_argument("--kube-api-qps", type=float, default=5,
+                   help="max QPS to the apiserver from this client")
+    p.add_argument("--kube-api-burst", type=int, default=10,
+                   help="maximum burst for the client throttle")
+    p.add_argument("--controller-queue-rate-limit", type=float, default=10,
+                   help="rate limit of the controller events queue")
+    p.add_argument("--controller-queue-burst", type=int, default=100,
+                   help="maximum burst of the controller events queue")
     p.add_argument("--cluster-domain", default="")
     p.add_argument("--leader-elect", action="store_true", default=True)
     p.add_argument("--no-leader-elect", dest="leader_elect", action="store_false")
@@ -143,7 +151,9 @@ class OperatorServer:
     :1262-1312)."""
 
     def __init__(self, client, controller: MPIJobController, namespace: str = "",
-                 threadiness: int = 2, resync_s: int = 30):
+                 threadiness: int = 2, resync_s: int = 30, rate_limiter=None):
+        from .ratelimit import default_controller_limiter
+
         self.client = client
         self.controller = controller
         self.namespace = namespace
@@ -153,6 +163,10 @@ class OperatorServer:
         self._queued: set = set()
         self._lock = threading.Lock()
         self.stop = threading.Event()
+        # MaxOf{per-key exponential 5ms→1000s, bucket} — a persistently
+        # failing job backs off instead of hammering the apiserver
+        # (reference mpi_job_controller.go:121-124)
+        self.rate_limiter = rate_limiter or default_controller_limiter()
 
     def enqueue(self, namespace: str, name: str):
         with self._lock:
@@ -173,20 +187,45 @@ class OperatorServer:
             return t.namespace(obj), labels[c.JOB_NAME_LABEL]
         return None
 
+    def _handle_event(self, ev: dict, mpijob: bool):
+        obj = ev.get("object", {})
+        if mpijob:
+            self.enqueue(t.namespace(obj), t.name(obj))
+        else:
+            owner = self._owner_job_of(obj)
+            if owner:
+                self.enqueue(*owner)
+        return t.meta(obj).get("resourceVersion")
+
     def _watch_loop(self, gvr, mpijob: bool):
+        """List once, then resume watches from the last seen resourceVersion
+        (client-go ListWatch). Without the resume, every 60 s watch timeout
+        re-LISTed from scratch — O(jobs) LISTs per minute per watcher.
+        A 410 Gone (RV expired) falls back to a fresh list."""
+        from .client.base import ApiError
+
+        rv = None
         while not self.stop.is_set():
             try:
                 rc = self.client.resource(gvr)
-                for ev in rc.watch(self.namespace or None, timeout_s=60):
+                if rv is None:
+                    items, rv = rc.list_with_rv(self.namespace or None)
+                    for obj in items:
+                        self._handle_event({"object": obj}, mpijob)
+                for ev in rc.watch(self.namespace or None, resource_version=rv,
+                                   timeout_s=60):
                     if self.stop.is_set():
                         break
-                    obj = ev.get("object", {})
-                    if mpijob:
-                        self.enqueue(t.namespace(obj), t.name(obj))
-                    else:
-                        owner = self._owner_job_of(obj)
-                        if owner:
-                            self.enqueue(*owner)
+                    if ev.get("type") == "ERROR":  # in-stream 410 Gone
+                        rv = None
+                        break
+                    rv = self._handle_event(ev, mpijob) or rv
+            except ApiError as e:
+                if e.code == 410:
+                    rv = None  # expired RV: re-list
+                else:
+                    log.warning("watch %s failed: %s; retrying", gvr.resource, e)
+                    time.sleep(2)
             except Exception as e:
                 log.warning("watch %s failed: %s; retrying", gvr.resource, e)
                 time.sleep(2)
@@ -210,9 +249,14 @@ class OperatorServer:
                 self._queued.discard((ns, name))
             try:
                 self.controller.sync(ns, name)
+                self.rate_limiter.forget((ns, name))
             except Exception as e:
-                log.warning("sync %s/%s failed: %s; requeueing", ns, name, e)
-                threading.Timer(1.0, self.enqueue, args=(ns, name)).start()
+                delay = self.rate_limiter.when((ns, name))
+                log.warning("sync %s/%s failed: %s; requeueing in %.3fs",
+                            ns, name, e, delay)
+                timer = threading.Timer(delay, self.enqueue, args=(ns, name))
+                timer.daemon = True
+                timer.start()
             finally:
                 self.queue.task_done()
 
@@ -237,7 +281,8 @@ def main(argv=None):
     cfg = RestConfig.from_kubeconfig(opt.kubeconfig) if opt.kubeconfig else RestConfig.auto()
     if opt.master:
         cfg.host = opt.master.rstrip("/")
-    client = RestKubeClient(cfg, user_agent="mpi-operator")
+    client = RestKubeClient(cfg, user_agent="mpi-operator",
+                            qps=opt.kube_api_qps, burst=opt.kube_api_burst)
 
     if not client.crd_exists("mpijobs.kubeflow.org"):
         raise SystemExit("CRD mpijobs.kubeflow.org not found — apply manifests/ first")

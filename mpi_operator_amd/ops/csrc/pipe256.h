@@ -51,7 +51,11 @@ template <int SWZ> DEV_INLINE int p256_swz(int q, int row) {
   return q;
 }
 
-template <bool C_F32, bool BIAS, int SWZ = 1, int NPB = 2>
+// SP: 0 = per-phase setprio(1) around the MFMA cluster (guide T5); 1 =
+// static form — the younger dispatch half (tid>=256) gets priority 1 once
+// before the loop, nothing per-phase (T5 "static form": the condition must
+// be wave-uniform via readfirstlane or s_setprio runs unconditionally).
+template <bool C_F32, bool BIAS, int SWZ = 1, int NPB = 2, int SP = 0>
 __global__ __launch_bounds__(512) void pipe256_gemm_k(
     const uint16_t *__restrict__ a, long lda, const uint16_t *__restrict__ b,
     long ldb, void *__restrict__ cptr, int M, int N, int K, long ldc,
@@ -129,6 +133,9 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
     __builtin_amdgcn_s_barrier();
   }
 
+  if (SP == 1 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
   bf16x8 af[2][2], bf_[2][2];
   for (int t = 0; t < nk; ++t) {
     int buf = t & 1;
@@ -148,7 +155,7 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
       if (pre) stage(t + 1, ph & 1, ph >> 1); // ph0:A-kh0 ph1:B-kh0 ...
       __builtin_amdgcn_s_barrier();
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_setprio(1);
+      if (SP == 0) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -157,10 +164,11 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
           for (int ni = 0; ni < 2; ++ni)
             acc[mh * 2 + mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 af[mi][kk], bf_[ni][kk], acc[mh * 2 + mi][ni], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
+      if (SP == 0) __builtin_amdgcn_s_setprio(0);
       if (NPB == 2) __builtin_amdgcn_s_barrier();
     }
   }
+  if (SP == 1) __builtin_amdgcn_s_setprio(0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // epilogue drain
 
   // 32x32 C/D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)

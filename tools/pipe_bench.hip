@@ -41,7 +41,7 @@ __global__ void tr_probe_k(unsigned int *out, int pattern) {
   else addr = ((t & 15) * 4 + (t >> 4) * 64) * 2;   // guide layout guess
   uint2v v;
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v) : "v"(addr));
+               : "=v"(v) : "v"(addr) : "memory");
   out[t * 2] = v.x;
   out[t * 2 + 1] = v.y;
 }
@@ -78,9 +78,13 @@ static float bf2f_h(uint16_t u) {
   return v.f;
 }
 
-enum Variant { V_GEMM256, V_PIPE_S0, V_PIPE_S1, V_PIPE_S2, V_PIPE_S1_1B, N_VAR };
+enum Variant {
+  V_GEMM256, V_PIPE_S0, V_PIPE_S1, V_PIPE_S2, V_PIPE_S1_1B,
+  V_PIPE_SPRIO, V_PIPE_1B_SPRIO, N_VAR
+};
 static const char *vname[] = {"gemm256(r1)", "pipe swz0", "pipe swz1",
-                              "pipe swz2", "pipe swz1 1bar"};
+                              "pipe swz2", "pipe swz1 1bar",
+                              "pipe sprio", "pipe 1bar+sprio"};
 
 static hipError_t launch_variant(int v, const uint16_t *a, const uint16_t *b,
                                  float *c, int M, int N, int K) {
@@ -107,6 +111,14 @@ static hipError_t launch_variant(int v, const uint16_t *a, const uint16_t *b,
   case V_PIPE_S1_1B:
     pipe256_gemm_k<true, false, 1, 1><<<nwg, 512>>>(a, (long)K, b, (long)K, c,
                                                     M, N, K, N, tiles_n, cpx, 0);
+    break;
+  case V_PIPE_SPRIO:
+    pipe256_gemm_k<true, false, 1, 2, 1><<<nwg, 512>>>(
+        a, (long)K, b, (long)K, c, M, N, K, N, tiles_n, cpx, 0);
+    break;
+  case V_PIPE_1B_SPRIO:
+    pipe256_gemm_k<true, false, 1, 1, 1><<<nwg, 512>>>(
+        a, (long)K, b, (long)K, c, M, N, K, N, tiles_n, cpx, 0);
     break;
   }
   return hipGetLastError();
@@ -203,5 +215,10 @@ int main(int argc, char **argv) {
   if (fails) { printf("NUMERICS FAILURES: %d\n", fails); return 1; }
   bench(4096, 4096, 4096, 25);
   bench(8192, 8192, 8192, 8);
+  // real model shapes on this path: ResNet stage-2/3 1x1 convs (M=NHW),
+  // BERT-Large MLM head (vocab-padded 30720)
+  bench(50176, 512, 256, 25);
+  bench(12544, 1024, 512, 25);
+  bench(4096, 30720, 1024, 12);
   return 0;
 }
