@@ -200,19 +200,23 @@ static hipError_t launch_pipe256(const LA &la, const LB &lb, void *c, int M,
   int tiles_m = M / 256, tiles_n = N / 256;
   int nwg = tiles_m * tiles_n;
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
+  // production config = measured best (tools/pipe_bench, same-box sweep):
+  // swz1 + 1 barrier/phase + static setprio: 1146 TF @4k³ vs 789 for the
+  // round-1 gemm256 (+44%); +17-42% on the real model shapes
+  constexpr int SWZ = 1, NPB = 1, SP = 1;
   if (c_f32) {
     if (bias)
-      pipe256_gemm_k<true, true><<<nwg, 512, 0, s>>>(
+      pipe256_gemm_k<true, true, SWZ, NPB, SP><<<nwg, 512, 0, s>>>(
           la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
     else
-      pipe256_gemm_k<true, false><<<nwg, 512, 0, s>>>(
+      pipe256_gemm_k<true, false, SWZ, NPB, SP><<<nwg, 512, 0, s>>>(
           la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
   } else {
     if (bias)
-      pipe256_gemm_k<false, true><<<nwg, 512, 0, s>>>(
+      pipe256_gemm_k<false, true, SWZ, NPB, SP><<<nwg, 512, 0, s>>>(
           la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
     else
-      pipe256_gemm_k<false, false><<<nwg, 512, 0, s>>>(
+      pipe256_gemm_k<false, false, SWZ, NPB, SP><<<nwg, 512, 0, s>>>(
           la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
   }
   return hipGetLastError();

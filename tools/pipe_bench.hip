@@ -33,12 +33,20 @@ typedef __attribute__((ext_vector_type(2))) unsigned int uint2v;
 __global__ void tr_probe_k(unsigned int *out, int pattern) {
   __shared__ unsigned short l[2048];
   int t = threadIdx.x;
-  for (int i = t; i < 2048; i += 64) l[i] = i;
+  // volatile: the array address only reaches the asm as an integer, so
+  // without this the compiler proves the fill dead and drops the LDS
+  // allocation entirely (observed: group_segment_fixed_size 0)
+  volatile unsigned short *vl = l;
+  for (int i = t; i < 2048; i += 64) vl[i] = i;
   __syncthreads();
-  unsigned int addr;
-  if (pattern == 0) addr = t * 8;                   // lane-linear 8 B
-  else if (pattern == 1) addr = (t & 15) * 8;       // group-constant rows
-  else addr = ((t & 15) * 4 + (t >> 4) * 64) * 2;   // guide layout guess
+  unsigned int off;
+  if (pattern == 0) off = t * 8;                   // lane-linear 8 B
+  else if (pattern == 1) off = (t & 15) * 8;       // group-constant rows
+  else off = ((t & 15) * 4 + (t >> 4) * 64) * 2;   // guide layout guess
+  // address = LDS byte offset of l + per-lane offset (AS3 cast)
+  unsigned int addr =
+      (unsigned int)(unsigned long)(__attribute__((address_space(3)))
+                                    unsigned short *)l + off;
   uint2v v;
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                : "=v"(v) : "v"(addr) : "memory");
