@@ -231,6 +231,12 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
 extern "C" hipError_t conv_dgrad_1x1_acc(const void *dy, const void *w,
                                          void *dx_acc, long M, int C, int Kout,
                                          hipStream_t strm) {
+  if (use_pipemix()) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, (long)Kout, (int)M, Kout}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, (long)C, Kout, C}};
+    return launch_pipe_mix_wr(sa, sb, dx_acc, (int)M, C, Kout,
+                              LinearAccWriter{(long)C}, C, false, strm);
+  }
   GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
   TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
   return launch_mix_gemm_wr(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
@@ -394,11 +400,18 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
   bool f32 = splits > 1;
   hipError_t e;
   if (R == 1 && S == 1 && pad == 0) {
-    // 1x1 dgrad: dx[M][C] = dy[M][Q] · w[Q][C] (w TN-staged, no gather)
-    GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
-    TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
-    e = launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, out,
-                        (int)M, C, Kout, C, f32, strm, splits);
+    // 1x1 dgrad: dx[M][C] = dy[M][Q] · w[Q][C] (w TN via tr_b16 pipeline)
+    if (use_pipemix()) {
+      NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, (long)Kout, (int)M, Kout}};
+      TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, (long)C, Kout, C}};
+      e = launch_pipe_mix_wr(sa, sb, out, (int)M, C, Kout, LinearWriter{(long)C},
+                             C, f32, strm, splits);
+    } else {
+      GemmLoader la{(const uint16_t *)dy, (int)M, (long)Kout, Kout};
+      TnRowMajor lb{(const uint16_t *)w, (long)C, Kout, C};
+      e = launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, out,
+                          (int)M, C, Kout, C, f32, strm, splits);
+    }
   } else {
     // TN B: k=(r,s,q) with q fastest; element (c, k) = w[q][(r*S+s)*C + c]
     DgradWTn lb{(const uint16_t *)w, C, Kout, K, R * S * C};
@@ -468,9 +481,16 @@ extern "C" hipError_t conv_wgrad_implicit(const void *dy, const void *x,
   hipError_t e;
   if (R == 1 && S == 1 && stride == 1 && pad == 0) {
     // 1x1 wgrad: the im2col column of x IS x itself — plain TN view
-    TnRowMajor lb{(const uint16_t *)x, (long)C, (int)M, C};
-    e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
-                        partial, Kout, RSC, (int)M, RSC, true, strm, splits);
+    if (use_pipemix()) {
+      TnPipe<PlainTnSrc> sa{{(const uint16_t *)dy, (long)Kout, (int)M, Kout}};
+      TnPipe<PlainTnSrc> sb{{(const uint16_t *)x, (long)C, (int)M, C}};
+      e = launch_pipe_mix_wr(sa, sb, partial, Kout, RSC, (int)M,
+                             LinearWriter{(long)RSC}, RSC, true, strm, splits);
+    } else {
+      TnRowMajor lb{(const uint16_t *)x, (long)C, (int)M, C};
+      e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
+                          partial, Kout, RSC, (int)M, RSC, true, strm, splits);
+    }
   } else {
     XcolStage sb{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, RSC, M};
     e = launch_mix_gemm(TnStage<TnRowMajor>{la}, sb, partial, Kout, RSC,

@@ -30,6 +30,12 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
 extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
                                  int c_f32, hipStream_t s) {
+  if (use_pipemix()) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
+    return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
+                              c_f32 != 0, s);
+  }
   GemmLoader la{(const uint16_t *)a, M, lda, K};
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   return launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, c,
@@ -57,6 +63,17 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
+  if (use_pipemix()) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
+    if (splits <= 1)
+      return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
+                                false, s);
+    hipError_t e = launch_pipe_mix_wr(sa, sb, partial, M, N, K,
+                                      LinearWriter{ldc}, ldc, true, s, splits);
+    if (e != hipSuccess) return e;
+    return splitk_reduce(partial, splits, (long)M * ldc, c, 1, s);
+  }
   GemmLoader la{(const uint16_t *)a, M, lda, K};
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   if (splits <= 1)
@@ -74,6 +91,12 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
 extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
                                  int c_f32, hipStream_t s) {
+  if (use_pipemix()) {
+    TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
+    return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
+                              c_f32 != 0, s);
+  }
   TnRowMajor la{(const uint16_t *)a, lda, K, M};
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   return launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb}, c,
@@ -106,6 +129,17 @@ extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
+  if (use_pipemix()) {
+    TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
+    if (splits <= 1)
+      return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
+                                true, s);
+    hipError_t e = launch_pipe_mix_wr(sa, sb, partial, M, N, K,
+                                      LinearWriter{ldc}, ldc, true, s, splits);
+    if (e != hipSuccess) return e;
+    return splitk_reduce(partial, splits, (long)M * ldc, c, 0, s);
+  }
   TnRowMajor la{(const uint16_t *)a, lda, K, M};
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   if (splits <= 1)

@@ -415,6 +415,7 @@ static hipError_t launch_mix_gemm(const SA &sa, const SB &sb, void *c, int M,
 
 #include "gemm256.h"
 #include "pipe256.h"
+#include "pipe_mix.h"
 
 // Plain NT×NT entry (both operands k-contiguous row-major). glds staging
 // vs register staging selectable for same-box A/B (MPIAMD_GLDS=0 reverts).
@@ -445,6 +446,17 @@ static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
     if (use_pipe && K % 64 == 0)
       return launch_pipe256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
     return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
+  }
+  // non-full-tile / small-grid shapes: the 128²-tile deep pipeline
+  // (pipe_mix.h) — ragged edges handled by its zeros-page staging
+  if (use_pipemix()) {
+    NtPipe<PlainNtSrc> ga{{la.p, la.ld, la.rows, la.kdim}};
+    NtPipe<PlainNtSrc> gb{{lb.p, lb.ld, lb.rows, lb.kdim}};
+    if (bias)
+      return launch_pipe_mix_wr(ga, gb, c, M, N, K, BiasWriter{ldc, bias},
+                                ldc, c_f32, s, splits);
+    return launch_pipe_mix_wr(ga, gb, c, M, N, K, LinearWriter{ldc}, ldc,
+                              c_f32, s, splits);
   }
   static const bool use_glds = [] {
     const char *e = getenv("MPIAMD_GLDS");

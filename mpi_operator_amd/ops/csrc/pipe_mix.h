@@ -1,0 +1,311 @@
+// Deep-pipelined 128²-tile GEMM with NT *and* TN (k-strided) operand
+// staging — the round-2 replacement for mix_gemm's 2-barrier register-
+// staged tile (measured there: staging loads ~60% of the TN kernel).
+//
+// Everything is glds (HBM→LDS DMA, no staging registers, no ds_write
+// pass), kept in flight across raw barriers by counted vmcnt:
+//
+//   - 256 threads, 4 waves as 2×2; per-wave C = 64×64 as 2×2 frags of
+//     32×32 (mfma_f32_32x32x16_bf16)
+//   - K-tile BK=64 as two 32-deep k-halves; 2 phases per tile (one per
+//     k-half); per phase: [vmcnt(4)] → frag ds_reads → stage one k-half
+//     of A+B for a FUTURE tile (4 glds) → s_barrier → lgkmcnt(0) →
+//     8 MFMA → [optional s_barrier]
+//   - LDS: 2 tile-buffers × 2 halves × 2 operands × 8 KiB = 64 KiB →
+//     2 blocks/CU (8 waves/CU)
+//   - stage schedule (slot freed by the k-half-split read pattern):
+//       t.ph0 stages (t+1, k1);  t.ph1 stages (t+2, k0)
+//     every consumed half was staged 3 phases earlier and is proven by a
+//     vmcnt(4) one full phase (≥1 barrier) before its first read; the
+//     LAST tile's checks drain (vmcnt 0) because the skipped stages would
+//     otherwise leave the final halves uncounted.
+//
+// Operand layouts:
+//   NT (k-contiguous rows): LDS image [128 rows][4 k-octets] lane-linear,
+//     source octet pre-swizzled q ^= (row>>2)&3 (pipe256's measured-best),
+//     consumed by ds_read_b128.
+//   TN (k-strided, memory rows ARE k): LDS image = [4k][16col] subtiles,
+//     cq-fastest (64 subtiles per half), filled by glds of 8 columns at
+//     one k (coalesced 32-B runs that the coalescer merges into full
+//     lines), consumed by ds_read_b64_tr_b16 PAIRS — the gfx950 LDS
+//     transpose read. Probed semantics (tools/pipe_bench --probe): lane l
+//     receives column (l&15) of the 128-B-aligned block at its address,
+//     elems j = block[(l&15) + 16j], j=0..3 — so two tr reads at k-subtiles
+//     (4k apart) assemble the 8-deep k fragment with zero VALU repacking,
+//     and a 32-lane service group touching two ADJACENT blocks covers all
+//     64 banks conflict-free.
+//
+// Sources are SrcMap structs: ptr16(k, col8) → global address of the
+// 16-B granule, or nullptr for out-of-range/padding — the launcher passes
+// a 16-B zeros page those lanes load instead, so EVERY lane issues the
+// same number of VM ops (the counted-vmcnt contract) and ragged M/N/K
+// need no edge kernels.
+//
+// Replaces (reference role): the cuDNN/cuBLAS backward GEMMs of the
+// tf_cnn_benchmarks image (reference README.md:127-130) — conv wgrad /
+// dgrad and linear dw/dx shapes — as MI355X-native code.
+#pragma once
+
+// included from mix_gemm.h after mfma_tile.h; needs LinearWriter etc.
+
+typedef __attribute__((ext_vector_type(2))) unsigned int uint2v_pm;
+
+// MPIAMD_PIPEMIX=0 reverts every pipe_mix route to the round-1 register-
+// staged mix_gemm tile (A/B lever).
+static inline bool use_pipemix() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_PIPEMIX");
+    return !(e && e[0] == '0');
+  }();
+  return on;
+}
+
+// ---- SrcMaps ----------------------------------------------------------
+// concept: const uint16_t *ptr16(int k, int col) — address of the 16-B
+// granule holding cols [col, col+8) at reduce index k, or nullptr.
+
+struct PlainNtSrc { // row-major [rows][kdim]: rows along output dim
+  const uint16_t *p;
+  long ld;
+  int rows, kdim;
+  DEV_INLINE const uint16_t *ptr16(int k, int row) const {
+    // NT: "col" param is the output row; k octet-granular
+    return (row < rows && k < kdim) ? p + (long)row * ld + k : nullptr;
+  }
+};
+
+struct PlainTnSrc { // row-major [kdim][cols]: rows along k
+  const uint16_t *p;
+  long ld;
+  int kdim, cols;
+  DEV_INLINE const uint16_t *ptr16(int k, int col) const {
+    return (k < kdim && col < cols) ? p + (long)k * ld + col : nullptr;
+  }
+};
+
+// ---- stagers ----------------------------------------------------------
+constexpr int PM_BM = 128, PM_BK = 64; // tile 128x128, k-halves of 32
+constexpr int PM_THREADS = 256;
+constexpr int PM_HSZ = 512; // ushort8 slots per (operand, k-half) = 8 KiB
+
+DEV_INLINE int pm_swz(int q, int row) { return q ^ ((row >> 2) & 3); }
+
+// NT k-half: [128 rows][4 octets], lane-linear glds, source-swizzled.
+template <class SRC> struct NtPipe {
+  static constexpr bool TR = false;
+  SRC s;
+  DEV_INLINE void stage(int tid, int kb, int base, ushort8 *img,
+                        const uint16_t *zeros) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * PM_THREADS + tid; // [0,512)
+      int row = idx >> 2;
+      int k = kb + pm_swz(idx & 3, row) * 8;
+      const uint16_t *src = s.ptr16(k, base + row);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void *)(src ? src : zeros),
+          (__attribute__((address_space(3))) void *)(img + idx), 16, 0, 0);
+    }
+  }
+  // frag read: 32 rows starting at frag0, k-step ks (16 deep) of the half
+  DEV_INLINE bf16x8 read(const ushort8 *img, int lane, int frag0, int ks) const {
+    int row = frag0 + (lane & 31);
+    int q = ks * 2 + (lane >> 5);
+    return us8_to_bf8v(img[row * 4 + pm_swz(q, row)]);
+  }
+};
+
+// TN k-half: 64 subtiles of [4k][16col], cq-fastest; tr_b16 consumption.
+template <class SRC> struct TnPipe {
+  static constexpr bool TR = true;
+  SRC s;
+  DEV_INLINE void stage(int tid, int kb, int base, ushort8 *img,
+                        const uint16_t *zeros) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * PM_THREADS + tid;     // [0,512) 16-B slots
+      int st = idx >> 3;                  // subtile: kq*8 + cq
+      int kq = st >> 3, cq = st & 7;
+      int kl = (idx & 7) >> 1, ch = idx & 1;
+      int k = kb + kq * 4 + kl;
+      int col = base + cq * 16 + ch * 8;
+      const uint16_t *src = s.ptr16(k, col);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void *)(src ? src : zeros),
+          (__attribute__((address_space(3))) void *)(img + idx), 16, 0, 0);
+    }
+  }
+  DEV_INLINE bf16x8 read(const ushort8 *img, int lane, int frag0, int ks) const {
+    // two tr reads at k-subtiles (ks*4 + (lane>>5)*2 + {0,1}) deliver
+    // column (lane&15) of block (kq, cq) = k 0..3 / 4..7 of this lane's
+    // operand row — no repacking
+    unsigned base = (unsigned)(unsigned long)(
+        __attribute__((address_space(3))) const void *)img;
+    int kq0 = ks * 4 + ((lane >> 5) & 1) * 2;
+    int cq = (frag0 >> 4) + ((lane >> 4) & 1);
+    unsigned a0 = base + (unsigned)((kq0 * 8 + cq) * 128 + (lane & 15) * 8);
+    uint2v_pm lo, hi;
+    asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                 "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
+                 "s_waitcnt lgkmcnt(0)"
+                 : "=v"(lo), "=v"(hi)
+                 : "v"(a0)
+                 : "memory");
+    union { unsigned u[4]; bf16x8 v; } r;
+    r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
+    return r.v;
+  }
+};
+
+// ---- kernel -----------------------------------------------------------
+template <class SA, class SB, bool C_F32, class WR, int NPB = 1>
+__global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
+    SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
+    int tiles_n, int kt_per_split, long split_stride, int xcd_cpx, int swap,
+    const uint16_t *__restrict__ zeros) {
+  int tile = swap ? blockIdx.y : blockIdx.x;
+  int split = swap ? blockIdx.x : blockIdx.y;
+  if (xcd_cpx) tile = (tile & 7) * xcd_cpx + (tile >> 3);
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * PM_BM, col0 = tn * PM_BM;
+  int tid = threadIdx.x;
+  int lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 1, wc = wave & 1;
+
+  // 128-B aligned: the tr_b16 read resolves its block as addr & ~127
+  __shared__ __align__(128) ushort8 lds[8 * PM_HSZ]; // [buf][op][kh] = 64 KiB
+#define PM_IMG(buf, op, kh) (lds + (((buf) * 2 + (op)) * 2 + (kh)) * PM_HSZ)
+
+  float16v acc[2][2] = {};
+
+  int nk_total = (K + PM_BK - 1) / PM_BK;
+  int t0 = split * kt_per_split;
+  int nk = min(kt_per_split, nk_total - t0);
+  if (nk < 0) nk = 0;
+
+  auto stage_half = [&](int t, int kh) {
+    int kb = (t0 + t) * PM_BK + kh * 32;
+    sa.stage(tid, kb, row0, PM_IMG(t & 1, 0, kh), zeros);
+    sb.stage(tid, kb, col0, PM_IMG(t & 1, 1, kh), zeros);
+  };
+
+  if (nk > 0) {
+    stage_half(0, 0);
+    stage_half(0, 1);
+    if (nk > 1) stage_half(1, 0);
+    // prove (0,k0): allow (0,k1)+(1,k0) in flight
+    if (nk > 1)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      // counted check: proves the half read THIS phase's successor needs.
+      // ph0 proves (t,k1) [staged 3 phases ago]; ph1 proves (t+1,k0).
+      // When the pipeline stops staging (last tiles), drain instead —
+      // vmcnt(4) would leave the final halves unproven.
+      if (t > 0 || kh > 0) { // tile0.ph0 was proven by the prologue wait
+        bool steady = (kh == 0) ? (t + 1 < nk) : (t + 2 < nk);
+        if (steady)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      bf16x8 af[2][2], bf_[2][2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          af[mi][ks] = sa.read(PM_IMG(buf, 0, kh), lane, wr * 64 + mi * 32, ks);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          bf_[ni][ks] = sb.read(PM_IMG(buf, 1, kh), lane, wc * 64 + ni * 32, ks);
+      }
+      if (kh == 0) {
+        if (t + 1 < nk) stage_half(t + 1, 1);
+      } else {
+        if (t + 2 < nk) stage_half(t + 2, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 2; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af[mi][ks], bf_[ni][ks], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (NPB == 2) __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  // 32x32 C/D map (shared with mix_gemm): row-major loop, RowCtx hoisted
+  cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      if (row >= M) continue;
+      typename WR::RowCtx rc = wrt.row_ctx(row);
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+        if (col >= N) continue;
+        if (C_F32)
+          wrt.store_f32((float *)cptr, rc, col, acc[mi][ni][r]);
+        else
+          wrt.store_bf16((uint16_t *)cptr, rc, col, acc[mi][ni][r]);
+      }
+    }
+  }
+}
+#undef PM_IMG
+
+// 16-B zeros page for padding/ragged lanes (per-device, lazily allocated;
+// glds must read SOMETHING for the counted-vmcnt contract to hold).
+inline const uint16_t *pm_zeros_page() {
+  static uint16_t *pages[64] = {};
+  int dev = 0;
+  hipGetDevice(&dev);
+  if (!pages[dev]) {
+    void *p = nullptr;
+    if (hipMalloc(&p, 256) != hipSuccess) return nullptr;
+    hipMemset(p, 0, 256);
+    pages[dev] = (uint16_t *)p;
+  }
+  return pages[dev];
+}
+
+template <class SA, class SB, class WR>
+static hipError_t launch_pipe_mix_wr(const SA &sa, const SB &sb, void *c,
+                                     int M, int N, int K, const WR &wrt,
+                                     long ldc, bool c_f32, hipStream_t s,
+                                     int splits = 1) {
+  const uint16_t *zeros = pm_zeros_page();
+  if (!zeros) return hipErrorOutOfMemory;
+  int tiles_m = (M + PM_BM - 1) / PM_BM, tiles_n = (N + PM_BM - 1) / PM_BM;
+  int nk = (K + PM_BK - 1) / PM_BK;
+  if (splits > nk) splits = nk > 0 ? nk : 1;
+  int kts = (nk + splits - 1) / splits;
+  long split_stride = (long)M * ldc;
+  int nwg = tiles_m * tiles_n;
+  int cpx = (nwg % 8 == 0 && nwg >= 32 && splits == 1) ? nwg / 8 : 0;
+  dim3 grid(nwg, splits);
+  if (c_f32)
+    pipe_mix_k<SA, SB, true, WR><<<grid, PM_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, 0, zeros);
+  else
+    pipe_mix_k<SA, SB, false, WR><<<grid, PM_THREADS, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, 0, zeros);
+  return hipGetLastError();
+}

@@ -56,3 +56,23 @@ def test_bench_under_torchrun_two_ranks():
     out = json.loads(lines[0])
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "dp2"
+
+
+def test_bench_under_torchrun_eight_ranks():
+    """the 8-rank shape of the driver's scaling run,
+    exercised on CPU/gloo so the first 8-GPU execution is not the first
+    8-rank execution of the bucket-overlap + MAX-over-ranks path."""
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29642", os.path.join(REPO, "bench.py"),
+         "--gpus", "8", "--steps", "1", "--warmup", "0",
+         "--model", "resnet50"],
+        capture_output=True, text=True, timeout=900, env=env)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 8
+    assert out["config"]["parallelism"] == "dp8"
