@@ -87,6 +87,13 @@ def main():
         model = model_fn()
     model.train()
 
+    p2p_busbw = None
+    if use_cuda and hvd.size() > 1:
+        # same-node P2P guard: fail loudly if RCCL fell back
+        # to TCP for ranks sharing a node (multi-pod config 4)
+        from mpi_operator_amd.parallel.p2p import verify_p2p
+        p2p_busbw = verify_p2p().get("busbw_gb")
+
     data = SyntheticImageData(batch, image, 1000, device=device, dtype=dtype,
                               channels_last=use_cuda)
     dopt = make_trainer(model, bucket_bytes=args.bucket_mb * 1024 * 1024)
@@ -118,6 +125,7 @@ def main():
                 "impl": args.impl,
                 "graph": res.get("graph", False),
                 "loss": res["loss"],
+                "p2p_busbw_gb": p2p_busbw,
             },
         }
         print(json.dumps(out), flush=True)

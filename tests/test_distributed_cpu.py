@@ -182,3 +182,41 @@ def test_checkpoint_single_process(tmp_path):
     assert extra == {"epoch": 1}
     for a, b in zip(m.parameters(), m2.parameters()):
         assert torch.equal(a, b)
+
+
+# ---- same-node P2P guard (mpi_operator_amd.parallel.p2p) ----
+
+def _worker_p2p_groups(rank, world, port, q):
+    try:
+        _dist_env(rank, world, port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo")
+        from mpi_operator_amd.parallel import p2p
+        groups = p2p.same_node_groups()
+        # both CPU ranks run in one container -> one node group with both
+        assert sorted(groups[0]) == [0, 1], groups
+        # no CUDA here: verify_p2p is a no-op report
+        rep = p2p.verify_p2p()
+        assert rep["checked"] in (False, True)
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"err {e}"))
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_p2p_same_node_groups():
+    _run_workers(_worker_p2p_groups, world=2, port=29741)
+
+
+def test_p2p_static_preconditions(monkeypatch):
+    from mpi_operator_amd.parallel import p2p
+    monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "1")
+    probs = p2p.static_preconditions()
+    assert any("HSA_ENABLE_IPC_MODE_LEGACY" in p for p in probs)
+    monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    monkeypatch.setenv("NCCL_P2P_DISABLE", "1")
+    probs = p2p.static_preconditions()
+    assert any("NCCL_P2P_DISABLE" in p for p in probs)

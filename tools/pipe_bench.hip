@@ -330,9 +330,104 @@ static void bench_tn(int kind, int M, int N, int K, int splits, int iters) {
   hipFree(da); hipFree(db); hipFree(dc);
 }
 
+// ---- TN stage/read isolation debug ------------------------------------
+// Stage ONE k-half of a [K][cols] matrix filled with value k*1000+col via
+// TnPipe::stage, then (a) dump raw LDS halfwords, (b) dump what
+// TnPipe::read delivers per lane — separates a staging-layout bug from a
+// tr-read bug without the pipeline.
+__global__ void tn_debug_k(const uint16_t *src, long ld, int kdim, int cols,
+                           uint16_t *raw, uint16_t *frag,
+                           const uint16_t *zeros, int base) {
+  __shared__ __align__(128) ushort8 lds[PM_HSZ];
+  TnPipe<PlainTnSrc> s{{src, ld, kdim, cols}};
+  int tid = threadIdx.x;
+  s.stage(tid, 0, base, lds, zeros);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  // (a) raw image: 512 slots x 8 halfwords (each thread dumps 2 slots)
+  {
+    ushort8 v = lds[tid];
+    for (int j = 0; j < 8; ++j) raw[tid * 8 + j] = v[j];
+    v = lds[256 + tid];
+    for (int j = 0; j < 8; ++j) raw[(256 + tid) * 8 + j] = v[j];
+  }
+  __syncthreads();
+  // (b) frag reads: waves pretend to be (wr=0..), frag0 0/32/64/96, ks 0/1
+  int lane = tid & 63, wave = tid >> 6;
+  int frag0 = wave * 32; // 4 waves cover frag0 = 0,32,64,96
+  for (int ks = 0; ks < 2; ++ks) {
+    bf16x8 v = s.read(lds, lane, frag0, ks);
+    union { bf16x8 b; uint16_t u[8]; } u;
+    u.b = v;
+    for (int j = 0; j < 8; ++j)
+      frag[((wave * 2 + ks) * 64 + lane) * 8 + j] = u.u[j];
+  }
+}
+
+static void run_tn_debug() {
+  int K = 32, cols = 128;
+  std::vector<uint16_t> h(K * cols);
+  for (int k = 0; k < K; ++k)
+    for (int c = 0; c < cols; ++c) h[k * cols + c] = (uint16_t)(k * 200 + c);
+  uint16_t *dsrc, *draw, *dfrag, *dz;
+  CHECK(hipMalloc(&dsrc, h.size() * 2));
+  CHECK(hipMalloc(&draw, 4096 * 2));
+  CHECK(hipMalloc(&dfrag, 8 * 64 * 8 * 2));
+  CHECK(hipMalloc(&dz, 256));
+  CHECK(hipMemset(dz, 0, 256));
+  CHECK(hipMemcpy(dsrc, h.data(), h.size() * 2, hipMemcpyHostToDevice));
+  tn_debug_k<<<1, PM_THREADS>>>(dsrc, cols, K, cols, draw, dfrag, dz, 0);
+  CHECK(hipDeviceSynchronize());
+  std::vector<uint16_t> raw(4096), frag(8 * 64 * 8);
+  CHECK(hipMemcpy(raw.data(), draw, raw.size() * 2, hipMemcpyDeviceToHost));
+  CHECK(hipMemcpy(frag.data(), dfrag, frag.size() * 2, hipMemcpyDeviceToHost));
+  // expected raw: slot idx -> subtile st=idx>>3 (kq=st>>3,cq=st&7),
+  // kl=(idx&7)>>1, ch=idx&1: halfword j = (kq*4+kl)*200 + cq*16+ch*8+j
+  int bad = 0;
+  for (int idx = 0; idx < 512 && bad < 10; ++idx) {
+    int st = idx >> 3, kq = st >> 3, cq = st & 7;
+    int kl = (idx & 7) >> 1, ch = idx & 1;
+    for (int j = 0; j < 8; ++j) {
+      uint16_t want = (uint16_t)((kq * 4 + kl) * 200 + cq * 16 + ch * 8 + j);
+      if (raw[idx * 8 + j] != want) {
+        printf("RAW MISMATCH idx %d j %d: got %d want %d\n", idx, j,
+               raw[idx * 8 + j], want);
+        ++bad;
+        break;
+      }
+    }
+  }
+  if (!bad) printf("raw LDS image: OK (all 512 slots)\n");
+  // expected frag: lane l of frag0 f, ks: row f+(l&31), k = ks*16+(l>>5)*8+j
+  bad = 0;
+  for (int w = 0; w < 4 && bad < 12; ++w)
+    for (int ks = 0; ks < 2; ++ks)
+      for (int l = 0; l < 64; ++l) {
+        int f0 = w * 32;
+        int col = f0 + (l & 31);
+        for (int j = 0; j < 8; ++j) {
+          int k = ks * 16 + (l >> 5) * 8 + j;
+          uint16_t want = (uint16_t)(k * 200 + col);
+          uint16_t got = frag[((w * 2 + ks) * 64 + l) * 8 + j];
+          if (got != want) {
+            printf("FRAG MISMATCH w%d ks%d lane%d j%d: got %d want %d\n", w,
+                   ks, l, j, got, want);
+            ++bad;
+            break;
+          }
+        }
+      }
+  if (!bad) printf("tr_b16 frag reads: OK (4 frag0 x 2 ks x 64 lanes)\n");
+  hipFree(dsrc); hipFree(draw); hipFree(dfrag); hipFree(dz);
+}
+
 int main(int argc, char **argv) {
   if (argc > 1 && !strcmp(argv[1], "--probe")) {
     run_probe();
+    return 0;
+  }
+  if (argc > 1 && !strcmp(argv[1], "--tn-debug")) {
+    run_tn_debug();
     return 0;
   }
   if (argc > 1 && !strcmp(argv[1], "--tn")) {
