@@ -145,10 +145,13 @@ template <class SRC> struct TnPipe {
     int cq = (frag0 >> 4) + ((lane >> 4) & 1);
     unsigned a0 = base + (unsigned)((kq0 * 8 + cq) * 128 + (lane & 15) * 8);
     uint2v_pm lo, hi;
+    // "=&v" early-clobber is load-bearing: with plain "=v" the allocator
+    // may overlap lo with a0, and the first read clobbers the address the
+    // second read consumes (manifested only under register pressure)
     asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
                  "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
                  "s_waitcnt lgkmcnt(0)"
-                 : "=v"(lo), "=v"(hi)
+                 : "=&v"(lo), "=&v"(hi)
                  : "v"(a0)
                  : "memory");
     union { unsigned u[4]; bf16x8 v; } r;

@@ -295,7 +295,27 @@ static int check_tn(int kind, int M, int N, int K, int splits) {
     bool ok = maxerr < 0.01f * scale + 0.05f;
     printf("tn-numerics kind%d %dx%dx%d sk%d %-8s maxerr %.4g %s\n", kind, M,
            N, K, splits, v ? "pipe_mix" : "mix", maxerr, ok ? "OK" : "FAIL");
-    if (!ok) ++fails;
+    if (!ok) {
+      ++fails;
+      // per-128x128-tile / per-wave-quadrant breakdown localizes the bug:
+      // tile = which block; 64x64 quadrant within tile = which wave
+      for (int tr_ = 0; tr_ < (M + 127) / 128; ++tr_)
+        for (int tc = 0; tc < (N + 127) / 128; ++tc) {
+          float te = 0;
+          int bi = -1, bj = -1;
+          for (int i = tr_ * 128; i < M && i < tr_ * 128 + 128; ++i)
+            for (int j = tc * 128; j < N && j < tc * 128 + 128; ++j) {
+              float e = fabsf(out[(long)i * N + j] - ref[(long)i * N + j]);
+              if (e > te) { te = e; bi = i; bj = j; }
+            }
+          if (te > 0.05f)
+            printf("  tile(%d,%d) maxerr %.3g at (%d,%d) got %.4f want %.4f "
+                   "[quad %d,%d]\n", tr_, tc, te, bi, bj,
+                   bi >= 0 ? out[(long)bi * N + bj] : 0.f,
+                   bi >= 0 ? ref[(long)bi * N + bj] : 0.f,
+                   bi >= 0 ? (bi % 128) / 64 : -1, bj >= 0 ? (bj % 128) / 64 : -1);
+        }
+    }
   }
   hipFree(da); hipFree(db); hipFree(dc);
   return fails;
