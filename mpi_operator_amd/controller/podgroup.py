@@ -36,7 +36,15 @@ def parse_quantity(q) -> Fraction:
 
 def format_quantity(v: Fraction) -> str:
     if v.denominator == 1:
-        return str(v.numerator)
+        n = v.numerator
+        # canonicalize large byte counts back to binary suffixes, matching
+        # k8s resource.Quantity String() ("1Gi"+2×"32Gi" prints "65Gi")
+        if n != 0 and n % 1024 == 0:
+            for suf, mult in (("Ei", 1 << 60), ("Pi", 1 << 50), ("Ti", 1 << 40),
+                              ("Gi", 1 << 30), ("Mi", 1 << 20), ("Ki", 1 << 10)):
+                if n % mult == 0:
+                    return f"{n // mult}{suf}"
+        return str(n)
     milli = v * 1000
     if milli.denominator == 1:
         return f"{milli.numerator}m"

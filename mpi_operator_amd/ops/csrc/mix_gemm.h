@@ -448,8 +448,13 @@ static hipError_t launch_nt_gemm(const LA &la, const LB &lb, void *c, int M,
     return launch_nt256(la, lb, c, M, N, K, ldc, c_f32, s, bias);
   }
   // non-full-tile / small-grid shapes: the 128²-tile deep pipeline
-  // (pipe_mix.h) — ragged edges handled by its zeros-page staging
-  if (use_pipemix()) {
+  // (pipe_mix.h) — ragged edges handled by its zeros-page staging.
+  // MPIAMD_PIPENT=0 reverts just this NT fallback (bisect lever).
+  static const bool use_pipent = [] {
+    const char *e = getenv("MPIAMD_PIPENT");
+    return !(e && e[0] == '0');
+  }();
+  if (use_pipemix() && use_pipent) {
     NtPipe<PlainNtSrc> ga{{la.p, la.ld, la.rows, la.kdim}};
     NtPipe<PlainNtSrc> gb{{lb.p, lb.ld, lb.rows, lb.kdim}};
     if (bias)

@@ -1,5 +1,6 @@
 """API defaulting/validation + podgroup math tests — mirrors the reference's
 default_test.go / validation_test.go / podgroup_test.go matrices."""
+import os
 import pytest
 
 from mpi_operator_amd.controller.api import constants as c
@@ -144,3 +145,105 @@ def test_gang_scheduling_creates_podgroup_and_decorates_pods():
     assert pod["metadata"]["annotations"]["scheduling.k8s.io/group-name"] == "test"
     launcher = client.jobs.get("default", "test-launcher")
     assert launcher["spec"]["template"]["spec"]["schedulerName"] == "volcano"
+
+
+# ---- CRD schema round-trip (hack/gen_crd.py) ----
+
+def _validate_schema(obj, schema, path="$"):
+    """Minimal OpenAPI v3 structural validator (enough for the CRD test:
+    type/enum/required/properties/items/additionalProperties)."""
+    t_ = schema.get("type")
+    errs = []
+    if "anyOf" in schema:
+        if not any(not _validate_schema(obj, s, path) for s in schema["anyOf"]):
+            errs.append(f"{path}: matches no anyOf branch")
+        return errs
+    if t_ == "object":
+        if not isinstance(obj, dict):
+            return [f"{path}: expected object, got {type(obj).__name__}"]
+        for req in schema.get("required", []):
+            if req not in obj:
+                errs.append(f"{path}: missing required {req}")
+        props = schema.get("properties", {})
+        addl = schema.get("additionalProperties")
+        preserve = schema.get("x-kubernetes-preserve-unknown-fields")
+        for k, v in obj.items():
+            if k in props:
+                errs += _validate_schema(v, props[k], f"{path}.{k}")
+            elif isinstance(addl, dict):
+                errs += _validate_schema(v, addl, f"{path}.{k}")
+            elif not preserve and props and addl is None:
+                errs.append(f"{path}: unknown field {k}")
+    elif t_ == "array":
+        if not isinstance(obj, list):
+            return [f"{path}: expected array"]
+        for i, v in enumerate(obj):
+            errs += _validate_schema(v, schema.get("items", {}), f"{path}[{i}]")
+    elif t_ == "string":
+        if not isinstance(obj, str):
+            return [f"{path}: expected string, got {obj!r}"]
+        if "enum" in schema and obj not in schema["enum"]:
+            errs.append(f"{path}: {obj!r} not in {schema['enum']}")
+    elif t_ == "integer":
+        if not isinstance(obj, int) or isinstance(obj, bool):
+            return [f"{path}: expected integer, got {obj!r}"]
+        if "minimum" in schema and obj < schema["minimum"]:
+            errs.append(f"{path}: {obj} < minimum {schema['minimum']}")
+    elif t_ == "boolean":
+        if not isinstance(obj, bool):
+            return [f"{path}: expected boolean, got {obj!r}"]
+    return errs
+
+
+def _crd_schema():
+    import yaml
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    with open(os.path.join(root, "manifests", "base",
+                           "kubeflow.org_mpijobs.yaml")) as f:
+        crd = yaml.safe_load(f)
+    return crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+
+
+def test_crd_generator_in_sync():
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable, os.path.join(root, "hack", "gen_crd.py"),
+                        "--check"], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+
+
+def test_example_yamls_validate_against_crd_schema():
+    import glob
+    import yaml
+    schema = _crd_schema()
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    files = glob.glob(os.path.join(root, "examples", "v2beta1", "**", "*.yaml"),
+                      recursive=True)
+    checked = 0
+    for f in files:
+        with open(f) as fh:
+            for doc in yaml.safe_load_all(fh):
+                if not doc or doc.get("kind") != "MPIJob":
+                    continue
+                errs = _validate_schema(doc, schema)
+                assert not errs, f"{f}: {errs}"
+                checked += 1
+    assert checked >= 3  # pi, resnet, bert examples at minimum
+
+
+def test_crd_schema_rejects_malformed():
+    schema = _crd_schema()
+    bad = {"apiVersion": "kubeflow.org/v2beta1", "kind": "MPIJob",
+           "metadata": {"name": "x"},
+           "spec": {"mpiImplementation": "NotReal",
+                    "slotsPerWorker": 0,
+                    "mpiReplicaSpecs": {"Worker": {
+                        "replicas": "two",
+                        "template": {"spec": {"containers": [
+                            {"image": "x"}]}}}}}}
+    errs = _validate_schema(bad, schema)
+    joined = "\n".join(errs)
+    assert "NotReal" in joined          # bad enum
+    assert "slotsPerWorker: 0" in joined or "minimum" in joined
+    assert "expected integer" in joined  # replicas: "two"
+    assert "missing required name" in joined  # container without name

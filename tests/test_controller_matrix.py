@@ -221,31 +221,40 @@ def _job_with_resources(launcher_req, worker_req, workers=2, sched=None):
     return job
 
 
+def _pg_ctrl(priority_classes=None):
+    from mpi_operator_amd.controller.podgroup import VolcanoCtrl
+    return VolcanoCtrl(FakeKubeClient(), priority_classes=priority_classes)
+
+
+def _defaulted(job):
+    from mpi_operator_amd.controller.api import defaults
+    defaults.set_defaults_mpijob(job)
+    return job
+
+
 def test_pg_min_resources_explicit_wins():
-    from mpi_operator_amd.controller.podgroup import calc_pg_min_resources
-    job = make_job()
+    job = _defaulted(make_job())
     job["spec"]["runPolicy"] = {"schedulingPolicy": {
         "minResources": {"cpu": "7", "memory": "10Gi"}}}
-    got = calc_pg_min_resources(job, 3, lambda name: None)
+    got = _pg_ctrl().calculate_pg_min_resources(3, job)
     assert got == {"cpu": "7", "memory": "10Gi"}
 
 
 def test_pg_min_resources_no_scheduling_policy_sums_all():
-    from mpi_operator_amd.controller.podgroup import calc_pg_min_resources
-    job = _job_with_resources({"cpu": "2", "memory": "1Gi"},
-                              {"cpu": "10", "memory": "32Gi"}, workers=2)
-    got = calc_pg_min_resources(job, 3, lambda name: None)
+    job = _defaulted(_job_with_resources({"cpu": "2", "memory": "1Gi"},
+                                         {"cpu": "10", "memory": "32Gi"},
+                                         workers=2))
+    got = _pg_ctrl().calculate_pg_min_resources(3, job)
     assert got["cpu"] == "22"           # 2 + 2*10
     assert got["memory"] == "65Gi"      # 1 + 2*32
 
 
 def test_pg_min_resources_launcher_only():
-    from mpi_operator_amd.controller.podgroup import calc_pg_min_resources
     job = make_job(workers=2)
     del job["spec"]["mpiReplicaSpecs"]["Worker"]
     job["spec"]["mpiReplicaSpecs"]["Launcher"]["template"]["spec"][
         "containers"][0]["resources"] = {"requests": {"cpu": "2", "memory": "1Gi"}}
-    got = calc_pg_min_resources(job, 1, lambda name: None)
+    got = _pg_ctrl().calculate_pg_min_resources(1, _defaulted(job))
     assert got["cpu"] == "2" and got["memory"] == "1Gi"
 
 
@@ -253,25 +262,25 @@ def test_pg_min_resources_priority_order_counts_first_min_member():
     """With worker priority > launcher priority, minMember=2 counts the two
     workers, not the launcher (reference podgroup_test.go:803-929: replicas
     sorted by PriorityClass value; only the first minMember pods counted)."""
-    from mpi_operator_amd.controller.podgroup import calc_pg_min_resources
     job = _job_with_resources({"cpu": "2", "memory": "1Gi"},
                               {"cpu": "10", "memory": "32Gi"}, workers=2)
     job["spec"]["mpiReplicaSpecs"]["Launcher"]["template"]["spec"][
         "priorityClassName"] = "low"
     job["spec"]["mpiReplicaSpecs"]["Worker"]["template"]["spec"][
         "priorityClassName"] = "high"
-    prio = {"low": 10, "high": 100}
-    got = calc_pg_min_resources(job, 2, lambda name: prio.get(name))
+    got = _pg_ctrl({"low": 10, "high": 100}).calculate_pg_min_resources(
+        2, _defaulted(job))
     assert got["cpu"] == "20"           # two workers only
     assert got["memory"] == "64Gi"
 
 
 def test_pg_min_resources_equal_priority_launcher_first():
-    """Equal priorities: launcher sorts first (reference order), so
-    minMember=2 = launcher + 1 worker."""
-    from mpi_operator_amd.controller.podgroup import calc_pg_min_resources
-    job = _job_with_resources({"cpu": "2", "memory": "1Gi"},
-                              {"cpu": "10", "memory": "32Gi"}, workers=2)
-    got = calc_pg_min_resources(job, 2, lambda name: None)
+    """Equal priorities: the worker replica count is truncated to
+    minMember-1 (reference podgroup.go:362-374), so minMember=2 counts
+    launcher + 1 worker."""
+    job = _defaulted(_job_with_resources({"cpu": "2", "memory": "1Gi"},
+                                         {"cpu": "10", "memory": "32Gi"},
+                                         workers=2))
+    got = _pg_ctrl().calculate_pg_min_resources(2, job)
     assert got["cpu"] == "12"
     assert got["memory"] == "33Gi"
