@@ -30,7 +30,10 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
 extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
                                  int c_f32, hipStream_t s) {
-  if (use_pipemix()) {
+  // pipe TN staging loads 16-B column granules: a cols%8!=0 operand would
+  // read past the allocation on its last k-row (BERT vocab 30522 class) —
+  // those shapes keep the elementwise-tail mix path
+  if (use_pipemix() && N % 8 == 0) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
@@ -63,7 +66,7 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
-  if (use_pipemix()) {
+  if (use_pipemix() && N % 8 == 0) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)
@@ -91,7 +94,7 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
 extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
                                  int c_f32, hipStream_t s) {
-  if (use_pipemix()) {
+  if (use_pipemix() && M % 8 == 0 && N % 8 == 0) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
@@ -129,7 +132,7 @@ extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
-  if (use_pipemix()) {
+  if (use_pipemix() && M % 8 == 0 && N % 8 == 0) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)

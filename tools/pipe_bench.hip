@@ -441,10 +441,118 @@ static void run_tn_debug() {
   hipFree(dsrc); hipFree(draw); hipFree(dfrag); hipFree(dz);
 }
 
+// ---- mix-vs-pipe device cross-check at exact model shapes -------------
+// mix_gemm is the round-1-trusted oracle; both paths accumulate fp32 in
+// identical k order, so outputs should agree to ~bf16 rounding exactly.
+// kind: 0 TN×TN, 1 NT×TN, 2 NT×TN with LinearAccWriter (dgrad-acc)
+static int xcheck(int kind, long M, int N, int K, int splits) {
+  uint16_t *da, *db, *dacc0 = nullptr;
+  float *dc0, *dc1;
+  CHECK(hipMalloc(&da, (long)M * K * 2));
+  CHECK(hipMalloc(&db, (long)N * K * 2));
+  long csz = (long)(splits > 1 ? splits : 1) * M * N;
+  CHECK(hipMalloc(&dc0, csz * 4));
+  CHECK(hipMalloc(&dc1, csz * 4));
+  // deterministic pseudo-random bf16 fill on host (values in ±2)
+  {
+    std::vector<uint16_t> h((long)M * K);
+    unsigned x = 0x1234567u;
+    for (auto &v : h) {
+      x = x * 1664525u + 1013904223u;
+      v = f2bf_h(((x >> 8) & 0xffff) / 32768.0f - 1.0f);
+    }
+    CHECK(hipMemcpy(da, h.data(), h.size() * 2, hipMemcpyHostToDevice));
+    h.resize((long)N * K);
+    for (auto &v : h) {
+      x = x * 1664525u + 1013904223u;
+      v = f2bf_h(((x >> 8) & 0xffff) / 32768.0f - 1.0f);
+    }
+    CHECK(hipMemcpy(db, h.data(), h.size() * 2, hipMemcpyHostToDevice));
+  }
+  auto run_path = [&](int v, float *dc) -> hipError_t {
+    if (kind == 2) { // acc: bf16 += ; dc reused as bf16 buffer
+      CHECK(hipMemset(dc, 0x11, M * N * 2)); // same nonzero init both paths
+      if (v == 0) {
+        GemmLoader la{da, (int)M, (long)K, K};
+        TnRowMajor lb{db, (long)N, K, N};
+        return launch_mix_gemm_wr(NtStage<GemmLoader>{la},
+                                  TnStage<TnRowMajor>{lb}, dc, (int)M, N, K,
+                                  LinearAccWriter{(long)N}, N, false, 0);
+      }
+      NtPipe<PlainNtSrc> sa{{da, (long)K, (int)M, K}};
+      TnPipe<PlainTnSrc> sb{{db, (long)N, K, N}};
+      return launch_pipe_mix_wr(sa, sb, dc, (int)M, N, K,
+                                LinearAccWriter{(long)N}, N, false, 0);
+    }
+    return launch_tn(v, kind, da, db, dc, (int)M, N, K, splits);
+  };
+  int bad = 0;
+  CHECK(run_path(0, dc0));
+  CHECK(hipDeviceSynchronize());
+  CHECK(run_path(1, dc1));
+  CHECK(hipDeviceSynchronize());
+  long n_out = kind == 2 ? (M * N + 1) / 2 : csz; // acc compares bf16 pairs
+  std::vector<float> h0(n_out), h1(n_out);
+  CHECK(hipMemcpy(h0.data(), dc0, n_out * 4, hipMemcpyDeviceToHost));
+  CHECK(hipMemcpy(h1.data(), dc1, n_out * 4, hipMemcpyDeviceToHost));
+  long diffs = 0;
+  float maxd = 0;
+  long first = -1;
+  for (long i = 0; i < n_out; ++i) {
+    float d = fabsf(h0[i] - h1[i]);
+    if (kind == 2) { // compare raw bf16 bit pairs via float-diff of bits
+      if (h0[i] != h1[i]) { ++diffs; if (first < 0) first = i; }
+      continue;
+    }
+    if (d > 1e-4f * (fabsf(h0[i]) + 1.f)) {
+      ++diffs;
+      if (d > maxd) maxd = d;
+      if (first < 0) first = i;
+    }
+  }
+  bad = diffs != 0;
+  printf("xcheck kind%d M%ld N%d K%d sk%d: %s (diffs %ld/%ld maxd %.4g "
+         "first %ld)\n", kind, M, N, K, splits, bad ? "FAIL" : "OK", diffs,
+         n_out, maxd, first);
+  hipFree(da); hipFree(db); hipFree(dc0); hipFree(dc1);
+  if (dacc0) hipFree(dacc0);
+  return bad;
+}
+
+static int conv_splits_h(long M, int Ncols, int K) {
+  long tiles = ((M + 127) / 128) * ((Ncols + 127) / 128);
+  int nk = (K + 63) / 64;
+  if (tiles >= 256 || nk < 16) return 1;
+  long s = 512 / tiles;
+  if (s > nk / 8) s = nk / 8;
+  if (s > 16) s = 16;
+  return s < 1 ? 1 : (int)s;
+}
+
 int main(int argc, char **argv) {
   if (argc > 1 && !strcmp(argv[1], "--probe")) {
     run_probe();
     return 0;
+  }
+  if (argc > 1 && !strcmp(argv[1], "--xcheck")) {
+    int bad = 0;
+    // ResNet101 bs64 1x1 dgrad shapes (M=NHW, N=C, K=Kout) + model splits
+    bad += xcheck(1, 200704, 64, 256, conv_splits_h(200704, 64, 256));
+    bad += xcheck(1, 200704, 256, 64, conv_splits_h(200704, 256, 64));
+    bad += xcheck(1, 50176, 128, 512, conv_splits_h(50176, 128, 512));
+    bad += xcheck(1, 12544, 256, 1024, conv_splits_h(12544, 256, 1024));
+    bad += xcheck(1, 3136, 512, 2048, conv_splits_h(3136, 512, 2048));
+    // dgrad-acc (bottleneck conv1 accumulate)
+    bad += xcheck(2, 50176, 256, 128, 1);
+    bad += xcheck(2, 12544, 512, 256, 1);
+    bad += xcheck(2, 3136, 1024, 512, 1);
+    // 1x1 wgrad (TN×TN over pixel K) + FC backward
+    bad += xcheck(0, 256, 64, 200704, 8);
+    bad += xcheck(0, 512, 2048, 3136, 8);
+    bad += xcheck(0, 1000, 2048, 64, 1);
+    bad += xcheck(1, 64, 2048, 1000, 1);
+    printf(bad ? "XCHECK FAILURES\n" : "xcheck all OK\n");
+    return bad != 0;
   }
   if (argc > 1 && !strcmp(argv[1], "--tn-debug")) {
     run_tn_debug();
