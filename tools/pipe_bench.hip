@@ -534,6 +534,72 @@ int main(int argc, char **argv) {
     run_probe();
     return 0;
   }
+  if (argc > 1 && !strcmp(argv[1], "--xcheck256")) {
+    // pipe256 (new default <swz1,1bar,sprio>) vs gemm256 (round-1 oracle)
+    // at the exact ResNet 1x1-fwd + BERT linear shapes that route there
+    struct S { int M, N, K; } shapes[] = {
+        {200704, 256, 64},  {200704, 64, 256},  {50176, 512, 128},
+        {50176, 256, 512},  {12544, 1024, 256}, {12544, 512, 1024},
+        {3136, 2048, 512},  {4096, 4096, 1024}, {4096, 1024, 4096},
+    };
+    int bad = 0;
+    for (auto &sh : shapes) {
+      long M = sh.M;
+      int N = sh.N, K = sh.K;
+      if (M % 256 || N % 256 || K % 32 || (M / 256) * (N / 256) < 128) {
+        printf("x256 M%ld N%d K%d: skipped (not a 256-route shape)\n", M, N, K);
+        continue;
+      }
+      uint16_t *da, *db;
+      float *d0, *d1;
+      CHECK(hipMalloc(&da, M * K * 2));
+      CHECK(hipMalloc(&db, (long)N * K * 2));
+      CHECK(hipMalloc(&d0, M * N * 4));
+      CHECK(hipMalloc(&d1, M * N * 4));
+      {
+        std::vector<uint16_t> h(M * K);
+        unsigned x = 0xbeef123u;
+        for (auto &v : h) {
+          x = x * 1664525u + 1013904223u;
+          v = f2bf_h(((x >> 8) & 0xffff) / 32768.0f - 1.0f);
+        }
+        CHECK(hipMemcpy(da, h.data(), h.size() * 2, hipMemcpyHostToDevice));
+        h.resize((long)N * K);
+        for (auto &v : h) {
+          x = x * 1664525u + 1013904223u;
+          v = f2bf_h(((x >> 8) & 0xffff) / 32768.0f - 1.0f);
+        }
+        CHECK(hipMemcpy(db, h.data(), h.size() * 2, hipMemcpyHostToDevice));
+      }
+      GemmLoader la{da, (int)M, (long)K, K};
+      GemmLoader lb{db, N, (long)K, K};
+      CHECK(launch_nt256(la, lb, d0, (int)M, N, K, N, true, 0));
+      CHECK(hipDeviceSynchronize());
+      CHECK(launch_pipe256(la, lb, d1, (int)M, N, K, N, true, 0));
+      CHECK(hipDeviceSynchronize());
+      std::vector<float> h0(M * N), h1(M * N);
+      CHECK(hipMemcpy(h0.data(), d0, M * N * 4, hipMemcpyDeviceToHost));
+      CHECK(hipMemcpy(h1.data(), d1, M * N * 4, hipMemcpyDeviceToHost));
+      long diffs = 0, first = -1;
+      float maxd = 0;
+      for (long i = 0; i < M * N; ++i) {
+        float d = fabsf(h0[i] - h1[i]);
+        if (d > 1e-4f * (fabsf(h0[i]) + 1.f)) {
+          ++diffs;
+          if (d > maxd) maxd = d;
+          if (first < 0) first = i;
+        }
+      }
+      printf("x256 M%ld N%d K%d: %s (diffs %ld/%ld maxd %.4g first %ld "
+             "row %ld col %ld)\n", M, N, K, diffs ? "FAIL" : "OK", diffs,
+             M * N, maxd, first, first >= 0 ? first / N : -1,
+             first >= 0 ? first % N : -1);
+      bad += diffs != 0;
+      hipFree(da); hipFree(db); hipFree(d0); hipFree(d1);
+    }
+    printf(bad ? "X256 FAILURES\n" : "x256 all OK\n");
+    return bad != 0;
+  }
   if (argc > 1 && !strcmp(argv[1], "--xcheck")) {
     int bad = 0;
     // ResNet101 bs64 1x1 dgrad shapes (M=NHW, N=C, K=Kout) + model splits
