@@ -92,6 +92,11 @@ hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
                               int, int, int, int, int, int, int, hipStream_t);
 hipError_t softmax_xent_fwd_launch(const void *, const long *, float *, float *,
                                    int, int, hipStream_t);
+hipError_t masked_xent_fwd_launch(const void *, const long *, float *, float *,
+                                  int, int, long, hipStream_t);
+hipError_t masked_xent_bwd_launch(const void *, const long *, const float *,
+                                  const float *, const float *, void *, int,
+                                  int, long, hipStream_t);
 hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, const float *, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
@@ -479,6 +484,40 @@ static Tensor softmax_xent_bwd(const Tensor &probs, const Tensor &target,
   return d;
 }
 
+// MLM masked CE: fwd reads bf16 logits once, saves [B][2] stats + the
+// (loss_sum, valid_count) pair; bwd recomputes probs from the bf16 logits
+// (no fp32 probs materialization — 4× less HBM at vocab scale).
+static std::vector<Tensor> masked_xent_fwd(const Tensor &logits,
+                                           const Tensor &target,
+                                           int64_t ignore_index) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(logits.is_contiguous());
+  const HIPDeviceGuard guard(logits.device());
+  int B = logits.size(0), V = logits.size(1);
+  auto f32 = logits.options().dtype(at::kFloat);
+  Tensor stats = at::empty({B, 2}, f32);
+  Tensor out = at::empty({2}, f32); // loss_sum, valid_count
+  CHK(masked_xent_fwd_launch(logits.data_ptr(), target.data_ptr<long>(),
+                             stats.data_ptr<float>(), out.data_ptr<float>(),
+                             B, V, ignore_index, cur_stream()));
+  return {out, stats};
+}
+
+static Tensor masked_xent_bwd(const Tensor &logits, const Tensor &target,
+                              const Tensor &stats, const Tensor &out,
+                              const Tensor &dloss, int64_t ignore_index) {
+  const HIPDeviceGuard guard(logits.device());
+  int B = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(dloss.is_cuda() && dloss.scalar_type() == at::kFloat &&
+              dloss.numel() == 1, "dloss must be a device fp32 scalar");
+  Tensor d = at::empty({B, V}, logits.options());
+  CHK(masked_xent_bwd_launch(logits.data_ptr(), target.data_ptr<long>(),
+                             stats.data_ptr<float>(), out.data_ptr<float>(),
+                             dloss.data_ptr<float>(), d.data_ptr(), B, V,
+                             ignore_index, cur_stream()));
+  return d;
+}
+
 // ------------------------- add-relu -------------------------
 static Tensor add_relu_fwd_b(const Tensor &a, const Tensor &b) {
   check_cl_bf16(a, "a");
@@ -549,6 +588,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("masked_xent_fwd", &masked_xent_fwd);
+  m.def("masked_xent_bwd", &masked_xent_bwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("add_relu_fwd", &add_relu_fwd_b);
   m.def("add_relu_bwd", &add_relu_bwd_b);

@@ -275,3 +275,46 @@ class AddReLUFn(torch.autograd.Function):
 
 def add_relu(a, b):
     return AddReLUFn.apply(a, b)
+
+
+class MaskedSoftmaxCrossEntropyFn(torch.autograd.Function):
+    """MLM-head CE with ignore_index (mean over valid rows). Forward reads
+    the bf16 logits once and keeps only [B,2] stats; backward recomputes
+    probabilities from the saved bf16 logits — no fp32 logits cast, no fp32
+    probs (the torch path's ~1.5 GB of extra HBM traffic at BERT-Large
+    scale). Reference role: the loss of the out-of-tree BERT images
+    (SURVEY §2.3 N7-class fused op)."""
+
+    @staticmethod
+    def forward(ctx, logits, target, ignore_index):
+        if logits.is_cuda:
+            out, stats = hip_ext().masked_xent_fwd(logits.contiguous(), target,
+                                                   ignore_index)
+            ctx.save_for_backward(logits, target, stats, out)
+            ctx.ignore_index = ignore_index
+            # loss = sum / max(count, 1) — stays on device (graph-safe)
+            return out[0] / out[1].clamp(min=1.0)
+        loss, probs = ref.masked_softmax_cross_entropy_fwd(logits, target,
+                                                           ignore_index)
+        ctx.save_for_backward(probs, target)
+        ctx.cpu_path = True
+        ctx.dtype = logits.dtype
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        if getattr(ctx, "cpu_path", False):
+            probs, target = ctx.saved_tensors
+            d = ref.masked_softmax_cross_entropy_bwd(
+                probs, target, float(dloss), ctx.ignore_index)
+            return d.to(ctx.dtype), None, None
+        logits, target, stats, out = ctx.saved_tensors
+        d = hip_ext().masked_xent_bwd(logits, target, stats, out,
+                                      dloss.reshape(1).float().contiguous(),
+                                      ctx.ignore_index)
+        return d, None, None
+
+
+def masked_softmax_cross_entropy(logits, target, ignore_index=-100):
+    return MaskedSoftmaxCrossEntropyFn.apply(logits, target, ignore_index)

@@ -119,3 +119,30 @@ def sgd_momentum_step(params32, grads32, momenta, bf16_outs, lr, momentum, weigh
         p.add_(step, alpha=-lr)
         if out is not None:
             out.copy_(p.to(out.dtype))
+
+
+def masked_softmax_cross_entropy_fwd(logits, target, ignore_index=-100):
+    """fp32 golden oracle for the MLM masked CE (mean over valid rows)."""
+    x = logits.float()
+    mx = x.max(dim=1, keepdim=True).values
+    e = (x - mx).exp()
+    probs = e / e.sum(dim=1, keepdim=True)
+    valid = target != ignore_index
+    n = int(valid.sum().item())
+    if n == 0:
+        return x.new_zeros(()), probs
+    lse = e.sum(dim=1).log() + mx.squeeze(1)
+    picked = x[valid].gather(1, target[valid].unsqueeze(1)).squeeze(1)
+    loss = (lse[valid] - picked).sum() / n
+    return loss, probs
+
+
+def masked_softmax_cross_entropy_bwd(probs, target, grad_scale: float,
+                                     ignore_index=-100):
+    valid = target != ignore_index
+    n = max(int(valid.sum().item()), 1)
+    d = probs.clone()
+    rows = torch.arange(d.shape[0], device=d.device)
+    d[rows[valid], target[valid]] -= 1.0
+    d[~valid] = 0.0
+    return d * (grad_scale / n)
