@@ -258,10 +258,11 @@ def test_pg_min_resources_launcher_only():
     assert got["cpu"] == "2" and got["memory"] == "1Gi"
 
 
-def test_pg_min_resources_priority_order_counts_first_min_member():
-    """With worker priority > launcher priority, minMember=2 counts the two
-    workers, not the launcher (reference podgroup_test.go:803-929: replicas
-    sorted by PriorityClass value; only the first minMember pods counted)."""
+def test_pg_min_resources_priority_order_truncates_lower_class():
+    """With worker priority > launcher priority and minMember=2, the
+    reference keeps the HIGHER class in full and truncates the lower one to
+    minMember-1 (podgroup.go:358-376: order[1].Replicas = minMember-1), so
+    2 workers + 1 launcher are counted."""
     job = _job_with_resources({"cpu": "2", "memory": "1Gi"},
                               {"cpu": "10", "memory": "32Gi"}, workers=2)
     job["spec"]["mpiReplicaSpecs"]["Launcher"]["template"]["spec"][
@@ -270,8 +271,8 @@ def test_pg_min_resources_priority_order_counts_first_min_member():
         "priorityClassName"] = "high"
     got = _pg_ctrl({"low": 10, "high": 100}).calculate_pg_min_resources(
         2, _defaulted(job))
-    assert got["cpu"] == "20"           # two workers only
-    assert got["memory"] == "64Gi"
+    assert got["cpu"] == "22"           # 2 workers (full) + 1 launcher
+    assert got["memory"] == "65Gi"
 
 
 def test_pg_min_resources_equal_priority_launcher_first():
