@@ -196,11 +196,17 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
     stage_half(0, 0);
     stage_half(0, 1);
     if (nk > 1) stage_half(1, 0);
-    // prove (0,k0): allow (0,k1)+(1,k0) in flight
+    // Prove BOTH halves of tile 0 here (allow only (1,k0) in flight).
+    // (0,k1) is read at t0.ph1, and the ph1-top check runs in the SAME
+    // phase as that read: a per-wave vmcnt without an intervening barrier
+    // cannot collectivize other waves' stages, so a fast wave could read
+    // (0,k1) while a slow wave's prologue glds is still in flight — a
+    // timing race that single-kernel microbenches never hit but a loaded
+    // full-model step does (manifested as rare corruption → divergence).
     if (nk > 1)
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    else
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
 
