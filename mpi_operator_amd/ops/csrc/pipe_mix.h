@@ -61,16 +61,28 @@ static inline bool use_pipemix() {
 }
 
 // ---- SrcMaps ----------------------------------------------------------
-// concept: const uint16_t *ptr16(int k, int col) — address of the 16-B
-// granule holding cols [col, col+8) at reduce index k, or nullptr.
+// concept (stateful: each thread stages the same 2 (row/col, k-offset)
+// granules every k-tile, so the expensive decomposition hoists into init):
+//   void init(int i, int rc, int koff)       — granule i: output-dim index
+//                                              rc, fixed k offset koff
+//   const uint16_t *ptr16(int i, int kb)     — address of granule i at
+//                                              reduce base kb (the 8 elems
+//                                              at k = kb + koff for NT /
+//                                              the 8 cols at k = kb + koff
+//                                              for TN), nullptr = zeros.
 
-struct PlainNtSrc { // row-major [rows][kdim]: rows along output dim
+struct PlainNtSrc { // row-major [rows][kdim]: rows along the output dim
   const uint16_t *p;
   long ld;
   int rows, kdim;
-  DEV_INLINE const uint16_t *ptr16(int k, int row) const {
-    // NT: "col" param is the output row; k octet-granular
-    return (row < rows && k < kdim) ? p + (long)row * ld + k : nullptr;
+  const uint16_t *base_[2];
+  int koff_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    base_[i] = rc < rows ? p + (long)rc * ld + koff : nullptr;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    return (base_[i] && kb + koff_[i] < kdim) ? base_[i] + kb : nullptr;
   }
 };
 
@@ -78,8 +90,131 @@ struct PlainTnSrc { // row-major [kdim][cols]: rows along k
   const uint16_t *p;
   long ld;
   int kdim, cols;
-  DEV_INLINE const uint16_t *ptr16(int k, int col) const {
-    return (k < kdim && col < cols) ? p + (long)k * ld + col : nullptr;
+  const uint16_t *base_[2];
+  int koff_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    base_[i] = rc < cols ? p + (long)koff * ld + rc : nullptr;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    return (base_[i] && kb + koff_[i] < kdim) ? base_[i] + (long)kb * ld
+                                              : nullptr;
+  }
+};
+
+// conv forward x-patch gather (NT: rows are output pixels, k = (r,s,c)).
+// Row decomposition (n, ho·stride-pad, wo·stride-pad) hoisted; per stage
+// only k's (r,s,c) split + 2 bounds checks remain.
+struct ConvFwdSrc {
+  const uint16_t *x;
+  int H, W, C, HO, WO, S, stride, pad, K;
+  long M;
+  long nbase_[2];
+  int hb_[2], wb_[2], koff_[2];
+  bool ok_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    long m = rc;
+    ok_[i] = m < M;
+    int wo = (int)(m % WO);
+    long t = m / WO;
+    int ho = (int)(t % HO);
+    int n = (int)(t / HO);
+    nbase_[i] = (long)n * H * W;
+    hb_[i] = ho * stride - pad;
+    wb_[i] = wo * stride - pad;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    int k = kb + koff_[i];
+    int c = k % C, rs = k / C;
+    int s_ = rs % S, rr = rs / S;
+    int h = hb_[i] + rr, w = wb_[i] + s_;
+    if (k < K && ok_[i] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+      return x + (nbase_[i] + (long)h * W + w) * C + c;
+    return nullptr;
+  }
+};
+
+// conv dgrad dy-gather (NT: rows are input pixels, k = (r,s,q)); STRIDE
+// a template param so the inner %/÷ are shifts.
+template <int STRIDE> struct ConvDgradSrc {
+  const uint16_t *dy;
+  int H, W, Q, HO, WO, S, pad, K;
+  long M;
+  long nbase_[2];
+  int hb_[2], wb_[2], koff_[2];
+  bool ok_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    long m = rc;
+    ok_[i] = m < M;
+    int w_ = (int)(m % W);
+    long t = m / W;
+    int h_ = (int)(t % H);
+    int n = (int)(t / H);
+    nbase_[i] = (long)n * HO * WO;
+    hb_[i] = h_ + pad;
+    wb_[i] = w_ + pad;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    int k = kb + koff_[i];
+    int q = k % Q, rs = k / Q;
+    int s_ = rs % S, rr = rs / S;
+    int hn = hb_[i] - rr, wn = wb_[i] - s_;
+    int ho = hn / STRIDE, wo = wn / STRIDE;
+    if (k < K && ok_[i] && hn >= 0 && wn >= 0 &&
+        (STRIDE == 1 || (hn % STRIDE == 0 && wn % STRIDE == 0)) && ho < HO &&
+        wo < WO)
+      return dy + (nbase_[i] + (long)ho * WO + wo) * Q + q;
+    return nullptr;
+  }
+};
+
+// dgrad weight view (TN: k = (r,s,q) with q fastest; cols are input
+// channels): element (c, k) = w[q][(r·S+s)·C + c].
+struct DgradWTnSrc {
+  const uint16_t *w;
+  int C, Q, K, RSC;
+  int col_[2], koff_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    col_[i] = rc;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    int k = kb + koff_[i];
+    if (k >= K || col_[i] >= C) return nullptr;
+    int q = k % Q, rs = k / Q;
+    return w + (long)q * RSC + rs * C + col_[i];
+  }
+};
+
+// wgrad implicit-im2col gather (TN: k = output pixel m, cols = (r,s,c)).
+// Column decomposition (rr, ss, c) hoisted; per stage one m decomposition.
+struct XcolSrc {
+  const uint16_t *x;
+  int H, W, C, HO, WO, S, stride, pad, RSC;
+  long M;
+  int rr_[2], ss_[2], coff_[2], koff_[2];
+  bool cok_[2];
+  DEV_INLINE void init(int i, int rc, int koff) {
+    cok_[i] = rc < RSC;
+    int c = rc % C, rs = rc / C;
+    ss_[i] = rs % S;
+    rr_[i] = rs / S;
+    coff_[i] = c;
+    koff_[i] = koff;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    long m = kb + koff_[i];
+    if (m >= M || !cok_[i]) return nullptr;
+    int wo = (int)(m % WO);
+    long t = m / WO;
+    int ho = (int)(t % HO);
+    int n = (int)(t / HO);
+    int h = ho * stride + rr_[i] - pad, w = wo * stride + ss_[i] - pad;
+    if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+      return x + ((long)(n * H + h) * W + w) * C + coff_[i];
+    return nullptr;
   }
 };
 
@@ -193,20 +328,23 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
   };
 
   if (nk > 0) {
+    // Prove BOTH halves of tile 0 at this barrier. (0,k1) is read at
+    // t0.ph1, and the ph1-top check runs in the SAME phase as that read:
+    // a per-wave vmcnt without an intervening barrier cannot collectivize
+    // other waves' stages, so a fast wave could read (0,k1) while a slow
+    // wave's prologue glds was still in flight — a timing race that
+    // single-kernel microbenches never hit but a loaded full-model step
+    // did (rare corruption → diverging loss). Staging ALL of tile 1 here
+    // keeps 8 glds in flight across the wait (prologue latency hiding).
     stage_half(0, 0);
     stage_half(0, 1);
-    if (nk > 1) stage_half(1, 0);
-    // Prove BOTH halves of tile 0 here (allow only (1,k0) in flight).
-    // (0,k1) is read at t0.ph1, and the ph1-top check runs in the SAME
-    // phase as that read: a per-wave vmcnt without an intervening barrier
-    // cannot collectivize other waves' stages, so a fast wave could read
-    // (0,k1) while a slow wave's prologue glds is still in flight — a
-    // timing race that single-kernel microbenches never hit but a loaded
-    // full-model step does (manifested as rare corruption → divergence).
-    if (nk > 1)
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    else
+    if (nk > 1) {
+      stage_half(1, 0);
+      stage_half(1, 1); // t0.ph0 skips its (1,k1) stage in exchange
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
   }
 
@@ -236,7 +374,8 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
           bf_[ni][ks] = sb.read(PM_IMG(buf, 1, kh), lane, wc * 64 + ni * 32, ks);
       }
       if (kh == 0) {
-        if (t + 1 < nk) stage_half(t + 1, 1);
+        // t==0: (1,k1) was already staged by the prologue
+        if (t > 0 && t + 1 < nk) stage_half(t + 1, 1);
       } else {
         if (t + 2 < nk) stage_half(t + 2, 0);
       }
