@@ -226,17 +226,25 @@ constexpr int PM_HSZ = 512; // ushort8 slots per (operand, k-half) = 8 KiB
 DEV_INLINE int pm_swz(int q, int row) { return q ^ ((row >> 2) & 3); }
 
 // NT k-half: [128 rows][4 octets], lane-linear glds, source-swizzled.
+// init() binds each thread's 2 fixed (row, k-octet-offset) granules into
+// the SrcMap so per-stage address work is the SrcMap's ptr16 only.
 template <class SRC> struct NtPipe {
   static constexpr bool TR = false;
   SRC s;
-  DEV_INLINE void stage(int tid, int kb, int base, ushort8 *img,
-                        const uint16_t *zeros) const {
+  DEV_INLINE void init(int tid, int base) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       int idx = i * PM_THREADS + tid; // [0,512)
       int row = idx >> 2;
-      int k = kb + pm_swz(idx & 3, row) * 8;
-      const uint16_t *src = s.ptr16(k, base + row);
+      s.init(i, base + row, pm_swz(idx & 3, row) * 8);
+    }
+  }
+  DEV_INLINE void stage(int tid, int kb, ushort8 *img,
+                        const uint16_t *zeros) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * PM_THREADS + tid;
+      const uint16_t *src = s.ptr16(i, kb);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void *)(src ? src : zeros),
           (__attribute__((address_space(3))) void *)(img + idx), 16, 0, 0);
@@ -254,17 +262,22 @@ template <class SRC> struct NtPipe {
 template <class SRC> struct TnPipe {
   static constexpr bool TR = true;
   SRC s;
-  DEV_INLINE void stage(int tid, int kb, int base, ushort8 *img,
-                        const uint16_t *zeros) const {
+  DEV_INLINE void init(int tid, int base) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       int idx = i * PM_THREADS + tid;     // [0,512) 16-B slots
       int st = idx >> 3;                  // subtile: kq*8 + cq
       int kq = st >> 3, cq = st & 7;
       int kl = (idx & 7) >> 1, ch = idx & 1;
-      int k = kb + kq * 4 + kl;
-      int col = base + cq * 16 + ch * 8;
-      const uint16_t *src = s.ptr16(k, col);
+      s.init(i, base + cq * 16 + ch * 8, kq * 4 + kl);
+    }
+  }
+  DEV_INLINE void stage(int tid, int kb, ushort8 *img,
+                        const uint16_t *zeros) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * PM_THREADS + tid;
+      const uint16_t *src = s.ptr16(i, kb);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void *)(src ? src : zeros),
           (__attribute__((address_space(3))) void *)(img + idx), 16, 0, 0);
@@ -321,10 +334,12 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
   int nk = min(kt_per_split, nk_total - t0);
   if (nk < 0) nk = 0;
 
+  sa.init(tid, row0);
+  sb.init(tid, col0);
   auto stage_half = [&](int t, int kh) {
     int kb = (t0 + t) * PM_BK + kh * 32;
-    sa.stage(tid, kb, row0, PM_IMG(t & 1, 0, kh), zeros);
-    sb.stage(tid, kb, col0, PM_IMG(t & 1, 1, kh), zeros);
+    sa.stage(tid, kb, PM_IMG(t & 1, 0, kh), zeros);
+    sb.stage(tid, kb, PM_IMG(t & 1, 1, kh), zeros);
   };
 
   if (nk > 0) {

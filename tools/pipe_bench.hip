@@ -361,7 +361,8 @@ __global__ void tn_debug_k(const uint16_t *src, long ld, int kdim, int cols,
   __shared__ __align__(128) ushort8 lds[PM_HSZ];
   TnPipe<PlainTnSrc> s{{src, ld, kdim, cols}};
   int tid = threadIdx.x;
-  s.stage(tid, 0, base, lds, zeros);
+  s.init(tid, base);
+  s.stage(tid, 0, lds, zeros);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   // (a) raw image: 512 slots x 8 halfwords (each thread dumps 2 slots)
