@@ -123,7 +123,15 @@ class BertLayer(nn.Module):
 
     def forward(self, x, attn_mask=None):
         x = self.ln1(x + self.attn(x, attn_mask))
-        y = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            # fused FFN: GELU lives in the GEMM epilogues (fwd emits
+            # h_pre + gelu(h); bwd's fc2-dx multiplies by gelu'(h_pre))
+            b, s, h = x.shape
+            y = Fx.ffn(x.reshape(-1, h).contiguous(),
+                       self.fc1.weight, self.fc1.bias,
+                       self.fc2.weight, self.fc2.bias).view(b, s, h)
+        else:
+            y = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
         return self.ln2(x + y)
 
 

@@ -105,6 +105,12 @@ hipError_t gemm_nt_bias(const void *, const void *, const float *, void *,
                         int, int, int, long, long, long, hipStream_t);
 hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
+hipError_t gemm_nt_gelu_bias(const void *, const void *, const float *,
+                             void *, void *, int, int, int, long, long, long,
+                             hipStream_t);
+hipError_t gemm_nt_tn_gelubwd(const void *, const void *, const void *,
+                              void *, int, int, int, long, long, long,
+                              hipStream_t);
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 int gemm_tn_tn_splits(int, int, int);
@@ -456,6 +462,60 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   return {dx, dw, db};
 }
 
+// dw/db only (FFN backward computes fc2's dx separately with the fused
+// dgelu epilogue — re-running the full linear_bwd would pay that GEMM twice)
+static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
+                                             const Tensor &x) {
+  const HIPDeviceGuard guard(x.device());
+  Tensor dyc = dy.contiguous(), xc = x.contiguous();
+  int M = xc.size(0), K = xc.size(1), N = dyc.size(1);
+  auto f32 = x.options().dtype(at::kFloat);
+  Tensor dw = at::empty({N, K}, f32);
+  int dw_splits = gemm_tn_tn_splits(N, K, M);
+  Tensor dwp = dw_splits > 1 ? at::empty({dw_splits, (long)N * K}, f32) : dw;
+  CHK(gemm_tn_tn_sk(dyc.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
+                    dw.data_ptr(), N, K, M, N, K, K, dw_splits,
+                    cur_stream()));
+  Tensor db = at::zeros({N}, f32);
+  int chunks = colsum_chunks(M, N);
+  Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
+  CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),
+                  db.data_ptr<float>(), M, N, cur_stream()));
+  return {dw, db};
+}
+
+// ------------------------- fused FFN (GELU) -------------------------
+// fc1 forward with the bias+GELU in the epilogue: returns (g, h_pre)
+static std::vector<Tensor> linear_gelu_fwd(const Tensor &x, const Tensor &w,
+                                           const Tensor &b) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.dim() == 2);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  const HIPDeviceGuard guard(x.device());
+  int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K);
+  Tensor bias = b.to(at::kFloat).contiguous();
+  Tensor g = at::empty({M, N}, x.options());
+  Tensor pre = at::empty({M, N}, x.options());
+  CHK(gemm_nt_gelu_bias(x.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
+                        pre.data_ptr(), g.data_ptr(), M, N, K, K, K, N,
+                        cur_stream()));
+  return {g, pre};
+}
+
+// fc2-dx with dgelu fused: dh = (dy·w2) ⊙ gelu'(h_pre)
+static Tensor linear_gelu_dgrad(const Tensor &dy, const Tensor &w,
+                                const Tensor &pre) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dy.is_contiguous() && w.is_contiguous() && pre.is_contiguous());
+  const HIPDeviceGuard guard(dy.device());
+  int M = dy.size(0), K = dy.size(1), N = w.size(1);
+  TORCH_CHECK(pre.size(0) == M && pre.size(1) == N);
+  Tensor dh = at::empty({M, N}, dy.options());
+  CHK(gemm_nt_tn_gelubwd(dy.data_ptr(), w.data_ptr(), pre.data_ptr(),
+                         dh.data_ptr(), M, N, K, K, N, N, cur_stream()));
+  return dh;
+}
+
 // ------------------------- softmax xent -------------------------
 static std::vector<Tensor> softmax_xent_fwd(const Tensor &logits,
                                             const Tensor &target) {
@@ -589,6 +649,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_bwd", &linear_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("masked_xent_fwd", &masked_xent_fwd);
+  m.def("linear_gelu_fwd", &linear_gelu_fwd);
+  m.def("linear_wgrad_only", &linear_wgrad_only);
+  m.def("linear_gelu_dgrad", &linear_gelu_dgrad);
   m.def("masked_xent_bwd", &masked_xent_bwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("add_relu_fwd", &add_relu_fwd_b);

@@ -26,6 +26,46 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, false, s, 1, bias);
 }
 
+// FFN fc1 fused forward: g = gelu(x·w1ᵀ + b), pre-activation h saved for
+// backward (GeluBiasWriter). Both operands NT k-contiguous.
+extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
+                                        const float *bias, void *pre, void *g,
+                                        int M, int N, int K, long lda,
+                                        long ldb, long ldc, hipStream_t s) {
+  GeluBiasWriter wrt{ldc, bias, (uint16_t *)pre};
+  GemmLoader la{(const uint16_t *)x, M, lda, K};
+  GemmLoader lb{(const uint16_t *)w, N, ldb, K};
+  if (M % 256 == 0 && N % 256 == 0 && K % 64 == 0 && ldc == N &&
+      (long)(M / 256) * (N / 256) >= 128)
+    return launch_pipe256_wr(la, lb, g, M, N, K, ldc, false, wrt, s);
+  if (use_pipemix()) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)x, lda, M, K}};
+    NtPipe<PlainNtSrc> sb{{(const uint16_t *)w, ldb, N, K}};
+    return launch_pipe_mix_wr(sa, sb, g, M, N, K, wrt, ldc, false, s);
+  }
+  GldsNt ga{la.p, la.rows, la.ld, la.kdim};
+  GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
+  return launch_mix_gemm_wr(ga, gb, g, M, N, K, wrt, ldc, false, s);
+}
+
+// FFN backward: dh = (dy·w2) ⊙ gelu'(h)  — the fc2-dx GEMM with the
+// dgelu applied in the epilogue (GeluBwdWriter); A NT, B (w2) TN.
+extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
+                                         const void *pre, void *dh, int M,
+                                         int N, int K, long lda, long ldb,
+                                         long ldc, hipStream_t s) {
+  GeluBwdWriter wrt{ldc, (const uint16_t *)pre};
+  if (use_pipemix() && N % 8 == 0) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, lda, M, K}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, ldb, K, N}};
+    return launch_pipe_mix_wr(sa, sb, dh, M, N, K, wrt, ldc, false, s);
+  }
+  GemmLoader la{(const uint16_t *)dy, M, lda, K};
+  TnRowMajor lb{(const uint16_t *)w, ldb, K, N};
+  return launch_mix_gemm_wr(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                            dh, M, N, K, wrt, ldc, false, s);
+}
+
 // C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx
 extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,

@@ -55,11 +55,13 @@ template <int SWZ> DEV_INLINE int p256_swz(int q, int row) {
 // static form — the younger dispatch half (tid>=256) gets priority 1 once
 // before the loop, nothing per-phase (T5 "static form": the condition must
 // be wave-uniform via readfirstlane or s_setprio runs unconditionally).
-template <bool C_F32, bool BIAS, int SWZ = 1, int NPB = 2, int SP = 0>
+template <bool C_F32, bool BIAS, int SWZ = 1, int NPB = 2, int SP = 0,
+          class WR = LinearWriter, bool USE_WR = false>
 __global__ __launch_bounds__(512) void pipe256_gemm_k(
     const uint16_t *__restrict__ a, long lda, const uint16_t *__restrict__ b,
     long ldb, void *__restrict__ cptr, int M, int N, int K, long ldc,
-    int tiles_n, int xcd_cpx, const float *__restrict__ bias) {
+    int tiles_n, int xcd_cpx, const float *__restrict__ bias,
+    WR wrt = WR{}) {
   int tile = blockIdx.x;
   if (xcd_cpx) tile = (tile & 7) * xcd_cpx + (tile >> 3);
   int tm = tile / tiles_n, tn = tile % tiles_n;
@@ -178,6 +180,18 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
     for (int r = 0; r < 16; ++r) {
       long row = row0 + wr * 128 + mi * 32 + (r & 3) + 8 * (r >> 2) +
                  4 * (lane >> 5);
+      if (USE_WR) {
+        typename WR::RowCtx rc = wrt.row_ctx((int)row);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+          if (C_F32)
+            wrt.store_f32((float *)cptr, rc, col, acc[mi][ni][r]);
+          else
+            wrt.store_bf16((uint16_t *)cptr, rc, col, acc[mi][ni][r]);
+        }
+        continue;
+      }
       long base = row * ldc;
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
@@ -192,6 +206,24 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
   }
 }
 #undef P256_IMG
+
+// Writer-parameterized launch (fused epilogues: GELU+bias, BN stats, ...)
+template <class WR, class LA, class LB>
+static hipError_t launch_pipe256_wr(const LA &la, const LB &lb, void *c,
+                                    int M, int N, int K, long ldc, bool c_f32,
+                                    const WR &wrt, hipStream_t s) {
+  int tiles_m = M / 256, tiles_n = N / 256;
+  int nwg = tiles_m * tiles_n;
+  int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
+  constexpr int SWZ = 1, NPB = 1, SP = 1;
+  if (c_f32)
+    pipe256_gemm_k<true, false, SWZ, NPB, SP, WR, true><<<nwg, 512, 0, s>>>(
+        la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, nullptr, wrt);
+  else
+    pipe256_gemm_k<false, false, SWZ, NPB, SP, WR, true><<<nwg, 512, 0, s>>>(
+        la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, nullptr, wrt);
+  return hipGetLastError();
+}
 
 template <class LA, class LB>
 static hipError_t launch_pipe256(const LA &la, const LB &lb, void *c, int M,

@@ -146,3 +146,12 @@ def masked_softmax_cross_entropy_bwd(probs, target, grad_scale: float,
     d[rows[valid], target[valid]] -= 1.0
     d[~valid] = 0.0
     return d * (grad_scale / n)
+
+
+def dgelu(x):
+    """tanh-approx GELU derivative (fp32) — oracle for the fused FFN bwd."""
+    xf = x.float()
+    c = 0.7978845608028654
+    u = c * (xf + 0.044715 * xf ** 3)
+    t = torch.tanh(u)
+    return 0.5 * (1 + t) + 0.5 * xf * (1 - t * t) * c * (1 + 3 * 0.044715 * xf * xf)

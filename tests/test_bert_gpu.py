@@ -152,3 +152,32 @@ def test_bert_mlm_head_on_hip_path_matches_torch():
     num = (g_hip.float() - g_t.float()).abs().max().item()
     den = g_t.float().abs().max().item() + 1e-6
     assert num < 0.1 * den + 1e-4, (num, den)
+
+
+def test_ffn_fused_gelu_matches_fp32():
+    """Fused FFN (GELU in GEMM epilogues) vs a plain fp32 torch FFN:
+    forward output and all five gradients."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(7)
+    M, H, I = 512, 256, 1024
+    x = (torch.randn(M, H, device="cuda") * 0.5).to(torch.bfloat16)
+    w1 = (torch.randn(I, H, device="cuda") * 0.05).to(torch.bfloat16)
+    b1 = torch.randn(I, device="cuda").to(torch.bfloat16) * 0.1
+    w2 = (torch.randn(H, I, device="cuda") * 0.05).to(torch.bfloat16)
+    b2 = torch.randn(H, device="cuda").to(torch.bfloat16) * 0.1
+    args = [t.clone().requires_grad_(True) for t in (x, w1, b1, w2, b2)]
+    y = Fx.ffn(*args)
+    y.float().square().mean().backward()
+
+    ref = [t.float().clone().requires_grad_(True) for t in (x, w1, b1, w2, b2)]
+    h = torch.nn.functional.linear(ref[0], ref[1], ref[2])
+    g = torch.nn.functional.gelu(h, approximate="tanh")
+    yr = torch.nn.functional.linear(g, ref[3], ref[4])
+    yr.square().mean().backward()
+
+    assert torch.allclose(y.float(), yr, rtol=0.05, atol=0.05), \
+        (y.float() - yr).abs().max()
+    for a, r, name in zip(args, ref, "x w1 b1 w2 b2".split()):
+        num = (a.grad.float() - r.grad).abs().max().item()
+        den = r.grad.abs().max().item() + 1e-6
+        assert num < 0.08 * den + 1e-4, (name, num, den)
