@@ -22,10 +22,33 @@ MI355X-first design decisions:
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.distributed as dist
 
 _XGMI_BUCKET_BYTES = 28 * 1024 * 1024
+
+# MPIAMD_GRAD_VIEWS=1: keep p.grad bound to the bucket views across steps so
+# backward ACCUMULATES directly into the flat (no gather copy) — the A/B
+# lever for the 8-GPU overlap tax. Default stays the
+# assign-then-gather scheme: one fused multi-tensor copy per bucket measured
+# faster than per-param accumulate kernels on 1 GPU.
+_GRAD_VIEWS = os.environ.get("MPIAMD_GRAD_VIEWS") == "1"
+
+
+def _adasum_combine(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Adasum of two gradient vectors (Horovod's adaptive summation): scale
+    each operand down by its projection on the other so correlated updates
+    are not double-counted (hvd.Adasum, reference tensorflow_mnist.py:127)."""
+    af, bf = a.float(), b.float()
+    dot = torch.dot(af.flatten(), bf.flatten())
+    na = torch.dot(af.flatten(), af.flatten())
+    nb = torch.dot(bf.flatten(), bf.flatten())
+    one = torch.ones((), device=af.device)
+    ca = torch.where(na > 0, 1.0 - dot / (2.0 * na), one)
+    cb = torch.where(nb > 0, 1.0 - dot / (2.0 * nb), one)
+    return (ca * af + cb * bf).to(a.dtype)
 
 
 class _Bucket:
@@ -55,7 +78,15 @@ class DistributedOptimizer:
     """
 
     def __init__(self, optimizer, named_parameters=None, bucket_bytes: int = _XGMI_BUCKET_BYTES,
-                 average: bool = True, process_group=None):
+                 average: bool = True, process_group=None, op: str = "average"):
+        """op: "average" (default), "sum", or "adasum" (Horovod's adaptive
+        summation via recursive doubling over P2P sendrecv; power-of-two
+        world sizes — reference hvd.Adasum)."""
+        if op not in ("average", "sum", "adasum"):
+            raise ValueError(f"unknown reduction op {op!r}")
+        self.op = op
+        if op == "sum":
+            average = False
         self.optimizer = optimizer
         self.average = average
         self.group = process_group
@@ -140,7 +171,30 @@ class DistributedOptimizer:
     def _launch(self, bucket: _Bucket):
         self._gather(bucket)
         if dist.is_available() and dist.is_initialized():
-            bucket.handle = dist.all_reduce(bucket.flat, async_op=True, group=self.group)
+            if self.op == "adasum":
+                # synchronous recursive doubling (log2(w) pairwise
+                # exchanges); Adasum is order-sensitive so it does not ride
+                # the async comm stream
+                self._adasum_bucket(bucket)
+            else:
+                bucket.handle = dist.all_reduce(bucket.flat, async_op=True,
+                                                group=self.group)
+
+    def _adasum_bucket(self, bucket: _Bucket):
+        w = dist.get_world_size(self.group)
+        if w & (w - 1):
+            raise RuntimeError("adasum requires a power-of-two world size")
+        rank = dist.get_rank(self.group)
+        buf = torch.empty_like(bucket.flat)
+        span = 1
+        while span < w:
+            partner = rank ^ span
+            ops = [dist.P2POp(dist.isend, bucket.flat, partner, group=self.group),
+                   dist.P2POp(dist.irecv, buf, partner, group=self.group)]
+            for req in dist.batch_isend_irecv(ops):
+                req.wait()
+            bucket.flat.copy_(_adasum_combine(bucket.flat, buf))
+            span <<= 1
 
     def _world(self):
         if dist.is_available() and dist.is_initialized():
@@ -176,7 +230,7 @@ class DistributedOptimizer:
         the zeroed bucket views, so the next backward ACCUMULATES."""
         for b in self.buckets:
             b.ready = 0
-            if set_to_none:
+            if set_to_none and not _GRAD_VIEWS:
                 for p in b.params:
                     p.grad = None
             else:

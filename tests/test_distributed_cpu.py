@@ -220,3 +220,50 @@ def test_p2p_static_preconditions(monkeypatch):
     monkeypatch.setenv("NCCL_P2P_DISABLE", "1")
     probs = p2p.static_preconditions()
     assert any("NCCL_P2P_DISABLE" in p for p in probs)
+
+
+# ---- Adasum reduction (Horovod's hvd.Adasum parity) ----
+
+def test_adasum_combine_orthogonal_sums():
+    """Orthogonal gradients add exactly; identical gradients average."""
+    import torch
+    from mpi_operator_amd.parallel.distributed_optimizer import _adasum_combine
+    a = torch.tensor([1.0, 0.0])
+    b = torch.tensor([0.0, 2.0])
+    assert torch.allclose(_adasum_combine(a, b), torch.tensor([1.0, 2.0]))
+    c = torch.tensor([3.0, 4.0])
+    assert torch.allclose(_adasum_combine(c, c.clone()), c)  # avg of equals
+
+
+def _worker_adasum(rank, world, port, q):
+    try:
+        _dist_env(rank, world, port)
+        import torch
+        import torch.distributed as dist
+        dist.init_process_group("gloo")
+        from mpi_operator_amd import parallel as hvd
+        from mpi_operator_amd.parallel import DistributedOptimizer
+        torch.manual_seed(100 + rank)
+        m = torch.nn.Linear(8, 4)
+        opt = DistributedOptimizer(torch.optim.SGD(m.parameters(), lr=0.1),
+                                   op="adasum")
+        x = torch.randn(4, 8)
+        m(x).sum().backward()
+        # fire + reduce
+        opt.synchronize()
+        # result must be identical on all ranks (adasum is symmetric)
+        flat = torch.cat([p.grad.flatten() for p in m.parameters()])
+        gathered = [torch.empty_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        assert torch.allclose(gathered[0], gathered[1], atol=1e-6)
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"err {e}"))
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_adasum_two_rank_symmetry():
+    _run_workers(_worker_adasum, world=2, port=29745)

@@ -46,6 +46,10 @@ def main():
 
     for _ in range(args.warmup):
         loss = step()
+    # drop the warmup autograd graph before capture: a live loss keeps
+    # AccumulateGrad nodes from the eager stream alive, and capture_end
+    # segfaults on the stream mismatch
+    loss = None
     g = None
     if args.graph:
         try:
