@@ -1,0 +1,238 @@
+// Fused multi-head self-attention forward for BERT (bf16, head_dim 64):
+//   ctx[b, s, h·64+d] = softmax(Q·Kᵀ / √64 + mask) · V
+//
+// Replaces the hipBLASLt batched-GEMM + torch-softmax chain (reference
+// role: the fused attention of the out-of-tree BERT images, SURVEY §2.3
+// N7). MI355X-first structure:
+//   - one workgroup per (batch, head): B·H blocks (BERT-Large bs32: 512 —
+//     2 blocks/CU), 256 threads = 4 waves; wave w owns Q rows w·32..+31
+//   - K staged in LDS [S][64] with the (row>>2)-XOR octet swizzle and read
+//     as b128 MFMA B-fragments (k = head_dim, NT)
+//   - V staged in LDS as [4 s_k][16 d] subtiles and consumed with
+//     ds_read_b64_tr_b16 — the hardware transpose read (probed semantics:
+//     lane receives column (l&15) of its 128-B block), so the PV operand
+//     needs no repacking
+//   - scores never leave registers: full-row softmax (fp32 max/exp/sum via
+//     4-wide frag reduce + half-wave shfl) — S ≤ 512 keeps the whole row
+//     resident, no online-softmax rescaling needed
+//   - P crosses to the PV MFMA via one per-wave LDS image (the C-layout →
+//     A-fragment transpose)
+//   - probs optionally written out (bf16) for the torch-side backward
+//
+// Layouts: qkv is the fused projection output [B, S, 3, H, 64] (the model's
+// qkv view before any transpose — the kernel does the head split itself);
+// ctx is [B, S, H·64] so the consumer (attn.out linear) needs no reshape.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8a;
+typedef __attribute__((ext_vector_type(4))) float float4a;
+typedef __attribute__((ext_vector_type(16))) float float16a;
+typedef __attribute__((ext_vector_type(2))) unsigned int uint2a;
+
+DEV_INLINE bf16x8a us8_to_bf8a(ushort8 u) {
+  union { ushort8 u; bf16x8a b; } v;
+  v.u = u;
+  return v.b;
+}
+
+DEV_INLINE int att_swz(int q, int row) { return q ^ ((row >> 2) & 7); }
+
+// S_MAX: compile-time sequence capacity (LDS sizing); launched per actual S.
+template <int S_MAX>
+__global__ __launch_bounds__(256) void attn_fwd_k(
+    const uint16_t *__restrict__ qkv, // [B, S, 3, H, 64]
+    uint16_t *__restrict__ ctx,       // [B, S, H*64]
+    uint16_t *__restrict__ probs,     // [B, H, S, S] or nullptr
+    const float *__restrict__ mask,   // [B, S] additive (or nullptr)
+    int B, int S, int H, float scale) {
+  constexpr int D = 64;
+  int bh = blockIdx.x;
+  int b = bh / H, h = bh % H;
+  int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+
+  // LDS: K [S_MAX][8 slots] (swizzled octets), V subtiles [S_MAX/4][4][16],
+  // P per-wave [32][S_MAX]
+  __shared__ __align__(128) ushort8 k_img[S_MAX * 8];
+  __shared__ __align__(128) ushort8 v_img[S_MAX * 8];
+  __shared__ __align__(128) uint16_t p_img[4][32][S_MAX];
+
+  const long qkv_row = (long)3 * H * D; // elements per (b, s)
+  const uint16_t *base = qkv + (long)b * S * qkv_row;
+  const uint16_t *kp = base + (long)1 * H * D + h * D;
+  const uint16_t *vp = base + (long)2 * H * D + h * D;
+
+  // stage K/V padded to a 32-row multiple with ZEROS — un-staged LDS
+  // garbage can be Inf/NaN bf16 patterns, and 0·Inf = NaN would leak
+  // through the padded score columns
+  int s_pad = (S + 31) & ~31;
+  ushort8 z8{0, 0, 0, 0, 0, 0, 0, 0};
+  // K: thread t covers octet slices; row = s_k, 8 octets per row
+  for (int idx = tid; idx < s_pad * 8; idx += 256) {
+    int row = idx >> 3, q = idx & 7;
+    // store octet q of row at slot q^swz(row): read side XORs the same
+    k_img[row * 8 + att_swz(q, row)] =
+        row < S ? *(const ushort8 *)(kp + (long)row * qkv_row + q * 8) : z8;
+  }
+  // V: subtile st = (kq, cq): rows kq*4..+3 (s_k), cols cq*16..+15 (d)
+  // slot layout: [st][kl*2 + ch] with ch = 8-col half (TnPipe image)
+  for (int idx = tid; idx < s_pad * 8; idx += 256) {
+    int st = idx >> 3, kq = st >> 2, cq = st & 3;
+    int kl = (idx & 7) >> 1, ch = idx & 1;
+    int sk = kq * 4 + kl, d0 = cq * 16 + ch * 8;
+    v_img[idx] =
+        sk < S ? *(const ushort8 *)(vp + (long)sk * qkv_row + d0) : z8;
+  }
+  __syncthreads();
+
+  // Q fragments straight from global: lane holds row (l&31), k-octet (l>>5)
+  int qrow0 = wave * 32;
+  if (qrow0 >= S) return; // short sequences: idle waves (after the barrier)
+  const uint16_t *qp = base + h * D;
+  bf16x8a qf[4]; // 4 k-steps of 16 over D=64
+  {
+    int r = qrow0 + (lane & 31);
+    bool rok = r < S;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      int q = ks * 2 + (lane >> 5);
+      qf[ks] = us8_to_bf8a(rok ? *(const ushort8 *)(qp + (long)r * qkv_row + q * 8)
+                               : ushort8{0, 0, 0, 0, 0, 0, 0, 0});
+    }
+  }
+
+  // scores: per wave 32 rows × S cols as S/32 col-frags
+  constexpr int NF = S_MAX / 32;
+  float16a acc[NF];
+  int nf = (S + 31) / 32;
+#pragma unroll 4
+  for (int ni = 0; ni < nf; ++ni) {
+    float16a a = {};
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      int kr = ni * 32 + (lane & 31); // key row for the B fragment
+      int q = ks * 2 + (lane >> 5);
+      bf16x8a kf = us8_to_bf8a(k_img[kr * 8 + att_swz(q, kr)]);
+      a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kf, a, 0, 0, 0);
+    }
+    acc[ni] = a;
+  }
+
+  // row softmax in fp32. value (row, col): reg r of frag ni belongs to
+  // row = qrow0 + (r&3) + 8*(r>>2) + 4*(lane>>5), col = ni*32 + (lane&31):
+  // a row's values live in one half-wave at fixed r across frags.
+  float mrow[16], srow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    float m = -3.4e38f;
+#pragma unroll 4
+    for (int ni = 0; ni < nf; ++ni) {
+      float v = acc[ni][r] * scale;
+      int col = ni * 32 + (lane & 31);
+      if (col >= S)
+        v = -3.4e38f; // padded key columns (SET, not add: acc was garbage)
+      else if (mask)
+        v += mask[(long)b * S + col];
+      acc[ni][r] = v;
+      m = fmaxf(m, v);
+    }
+    // half-wave reduce (the 32 lanes holding this row)
+#pragma unroll
+    for (int off = 16; off > 0; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off, 64));
+    float s = 0.f;
+#pragma unroll 4
+    for (int ni = 0; ni < nf; ++ni) {
+      float e = __expf(acc[ni][r] - m);
+      acc[ni][r] = e;
+      s += e;
+    }
+#pragma unroll
+    for (int off = 16; off > 0; off >>= 1)
+      s += __shfl_xor(s, off, 64);
+    mrow[r] = m;
+    srow[r] = 1.f / s;
+  }
+
+  // normalize + park P in this wave's LDS image (and optionally to global)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int prow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+#pragma unroll 4
+    for (int ni = 0; ni < nf; ++ni) {
+      int col = ni * 32 + (lane & 31);
+      uint16_t pv = f2bf(acc[ni][r] * srow[r]);
+      p_img[wave][prow][col] = pv;
+      if (probs && col < S && qrow0 + prow < S)
+        probs[((long)bh * S + qrow0 + prow) * S + col] = pv;
+    }
+  }
+  // P image is per-wave private; the producing lanes and consuming lanes
+  // are in the SAME wave, so a wave-level LDS fence suffices (no barrier:
+  // waves run decoupled from here on)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+  // O = P·V: A-frags from p_img (row-major b128), B-frags via tr_b16
+  float16a oacc[2] = {};
+  unsigned vbase = (unsigned)(unsigned long)(
+      __attribute__((address_space(3))) const void *)v_img;
+  int nks = (S + 15) / 16; // k-steps of 16 over s_k
+  for (int ks = 0; ks < nks; ++ks) {
+    bf16x8a pf;
+    {
+      int r = lane & 31, k0 = ks * 16 + (lane >> 5) * 8;
+      pf = us8_to_bf8a(*(const ushort8 *)&p_img[wave][r][k0]);
+    }
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      // V fragment: lane holds d-col (ni*32 + l&31), k 8 deep at
+      // ks*16 + (l>>5)*8: two tr reads at consecutive k-subtiles
+      int kq0 = ks * 4 + ((lane >> 5) & 1) * 2;
+      int cq = ni * 2 + ((lane >> 4) & 1);
+      unsigned a0 = vbase + (unsigned)((kq0 * 4 + cq) * 128 + (lane & 15) * 8);
+      uint2a lo, hi;
+      asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                   "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
+                   "s_waitcnt lgkmcnt(0)"
+                   : "=&v"(lo), "=&v"(hi)
+                   : "v"(a0)
+                   : "memory");
+      union { unsigned u[4]; bf16x8a v; } vv;
+      vv.u[0] = lo.x; vv.u[1] = lo.y; vv.u[2] = hi.x; vv.u[3] = hi.y;
+      oacc[ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vv.v, oacc[ni],
+                                                         0, 0, 0);
+    }
+  }
+
+  // write ctx [b][s][h*64 + d]
+  long crow = (long)H * D;
+  uint16_t *cp = ctx + (long)b * S * crow + h * D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = qrow0 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    if (row >= S) continue;
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int d = ni * 32 + (lane & 31);
+      cp[(long)row * crow + d] = f2bf(oacc[ni][r]);
+    }
+  }
+}
+
+extern "C" hipError_t attn_fwd_launch(const void *qkv, void *ctx, void *probs,
+                                      const float *mask, int B, int S, int H,
+                                      float scale, hipStream_t strm) {
+  // LDS: S_MAX=128 → 64 KiB (2 blocks/CU); S_MAX=256 → 128 KiB (1/CU).
+  // S > 256 needs a tiled/online-softmax variant (torch fallback upstream).
+  if (S <= 128)
+    attn_fwd_k<128><<<B * H, 256, 0, strm>>>(
+        (const uint16_t *)qkv, (uint16_t *)ctx, (uint16_t *)probs, mask, B, S,
+        H, scale);
+  else if (S <= 256)
+    attn_fwd_k<256><<<B * H, 256, 0, strm>>>(
+        (const uint16_t *)qkv, (uint16_t *)ctx, (uint16_t *)probs, mask, B, S,
+        H, scale);
+  else
+    return hipErrorInvalidValue;
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}

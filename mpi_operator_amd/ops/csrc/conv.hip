@@ -215,7 +215,7 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
     GemmLoader la{(const uint16_t *)x, (int)M, (long)C, C};
     GemmLoader lb{(const uint16_t *)w, Kout, (long)C, C};
     e = launch_nt_gemm(la, lb, out, (int)M, Kout, C, Kout, f32, strm, splits);
-  } else if (use_pipemix()) {
+  } else if (use_pipegather()) {
     // gather fwd on the deep pipeline: x patches via per-lane glds source
     // addresses (zeros page for padding), weights plain NT
     NtPipe<ConvFwdSrc> sa{
@@ -420,7 +420,7 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
       e = launch_mix_gemm(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb}, out,
                           (int)M, C, Kout, C, f32, strm, splits);
     }
-  } else if (use_pipemix()) {
+  } else if (use_pipegather()) {
     NtPipe<ConvDgradSrc<1>> sa{
         {(const uint16_t *)dy, H, W, Kout, HO, WO, S, pad, K, M}};
     TnPipe<DgradWTnSrc> sb{{(const uint16_t *)w, C, Kout, K, R * S * C}};
@@ -505,7 +505,7 @@ extern "C" hipError_t conv_wgrad_implicit(const void *dy, const void *x,
       e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
                           partial, Kout, RSC, (int)M, RSC, true, strm, splits);
     }
-  } else if (use_pipemix()) {
+  } else if (use_pipegather()) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)dy, (long)Kout, (int)M, Kout}};
     TnPipe<XcolSrc> sb{
         {(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, RSC, M}};

@@ -60,6 +60,16 @@ static inline bool use_pipemix() {
   return on;
 }
 
+// the conv GATHER routes separately (their per-stage address math is
+// heavier than the plain pipes): MPIAMD_PIPEGATHER=0 keeps them on mix
+static inline bool use_pipegather() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_PIPEGATHER");
+    return !(e && e[0] == '0');
+  }();
+  return use_pipemix() && on;
+}
+
 // ---- SrcMaps ----------------------------------------------------------
 // concept (stateful: each thread stages the same 2 (row/col, k-offset)
 // granules every k-tile, so the expensive decomposition hoists into init):
