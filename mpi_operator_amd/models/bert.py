@@ -102,6 +102,12 @@ class SelfAttention(nn.Module):
     def forward(self, x, attn_mask=None):
         b, s, h = x.shape
         qkv = self.qkv(x).view(b, s, 3, self.heads, self.head_dim)
+        if (x.is_cuda and x.dtype == torch.bfloat16 and self.head_dim == 64
+                and s <= 256 and attn_mask is None):
+            # fused QKᵀ→softmax→PV kernel (one launch per step instead of
+            # the batched-matmul + softmax chain); emits ctx in [b, s, h]
+            # directly — no transpose/reshape passes
+            return self.out(Fx.attention(qkv, self.heads, self.scale))
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # b, nh, s, hd
         scores = torch.matmul(q, k.transpose(-1, -2)) * self.scale
         if attn_mask is not None:

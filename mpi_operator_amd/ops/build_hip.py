@@ -19,7 +19,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "hipcc")
 
 KERNEL_SRCS = ["elementwise.hip", "batchnorm.hip", "pooling.hip",
-               "softmax_xent.hip", "masked_xent.hip", "gemm.hip", "conv.hip", "layernorm.hip"]
+               "softmax_xent.hip", "masked_xent.hip", "attention.hip", "gemm.hip", "conv.hip", "layernorm.hip"]
 BINDING_SRC = "bindings.cpp"
 
 BASE_FLAGS = [f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",

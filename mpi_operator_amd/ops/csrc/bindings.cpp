@@ -97,6 +97,8 @@ hipError_t masked_xent_fwd_launch(const void *, const long *, float *, float *,
 hipError_t masked_xent_bwd_launch(const void *, const long *, const float *,
                                   const float *, const float *, void *, int,
                                   int, long, hipStream_t);
+hipError_t attn_fwd_launch(const void *, void *, void *, const float *, int,
+                           int, int, float, hipStream_t);
 hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, const float *, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
@@ -484,6 +486,41 @@ static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
   return {dw, db};
 }
 
+// ------------------------- fused attention -------------------------
+// qkv: [B, S, 3, H, 64] (the fused projection output, untransposed);
+// returns (ctx [B, S, H*64], probs [B, H, S, S] for the torch backward)
+static std::vector<Tensor> attn_fwd(const Tensor &qkv, int64_t heads,
+                                    double scale,
+                                    const c10::optional<Tensor> &mask,
+                                    bool want_probs) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3 && qkv.size(4) == 64,
+              "qkv must be [B,S,3,H,64]");
+  TORCH_CHECK(qkv.is_contiguous());
+  const HIPDeviceGuard guard(qkv.device());
+  int B = qkv.size(0), S = qkv.size(1), H = qkv.size(3);
+  TORCH_CHECK(H == heads);
+  TORCH_CHECK(S <= 256, "fused attention supports seq <= 256");
+  Tensor ctx = at::empty({B, S, (long)H * 64}, qkv.options());
+  Tensor probs;
+  const float *mp = nullptr;
+  Tensor maskc;
+  if (mask.has_value()) {
+    maskc = mask->to(at::kFloat).contiguous();
+    TORCH_CHECK(maskc.numel() == (long)B * S, "mask must be [B,S] additive");
+    mp = maskc.data_ptr<float>();
+  }
+  void *pp = nullptr;
+  if (want_probs) {
+    probs = at::empty({B, (long)H, S, S}, qkv.options());
+    pp = probs.data_ptr();
+  }
+  CHK(attn_fwd_launch(qkv.data_ptr(), ctx.data_ptr(), pp, mp, B, S, H,
+                      (float)scale, cur_stream()));
+  if (!want_probs) probs = at::empty({0}, qkv.options());
+  return {ctx, probs};
+}
+
 // ------------------------- fused FFN (GELU) -------------------------
 // fc1 forward with the bias+GELU in the epilogue: returns (g, h_pre)
 static std::vector<Tensor> linear_gelu_fwd(const Tensor &x, const Tensor &w,
@@ -650,6 +687,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("masked_xent_fwd", &masked_xent_fwd);
   m.def("linear_gelu_fwd", &linear_gelu_fwd);
+  m.def("attn_fwd", &attn_fwd);
   m.def("linear_wgrad_only", &linear_wgrad_only);
   m.def("linear_gelu_dgrad", &linear_gelu_dgrad);
   m.def("masked_xent_bwd", &masked_xent_bwd);

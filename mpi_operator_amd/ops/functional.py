@@ -365,3 +365,40 @@ class FfnFn(torch.autograd.Function):
 
 def ffn(x, w1, b1, w2, b2):
     return FfnFn.apply(x, w1, b1, w2, b2)
+
+
+class AttentionFn(torch.autograd.Function):
+    """Fused MHA forward (QKᵀ→softmax→PV in one kernel, probs saved bf16);
+    backward runs the standard four batched GEMMs + softmax-grad through
+    torch (shapes are library-friendly; the forward was the launch-bound
+    chain). qkv: [B, S, 3, H, 64] → ctx [B, S, H·64]."""
+
+    @staticmethod
+    def forward(ctx, qkv, heads, scale, mask):
+        out, probs = hip_ext().attn_fwd(qkv.contiguous(), heads, scale, mask,
+                                        True)
+        ctx.save_for_backward(qkv, probs)
+        ctx.heads, ctx.scale = heads, scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, probs = ctx.saved_tensors
+        B, S, _, H, D = qkv.shape
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # B,H,S,D
+        do = dout.view(B, S, H, D).transpose(1, 2)
+        p = probs.float()
+        dv = torch.matmul(p.transpose(-1, -2).to(do.dtype), do)
+        dp = torch.matmul(do, v.transpose(-1, -2)).float()
+        ds = (dp - (dp * p).sum(-1, keepdim=True)) * p * ctx.scale
+        ds = ds.to(do.dtype)
+        dq = torch.matmul(ds, k)
+        dk = torch.matmul(ds.transpose(-1, -2), q)
+        dqkv = torch.stack(
+            [dq.transpose(1, 2), dk.transpose(1, 2), dv.transpose(1, 2)],
+            dim=2)
+        return dqkv, None, None, None
+
+
+def attention(qkv, heads, scale, mask=None):
+    return AttentionFn.apply(qkv, heads, scale, mask)
