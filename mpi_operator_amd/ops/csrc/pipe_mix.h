@@ -60,12 +60,16 @@ static inline bool use_pipemix() {
   return on;
 }
 
-// the conv GATHER routes separately (their per-stage address math is
-// heavier than the plain pipes): MPIAMD_PIPEGATHER=0 keeps them on mix
+// The conv GATHER routes measured NET-NEGATIVE on the pipeline (3296 vs
+// 3390 img/s same-box A/B): their ptr16 re-derives the pixel/tap split
+// per stage (two int64 divmods per granule) where the mix stagers hoist
+// it — the glds win doesn't cover the extra VALU at 2 blocks/CU. Default
+// OFF (mix path); MPIAMD_PIPEGATHER=1 re-enables for A/B after the
+// carry-chain SrcMap optimization.
 static inline bool use_pipegather() {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_PIPEGATHER");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   return use_pipemix() && on;
 }
