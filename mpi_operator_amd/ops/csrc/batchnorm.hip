@@ -319,6 +319,28 @@ extern "C" hipError_t bn_fwd_train_launch(
   return hipSuccess;
 }
 
+// Variant taking PRE-COMPUTED partials (the conv epilogue's BnStatsWriter
+// slab, [pre_grid][2][C]) — skips bn_partials' full activation re-read.
+extern "C" hipError_t bn_fwd_train_pre_launch(
+    const float *pre_partial, int pre_grid, const void *x, const void *res,
+    const float *gamma, const float *beta, float eps, int relu, void *y,
+    float *mean, float *invstd, float *scale, float *shift,
+    float *running_mean, float *running_var, float momentum, long M, int C,
+    hipStream_t s) {
+  int C8 = C / 8;
+  if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
+  float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  bn_finalize_fwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+      pre_partial, pre_grid, C, gamma, beta, 1.f / (float)M, eps, mean,
+      invstd, scale, shift, running_mean, running_var, momentum, unbias);
+  HIP_KERNEL_CHECK();
+  bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
+      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y, M,
+      C8, relu);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
 extern "C" hipError_t bn_fwd_eval_launch(const void *x, const float *scale,
                                          const float *shift, int relu, void *y,
                                          long M, int C, hipStream_t s) {

@@ -127,6 +127,14 @@ hipError_t conv_fwd(const void *, const void *, void *, int, int, int, int,
 hipError_t conv_dgrad(const void *, const void *, void *, int, int, int, int,
                       int, int, int, int, int, int, int, int, float *,
                       hipStream_t);
+hipError_t conv_fwd_bn(const void *, const void *, void *, int, int, int,
+                       int, int, int, int, int, int, int, int, float *,
+                       hipStream_t);
+hipError_t bn_fwd_train_pre_launch(const float *, int, const void *,
+                                   const void *, const float *, const float *,
+                                   float, int, void *, float *, float *,
+                                   float *, float *, float *, float *, float,
+                                   long, int, hipStream_t);
 hipError_t conv_dgrad_1x1_acc(const void *, const void *, void *, long, int,
                               int, hipStream_t);
 hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
@@ -254,6 +262,64 @@ static void conv2d_dgrad_acc(const Tensor &dy, const Tensor &w, Tensor &dx_acc) 
 // ------------------------- batchnorm -------------------------
 // res (optional, may be undefined): residual tensor folded into the apply
 // pass — y = [relu](bn(x) + res), the bottleneck-join fusion.
+// conv forward + fused BN partial stats: returns (y, slab) — slab empty
+// when the shape takes a split-K route (stats would see partial sums).
+static std::vector<Tensor> conv2d_fwd_bn(const Tensor &x, const Tensor &w,
+                                         int64_t stride, int64_t pad) {
+  check_cl_bf16(x, "x");
+  check_cl_bf16(w, "w");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kout = w.size(0), R = w.size(2), S = w.size(3);
+  int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
+  long M = (long)N * HO * WO;
+  int splits = conv_splits(M, Kout, R * S * C);
+  auto f32 = x.options().dtype(at::kFloat);
+  if (splits > 1 || Kout % 8 != 0) {
+    Tensor y = conv2d_fwd(x, w, stride, pad);
+    return {y, at::empty({0}, f32)};
+  }
+  Tensor y = empty_cl_bf16(N, Kout, HO, WO, x);
+  long bands = (M + 63) / 64;
+  Tensor slab = at::empty({bands, 2, (long)Kout}, f32);
+  CHK(conv_fwd_bn(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W, C, Kout,
+                  R, S, (int)stride, (int)pad, HO, WO,
+                  slab.data_ptr<float>(), cur_stream()));
+  return {y, slab};
+}
+
+// bn_fwd_train consuming the conv epilogue's stats slab (skips partials)
+static std::vector<Tensor> bn_fwd_train_pre(
+    const Tensor &x, const Tensor &slab, const Tensor &gamma,
+    const Tensor &beta, double eps, bool relu, const Tensor &running_mean,
+    const Tensor &running_var, double momentum, const Tensor &res) {
+  check_cl_bf16(x, "x");
+  const HIPDeviceGuard guard(x.device());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long M = (long)N * H * W;
+  TORCH_CHECK(slab.numel() == ((M + 63) / 64) * 2 * (long)C, "slab shape");
+  auto f32 = x.options().dtype(at::kFloat);
+  Tensor y = empty_cl_bf16(N, C, H, W, x);
+  Tensor mean = at::empty({C}, f32), invstd = at::empty({C}, f32);
+  Tensor scale = at::empty({C}, f32), shift = at::empty({C}, f32);
+  float *rm = running_mean.defined() && running_mean.numel() == C
+                  ? running_mean.data_ptr<float>() : nullptr;
+  float *rv = rm ? running_var.data_ptr<float>() : nullptr;
+  const void *resp = nullptr;
+  if (res.defined() && res.numel() > 0) {
+    check_cl_bf16(res, "res");
+    resp = res.data_ptr();
+  }
+  CHK(bn_fwd_train_pre_launch(
+      slab.data_ptr<float>(), (int)((M + 63) / 64), x.data_ptr(), resp,
+      gamma.data_ptr<float>(), beta.data_ptr<float>(), (float)eps,
+      relu ? 1 : 0, y.data_ptr(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(), scale.data_ptr<float>(),
+      shift.data_ptr<float>(), rm, rv, (float)momentum, M, C, cur_stream()));
+  return {y, mean, invstd};
+}
+
 static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
                                         const Tensor &beta, double eps,
                                         bool relu, const Tensor &running_mean,
@@ -670,6 +736,8 @@ static Tensor gemm_nt_b(const Tensor &a, const Tensor &b, bool c_f32) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_fwd_bn", &conv2d_fwd_bn);
+  m.def("bn_fwd_train_pre", &bn_fwd_train_pre);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_dgrad_acc", &conv2d_dgrad_acc);
   m.def("add_bf16", &add_bf16_b);

@@ -233,6 +233,47 @@ extern "C" hipError_t conv_fwd(const void *x, const void *w, void *y, int N,
   return splitk_reduce(partial, splits, M * Kout, y, 1, strm);
 }
 
+// conv forward that ALSO emits BatchNorm partial statistics from the GEMM
+// epilogue (BnStatsWriter slab [ceil(M/64)][2][Kout]). splits==1 only —
+// the caller (bindings conv2d_fwd_bn) checks and falls back otherwise.
+extern "C" hipError_t conv_fwd_bn(const void *x, const void *w, void *y,
+                                  int N, int H, int W, int C, int Kout, int R,
+                                  int S, int stride, int pad, int HO, int WO,
+                                  float *bn_slab, hipStream_t strm) {
+  long M = (long)N * HO * WO;
+  int K = R * S * C;
+  BnStatsWriter wrt{(long)Kout, bn_slab, Kout};
+  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+    GemmLoader la{(const uint16_t *)x, (int)M, (long)C, C};
+    GemmLoader lb{(const uint16_t *)w, Kout, (long)C, C};
+    if (M % 256 == 0 && Kout % 256 == 0 && C % 64 == 0 &&
+        (long)(M / 256) * (Kout / 256) >= 128)
+      return launch_pipe256_wr(la, lb, y, (int)M, Kout, C, Kout, false, wrt,
+                               strm);
+    if (use_pipemix()) {
+      NtPipe<PlainNtSrc> sa{{(const uint16_t *)x, (long)C, (int)M, C}};
+      NtPipe<PlainNtSrc> sb{{(const uint16_t *)w, (long)C, Kout, C}};
+      return launch_pipe_mix_wr(sa, sb, y, (int)M, Kout, C, wrt, Kout, false,
+                                strm);
+    }
+    GldsNt ga{la.p, la.rows, la.ld, la.kdim};
+    GldsNt gb{lb.p, lb.rows, lb.ld, lb.kdim};
+    return launch_mix_gemm_wr(ga, gb, y, (int)M, Kout, C, wrt, Kout, false,
+                              strm);
+  }
+  if (use_pipegather()) {
+    NtPipe<ConvFwdSrc> sa{
+        {(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, K, M}};
+    NtPipe<PlainNtSrc> sb{{(const uint16_t *)w, (long)K, Kout, K}};
+    return launch_pipe_mix_wr(sa, sb, y, (int)M, Kout, K, wrt, Kout, false,
+                              strm);
+  }
+  ConvFwdStage sa{(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, K, M};
+  GemmLoader lb{(const uint16_t *)w, Kout, (long)K, K};
+  return launch_mix_gemm_wr(sa, NtStage<GemmLoader>{lb}, y, (int)M, Kout, K,
+                            wrt, Kout, false, strm);
+}
+
 // 1x1 stride-1 dgrad that ACCUMULATES into dx (bottleneck backward: the
 // conv1 input-grad lands directly on the skip-connection gradient, so the
 // residual join needs no separate elementwise add pass).

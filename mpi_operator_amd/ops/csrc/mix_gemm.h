@@ -137,6 +137,7 @@ typedef __attribute__((ext_vector_type(16))) float float16v;
 // of the column loop; store_*(ptr, ctx, col, v) places one value.
 struct LinearWriter {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   long ldc;
   typedef long RowCtx;
   DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
@@ -153,6 +154,7 @@ struct LinearWriter {
 // BERT-Large step).
 struct BiasWriter {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   long ldc;
   const float *bias;
   typedef long RowCtx;
@@ -183,6 +185,7 @@ DEV_INLINE float dgelu_f(float x) {
 // whole intermediate activation (BERT-Large: 32 MB/layer at 8 TB/s).
 struct GeluBiasWriter {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   long ldc;
   const float *bias;
   uint16_t *pre; // h (pre-activation), bf16, same layout as C
@@ -205,6 +208,7 @@ struct GeluBiasWriter {
 // (2 reads + 1 write over the intermediate activation per layer).
 struct GeluBwdWriter {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   long ldc;
   const uint16_t *pre; // h (pre-activation) saved by GeluBiasWriter
   typedef long RowCtx;
@@ -217,10 +221,75 @@ struct GeluBwdWriter {
   }
 };
 
+// Conv-forward epilogue that ALSO produces the BatchNorm per-channel
+// partial statistics (Σy, Σy²) — removes bn_partials' full re-read of the
+// activation it just wrote (~9% of a ResNet step at 8 TB/s).
+//
+// Deterministic, atomic-free: the slab has one entry per 64 OUTPUT ROWS
+// ([ceil(M/64)][2][C], pre-zeroed by the host). In every tile kernel here
+// (mix/pipe_mix 2×2 waves, pipe256 2×4) a lane's epilogue rows all fall in
+// ONE 64-row band per m-frag half, and a (band, channel) pair is written
+// by exactly one lane pair (l, l^32) — combined with one shfl, stored
+// plain. Stats use the ROUNDED bf16 value (what bn_partials would read).
+struct BnStatsWriter {
+  static constexpr bool ACC = false;
+  static constexpr bool STATS = true;
+  long ldc;
+  float *slab; // [ceil(M/64)][2][C]
+  int C;
+  // mutable per-lane accumulators: [row-band half][col slot (bit5 of col)]
+  mutable float s0[2][2], s1[2][2];
+  mutable int tm[2], colc[2];
+  mutable int half_;
+  typedef long RowCtx;
+  DEV_INLINE void reset() const {
+    s0[0][0] = s0[0][1] = s0[1][0] = s0[1][1] = 0.f;
+    s1[0][0] = s1[0][1] = s1[1][0] = s1[1][1] = 0.f;
+    tm[0] = tm[1] = -1;
+    colc[0] = colc[1] = -1;
+    half_ = 0;
+  }
+  DEV_INLINE RowCtx row_ctx(int row) const {
+    // pipe256's 4 m-frags span two 64-row bands per lane; track which
+    half_ = (row >> 6) & 1;
+    tm[half_] = row >> 6;
+    return (long)row * ldc;
+  }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] = v; // split-K callers never attach stats (caller contract)
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    uint16_t r = f2bf(v);
+    p[b + col] = r;
+    float vr = bf2f(r);
+    int ci = (col >> 5) & 1;
+    colc[ci] = col;
+    s0[half_][ci] += vr;
+    s1[half_][ci] += vr * vr;
+  }
+  DEV_INLINE void flush(int lane) const {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      // bands are per-(lane-half) identical; tm/col agree across l, l^32
+#pragma unroll
+      for (int ci = 0; ci < 2; ++ci) {
+        float a = s0[h][ci] + __shfl_xor(s0[h][ci], 32, 64);
+        float b = s1[h][ci] + __shfl_xor(s1[h][ci], 32, 64);
+        if (lane < 32 && tm[h] >= 0 && colc[ci] >= 0) {
+          float *e = slab + (long)tm[h] * 2 * C;
+          e[colc[ci]] = a;
+          e[C + colc[ci]] = b;
+        }
+      }
+    }
+  }
+};
+
 // += into an existing bf16 tensor (bottleneck backward: conv1's dgrad
 // accumulates onto the skip-connection gradient — no separate add pass).
 struct LinearAccWriter {
   static constexpr bool ACC = true;
+  static constexpr bool STATS = false;
   long ldc;
   typedef long RowCtx;
   DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
@@ -234,6 +303,7 @@ struct LinearAccWriter {
 
 struct Stride2Writer {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   int W2, H2, ph, pw, W, H, C;
   typedef long RowCtx;
   DEV_INLINE RowCtx row_ctx(int row) const {
@@ -255,6 +325,7 @@ struct Stride2Writer {
 // epilogue zeroes the three sibling pixels itself — no 100 MB memset pass.
 struct Stride2ZeroWriter {
   static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
   int W2, H2, ph, pw, W, H, C;
   struct RowCtx {
     long base;
@@ -384,6 +455,7 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
 
   // 32x32 C/D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
   // Row-major loop order so each row's address context is computed once.
+  if constexpr (WR::STATS) wrt.reset();
   cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
@@ -403,6 +475,7 @@ __global__ __launch_bounds__(NT_THREADS) void mix_gemm_k(
       }
     }
   }
+  if constexpr (WR::STATS) wrt.flush(lane);
 }
 
 #undef MXG_A

@@ -75,6 +75,9 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
   __shared__ ushort8 lds[8 * HSZ]; // [buf][op][kh] = 128 KiB
 #define P256_IMG(buf, op, kh) (lds + (((buf) * 2 + (op)) * 2 + (kh)) * HSZ)
 
+  if (USE_WR) {
+    if constexpr (WR::STATS) wrt.reset();
+  }
   float16v acc[4][2] = {};
   int nk = K / 64; // contract: K % 64 == 0
 
@@ -203,6 +206,9 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
           ((uint16_t *)cptr)[base + col] = f2bf(v);
       }
     }
+  }
+  if (USE_WR) {
+    if constexpr (WR::STATS) wrt.flush(lane);
   }
 }
 #undef P256_IMG

@@ -339,6 +339,7 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
 
   // 128-B aligned: the tr_b16 read resolves its block as addr & ~127
   __shared__ __align__(128) ushort8 lds[8 * PM_HSZ]; // [buf][op][kh] = 64 KiB
+  if constexpr (WR::STATS) wrt.reset();
 #define PM_IMG(buf, op, kh) (lds + (((buf) * 2 + (op)) * 2 + (kh)) * PM_HSZ)
 
   float16v acc[2][2] = {};
@@ -445,6 +446,7 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
       }
     }
   }
+  if constexpr (WR::STATS) wrt.flush(lane);
 }
 #undef PM_IMG
 

@@ -37,20 +37,31 @@ class BottleneckFn(torch.autograd.Function):
                 wd, gd, bd, rmd, rvd):
         ext = hip_ext()
         e = _empty()
-        a1 = ext.conv2d_fwd(x, w1, 1, 0)
-        y1, m1, v1 = ext.bn_fwd_train(a1, g1, b1, eps, True, rm1, rv1, momentum, e)
-        a2 = ext.conv2d_fwd(y1, w2, stride, 1)
-        y2, m2, v2 = ext.bn_fwd_train(a2, g2, b2, eps, True, rm2, rv2, momentum, e)
-        a3 = ext.conv2d_fwd(y2, w3, 1, 0)
+
+        def conv_bn(xin, w, st, pad, g, b, eps_, relu, rm, rv, mom, res):
+            """conv with BN stats fused into the GEMM epilogue; the empty-
+            slab return (split-K shapes) falls back to the partials pass."""
+            a, slab = ext.conv2d_fwd_bn(xin, w, st, pad)
+            if slab.numel() > 0:
+                y, m, v = ext.bn_fwd_train_pre(a, slab, g, b, eps_, relu,
+                                               rm, rv, mom, res)
+            else:
+                y, m, v = ext.bn_fwd_train(a, g, b, eps_, relu, rm, rv, mom,
+                                           res)
+            return a, y, m, v
+
+        a1, y1, m1, v1 = conv_bn(x, w1, 1, 0, g1, b1, eps, True, rm1, rv1,
+                                 momentum, e)
+        a2, y2, m2, v2 = conv_bn(y1, w2, stride, 1, g2, b2, eps, True, rm2,
+                                 rv2, momentum, e)
         has_ds = wd is not None
         if has_ds:
-            ad = ext.conv2d_fwd(x, wd, stride, 0)
-            res, md, vd = ext.bn_fwd_train(ad, gd, bd, eps, False, rmd, rvd,
-                                           momentum, e)
+            ad, res, md, vd = conv_bn(x, wd, stride, 0, gd, bd, eps, False,
+                                      rmd, rvd, momentum, e)
         else:
             ad = res = md = vd = None
-        out, m3, v3 = ext.bn_fwd_train(a3, g3, b3, eps, True, rm3, rv3,
-                                       momentum, res if has_ds else x)
+        a3, out, m3, v3 = conv_bn(y2, w3, 1, 0, g3, b3, eps, True, rm3, rv3,
+                                  momentum, res if has_ds else x)
         saved = [x, a1, y1, a2, y2, a3, out, m1, v1, m2, v2, m3, v3,
                  w1, g1, w2, g2, w3, g3]
         if has_ds:
