@@ -319,6 +319,49 @@ extern "C" hipError_t bn_fwd_train_launch(
   return hipSuccess;
 }
 
+// Band-slab finalize: the conv-epilogue slab has ceil(M/64) entries (up to
+// ~3.1k at ResNet stage-1 sizes) vs the partials path's ≤512 — the 32-lane
+// lane8_sums loop went latency-bound there. 256 lanes per channel: 12
+// iterations and a full-chip grid.
+__global__ void bn_finalize_fwd_bands_k(
+    const float *__restrict__ partial, int grid, int C,
+    const float *__restrict__ gamma, const float *__restrict__ beta,
+    float inv_m, float eps, float *__restrict__ mean,
+    float *__restrict__ invstd, float *__restrict__ scale,
+    float *__restrict__ shift, float *__restrict__ running_mean,
+    float *__restrict__ running_var, float momentum, float unbias) {
+  int c = blockIdx.x; // one block (256 threads) per channel
+  if (c >= C) return;
+  float s = 0.f, sq = 0.f;
+  for (int g = threadIdx.x; g < grid; g += 256) {
+    s += partial[(long)g * 2 * C + c];
+    sq += partial[(long)g * 2 * C + C + c];
+  }
+  __shared__ float red[2][256 / WAVE];
+  s = wave_sum(s);
+  sq = wave_sum(sq);
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    red[0][threadIdx.x / WAVE] = s;
+    red[1][threadIdx.x / WAVE] = sq;
+  }
+  __syncthreads();
+  if (threadIdx.x != 0) return;
+  s = red[0][0] + red[0][1] + red[0][2] + red[0][3];
+  sq = red[1][0] + red[1][1] + red[1][2] + red[1][3];
+  float mu = s * inv_m;
+  float var = fmaxf(sq * inv_m - mu * mu, 0.f);
+  float is = rsqrtf(var + eps);
+  mean[c] = mu;
+  invstd[c] = is;
+  float sc = gamma[c] * is;
+  scale[c] = sc;
+  shift[c] = beta[c] - mu * sc;
+  if (running_mean) {
+    running_mean[c] = running_mean[c] * (1.f - momentum) + mu * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + var * unbias * momentum;
+  }
+}
+
 // Variant taking PRE-COMPUTED partials (the conv epilogue's BnStatsWriter
 // slab, [pre_grid][2][C]) — skips bn_partials' full activation re-read.
 extern "C" hipError_t bn_fwd_train_pre_launch(
@@ -330,7 +373,7 @@ extern "C" hipError_t bn_fwd_train_pre_launch(
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
-  bn_finalize_fwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+  bn_finalize_fwd_bands_k<<<C, 256, 0, s>>>(
       pre_partial, pre_grid, C, gamma, beta, 1.f / (float)M, eps, mean,
       invstd, scale, shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
