@@ -64,10 +64,11 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
   // stage K/V padded to a 32-row multiple with ZEROS — un-staged LDS
   // garbage can be Inf/NaN bf16 patterns, and 0·Inf = NaN would leak
   // through the padded score columns
-  int s_pad = (S + 31) & ~31;
+  // staged to the full S_MAX capacity so every compile-time-unrolled
+  // fragment read sees defined (zero) data
   ushort8 z8{0, 0, 0, 0, 0, 0, 0, 0};
   // K: thread t covers octet slices; row = s_k, 8 octets per row
-  for (int idx = tid; idx < s_pad * 8; idx += 256) {
+  for (int idx = tid; idx < S_MAX * 8; idx += 256) {
     int row = idx >> 3, q = idx & 7;
     // store octet q of row at slot q^swz(row): read side XORs the same
     k_img[row * 8 + att_swz(q, row)] =
@@ -75,7 +76,7 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
   }
   // V: subtile st = (kq, cq): rows kq*4..+3 (s_k), cols cq*16..+15 (d)
   // slot layout: [st][kl*2 + ch] with ch = 8-col half (TnPipe image)
-  for (int idx = tid; idx < s_pad * 8; idx += 256) {
+  for (int idx = tid; idx < S_MAX * 8; idx += 256) {
     int st = idx >> 3, kq = st >> 2, cq = st & 3;
     int kl = (idx & 7) >> 1, ch = idx & 1;
     int sk = kq * 4 + kl, d0 = cq * 16 + ch * 8;
@@ -100,12 +101,15 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
     }
   }
 
-  // scores: per wave 32 rows × S cols as S/32 col-frags
+  // scores: per wave 32 rows × S_MAX cols as compile-time col-frags.
+  // COMPILE-TIME bounds are load-bearing: a runtime `nf` bound makes
+  // acc[] runtime-indexed → every accumulator spills to scratch (the
+  // first cut of this kernel ran 168 µs; padded frags compute zeros
+  // instead and the -inf mask drops them)
   constexpr int NF = S_MAX / 32;
   float16a acc[NF];
-  int nf = (S + 31) / 32;
-#pragma unroll 4
-  for (int ni = 0; ni < nf; ++ni) {
+#pragma unroll
+  for (int ni = 0; ni < NF; ++ni) {
     float16a a = {};
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
@@ -124,8 +128,8 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     float m = -3.4e38f;
-#pragma unroll 4
-    for (int ni = 0; ni < nf; ++ni) {
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
       float v = acc[ni][r] * scale;
       int col = ni * 32 + (lane & 31);
       if (col >= S)
@@ -140,8 +144,8 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
     for (int off = 16; off > 0; off >>= 1)
       m = fmaxf(m, __shfl_xor(m, off, 64));
     float s = 0.f;
-#pragma unroll 4
-    for (int ni = 0; ni < nf; ++ni) {
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
       float e = __expf(acc[ni][r] - m);
       acc[ni][r] = e;
       s += e;
@@ -157,8 +161,8 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     int prow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-#pragma unroll 4
-    for (int ni = 0; ni < nf; ++ni) {
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
       int col = ni * 32 + (lane & 31);
       uint16_t pv = f2bf(acc[ni][r] * srow[r]);
       p_img[wave][prow][col] = pv;
@@ -175,8 +179,9 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
   float16a oacc[2] = {};
   unsigned vbase = (unsigned)(unsigned long)(
       __attribute__((address_space(3))) const void *)v_img;
-  int nks = (S + 15) / 16; // k-steps of 16 over s_k
-  for (int ks = 0; ks < nks; ++ks) {
+  constexpr int NKS = S_MAX / 16; // padded P cols / V rows are zero
+#pragma unroll
+  for (int ks = 0; ks < NKS; ++ks) {
     bf16x8a pf;
     {
       int r = lane & 31, k0 = ks * 16 + (lane >> 5) * 8;

@@ -168,13 +168,21 @@ struct BiasWriter {
 };
 
 // tanh-approx GELU (BERT's activation) and its derivative, fp32.
+// tanh via one hardware exp (v_exp_f32): libm tanhf costs ~10× as many
+// VALU ops and dominated the fused-GELU GEMM epilogues (113 vs 58 µs on
+// the fc2-dx kernel). |rel err| ~1e-7 — far below bf16 output rounding.
+DEV_INLINE float tanh_fast(float u) {
+  float a = __expf(-2.f * fabsf(u));
+  float t = (1.f - a) / (1.f + a);
+  return u < 0.f ? -t : t;
+}
 DEV_INLINE float gelu_f(float x) {
   float u = 0.7978845608028654f * (x + 0.044715f * x * x * x);
-  return 0.5f * x * (1.f + tanhf(u));
+  return 0.5f * x * (1.f + tanh_fast(u));
 }
 DEV_INLINE float dgelu_f(float x) {
   float u = 0.7978845608028654f * (x + 0.044715f * x * x * x);
-  float t = tanhf(u);
+  float t = tanh_fast(u);
   return 0.5f * (1.f + t) +
          0.5f * x * (1.f - t * t) * 0.7978845608028654f *
              (1.f + 3.f * 0.044715f * x * x);
