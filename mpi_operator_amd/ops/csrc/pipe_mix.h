@@ -116,6 +116,16 @@ struct PlainTnSrc { // row-major [kdim][cols]: rows along k
   }
 };
 
+// small-divisor helpers for the per-stage k decompositions: ResNet's C/Q
+// are powers of two (shift+mask) and rs = k/C stays < R·S ≤ 49, so a
+// 16-bit reciprocal multiply is exact — no integer divides in ptr16.
+DEV_INLINE int ilog2_if_pow2(int v) {
+  return (v & (v - 1)) == 0 ? (31 - __builtin_clz(v)) : -1;
+}
+DEV_INLINE int div_small(int num, int recip16) { // exact for num ≤ ~1000
+  return (num * recip16) >> 16;
+}
+
 // conv forward x-patch gather (NT: rows are output pixels, k = (r,s,c)).
 // Row decomposition (n, ho·stride-pad, wo·stride-pad) hoisted; per stage
 // only k's (r,s,c) split + 2 bounds checks remain.
@@ -125,6 +135,7 @@ struct ConvFwdSrc {
   long M;
   long nbase_[2];
   int hb_[2], wb_[2], koff_[2];
+  int csh_, srecip_;
   bool ok_[2];
   DEV_INLINE void init(int i, int rc, int koff) {
     long m = rc;
@@ -137,11 +148,21 @@ struct ConvFwdSrc {
     hb_[i] = ho * stride - pad;
     wb_[i] = wo * stride - pad;
     koff_[i] = koff;
+    csh_ = ilog2_if_pow2(C);
+    srecip_ = (1 << 16) / S + 1;
   }
   DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
     int k = kb + koff_[i];
-    int c = k % C, rs = k / C;
-    int s_ = rs % S, rr = rs / S;
+    int c, rs;
+    if (csh_ >= 0) {
+      c = k & (C - 1);
+      rs = k >> csh_;
+    } else {
+      c = k % C;
+      rs = k / C;
+    }
+    int rr = div_small(rs, srecip_);
+    int s_ = rs - rr * S;
     int h = hb_[i] + rr, w = wb_[i] + s_;
     if (k < K && ok_[i] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
       return x + (nbase_[i] + (long)h * W + w) * C + c;
@@ -157,6 +178,7 @@ template <int STRIDE> struct ConvDgradSrc {
   long M;
   long nbase_[2];
   int hb_[2], wb_[2], koff_[2];
+  int qsh_, srecip_;
   bool ok_[2];
   DEV_INLINE void init(int i, int rc, int koff) {
     long m = rc;
@@ -169,11 +191,21 @@ template <int STRIDE> struct ConvDgradSrc {
     hb_[i] = h_ + pad;
     wb_[i] = w_ + pad;
     koff_[i] = koff;
+    qsh_ = ilog2_if_pow2(Q);
+    srecip_ = (1 << 16) / S + 1;
   }
   DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
     int k = kb + koff_[i];
-    int q = k % Q, rs = k / Q;
-    int s_ = rs % S, rr = rs / S;
+    int q, rs;
+    if (qsh_ >= 0) {
+      q = k & (Q - 1);
+      rs = k >> qsh_;
+    } else {
+      q = k % Q;
+      rs = k / Q;
+    }
+    int rr = div_small(rs, srecip_);
+    int s_ = rs - rr * S;
     int hn = hb_[i] - rr, wn = wb_[i] - s_;
     int ho = hn / STRIDE, wo = wn / STRIDE;
     if (k < K && ok_[i] && hn >= 0 && wn >= 0 &&
@@ -203,12 +235,20 @@ struct DgradWTnSrc {
 };
 
 // wgrad implicit-im2col gather (TN: k = output pixel m, cols = (r,s,c)).
-// Column decomposition (rr, ss, c) hoisted; per stage one m decomposition.
+// Column decomposition (rr, ss, c) hoisted. The pixel decomposition rides
+// a CARRY CHAIN: pipe_mix stages k-halves in strictly increasing kb order
+// (prologue 0,32,64,96 then +32 per phase), so each granule's (n, ho, wo)
+// advances by a precomputed (Δho = 32/WO, Δwo = 32%WO) instead of two
+// int64 divmods per stage — the cost that made the first gather-pipeline
+// cut 2.9% slower end-to-end.
 struct XcolSrc {
   const uint16_t *x;
   int H, W, C, HO, WO, S, stride, pad, RSC;
   long M;
   int rr_[2], ss_[2], coff_[2], koff_[2];
+  mutable long mcur_[2];
+  mutable int n_[2], ho_[2], wo_[2];
+  int dho_, dwo_;
   bool cok_[2];
   DEV_INLINE void init(int i, int rc, int koff) {
     cok_[i] = rc < RSC;
@@ -217,18 +257,38 @@ struct XcolSrc {
     rr_[i] = rs / S;
     coff_[i] = c;
     koff_[i] = koff;
+    mcur_[i] = -1;
+    dho_ = 32 / WO;
+    dwo_ = 32 % WO;
   }
   DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
     long m = kb + koff_[i];
-    if (m >= M || !cok_[i]) return nullptr;
-    int wo = (int)(m % WO);
-    long t = m / WO;
-    int ho = (int)(t % HO);
-    int n = (int)(t / HO);
-    int h = ho * stride + rr_[i] - pad, w = wo * stride + ss_[i] - pad;
-    if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-      return x + ((long)(n * H + h) * W + w) * C + coff_[i];
-    return nullptr;
+    if (mcur_[i] != m) { // first call of this split: full decomposition
+      wo_[i] = (int)(m % WO);
+      long t = m / WO;
+      ho_[i] = (int)(t % HO);
+      n_[i] = (int)(t / HO);
+    }
+    const uint16_t *out = nullptr;
+    if (m < M && cok_[i]) {
+      int h = ho_[i] * stride + rr_[i] - pad, w = wo_[i] * stride + ss_[i] - pad;
+      if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        out = x + ((long)(n_[i] * H + h) * W + w) * C + coff_[i];
+    }
+    // advance to the next stage's pixel (m + 32), carrying wo→ho→n
+    int wo = wo_[i] + dwo_, ho = ho_[i] + dho_;
+    if (wo >= WO) {
+      wo -= WO;
+      ++ho;
+    }
+    while (ho >= HO) { // bounded by 32/(WO*HO)+1 — ≥7×7 outputs: once
+      ho -= HO;
+      ++n_[i];
+    }
+    wo_[i] = wo;
+    ho_[i] = ho;
+    mcur_[i] = m + 32;
+    return out;
   }
 };
 

@@ -18,6 +18,12 @@ import torch
 
 from ._backend import hip_ext
 
+import os as _os
+
+# MPIAMD_FUSEBN=0 reverts to the separate bn_partials pass (A/B lever for
+# the conv-epilogue BN-stats fusion)
+_FUSEBN = _os.environ.get("MPIAMD_FUSEBN", "1") == "1"
+
 _EMPTY = None
 
 
@@ -41,6 +47,11 @@ class BottleneckFn(torch.autograd.Function):
         def conv_bn(xin, w, st, pad, g, b, eps_, relu, rm, rv, mom, res):
             """conv with BN stats fused into the GEMM epilogue; the empty-
             slab return (split-K shapes) falls back to the partials pass."""
+            if not _FUSEBN:
+                a = ext.conv2d_fwd(xin, w, st, pad)
+                y, m, v = ext.bn_fwd_train(a, g, b, eps_, relu, rm, rv, mom,
+                                           res)
+                return a, y, m, v
             a, slab = ext.conv2d_fwd_bn(xin, w, st, pad)
             if slab.numel() > 0:
                 y, m, v = ext.bn_fwd_train_pre(a, slab, g, b, eps_, relu,
