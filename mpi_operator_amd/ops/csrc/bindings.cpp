@@ -62,6 +62,8 @@ extern "C" {
 hipError_t add_relu_fwd(const void *, const void *, void *, long, hipStream_t);
 hipError_t add_bf16(const void *, const void *, void *, long, hipStream_t);
 hipError_t add_relu_bwd(const void *, const void *, void *, long, hipStream_t);
+hipError_t add_relu_bwd_add(const void *, const void *, const void *, void *,
+                            long, hipStream_t);
 hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t bias_add(void *, const float *, long, int, hipStream_t);
@@ -697,6 +699,17 @@ static Tensor add_relu_bwd_b(const Tensor &dy, const Tensor &y) {
   return dx;
 }
 
+static Tensor add_relu_bwd_add_b(const Tensor &dy, const Tensor &y,
+                                 const Tensor &dx0) {
+  const HIPDeviceGuard guard(dy.device());
+  TORCH_CHECK(dy.numel() == y.numel() && dy.numel() == dx0.numel() &&
+              dy.numel() % 8 == 0);
+  Tensor out = at::empty_like(dx0);
+  CHK(add_relu_bwd_add(dy.data_ptr(), y.data_ptr(), dx0.data_ptr(),
+                       out.data_ptr(), dy.numel(), cur_stream()));
+  return out;
+}
+
 // ------------------------- sgd -------------------------
 static void sgd_step(std::vector<Tensor> masters, std::vector<Tensor> grads,
                      std::vector<Tensor> momenta, std::vector<Tensor> outs,
@@ -762,6 +775,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("add_relu_fwd", &add_relu_fwd_b);
   m.def("add_relu_bwd", &add_relu_bwd_b);
+  m.def("add_relu_bwd_add", &add_relu_bwd_add_b);
   m.def("sgd_step", &sgd_step);
   m.def("gemm_nt", &gemm_nt_b);
   m.def("mfma_probe", [](const Tensor &a, const Tensor &b) {

@@ -86,6 +86,39 @@ __global__ void add_relu_bwd_k(const ushort8 *__restrict__ dy,
   }
 }
 
+// residual-join backward in ONE pass: dxt = dx0 + dout·(y > 0).
+// Replaces the materialized g = add_relu_bwd(dout, y) followed by
+// add_bf16(g, dx0): bn_bwd applies the ReLU mask itself (relu=1 reads y
+// anyway), so g was pure data movement — 2 passes (4 reads 2 writes) of
+// the block activation become 1 (3 reads 1 write).
+__global__ void add_relu_bwd_add_k(const ushort8 *__restrict__ dy,
+                                   const ushort8 *__restrict__ y,
+                                   const ushort8 *__restrict__ dx0,
+                                   ushort8 *__restrict__ dxt, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8 vd = dy[i], vy = y[i], v0 = dx0[i];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = bf2f(vy[j]) > 0.f ? bf2f(vd[j]) : 0.f;
+      o[j] = f2bf(g + bf2f(v0[j]));
+    }
+    dxt[i] = o;
+  }
+}
+
+extern "C" hipError_t add_relu_bwd_add(const void *dy, const void *y,
+                                       const void *dx0, void *dxt, long n,
+                                       hipStream_t s) {
+  long n8 = n / 8;
+  add_relu_bwd_add_k<<<ew_grid(n8), 256, 0, s>>>(
+      (const ushort8 *)dy, (const ushort8 *)y, (const ushort8 *)dx0,
+      (ushort8 *)dxt, n8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
 extern "C" hipError_t add_relu_bwd(const void *dy, const void *y, void *dx,
                                    long n, hipStream_t s) {
   long n8 = n / 8;

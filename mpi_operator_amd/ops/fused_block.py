@@ -93,10 +93,10 @@ class BottleneckFn(torch.autograd.Function):
             ad, md, vd, wd, gd = ctx.saved_tensors[19:]
 
         dout = dout.contiguous(memory_format=torch.channels_last)
-        # through the join's relu: g = dout * (out > 0) — also the skip grad
-        g = ext.add_relu_bwd(dout, out)
-        # bn3 (relu already applied via g → relu=False here)
-        dx3, dg3, db3 = ext.bn_bwd(g, a3, out, g3, m3, v3, False)
+        # The join's ReLU mask (out > 0) is applied INSIDE each consumer
+        # (bn_bwd's relu path reads `out` anyway) — the old materialized
+        # g = dout·(out>0) was pure data movement (2 extra passes/block).
+        dx3, dg3, db3 = ext.bn_bwd(dout, a3, out, g3, m3, v3, True)
         dw3 = ext.conv2d_wgrad(y2, dx3, 1, 1, 1, 0)
         dy2 = ext.conv2d_dgrad(dx3, w3, y2.shape[2], y2.shape[3], 1, 0)
         dx2, dg2, db2 = ext.bn_bwd(dy2, a2, y2, g2, m2, v2, True)
@@ -105,18 +105,18 @@ class BottleneckFn(torch.autograd.Function):
         dx1, dg1, db1 = ext.bn_bwd(dy1, a1, y1, g1, m1, v1, True)
         dw1 = ext.conv2d_wgrad(x, dx1, 1, 1, 1, 0)
 
+        dx0 = ext.conv2d_dgrad(dx1, w1, x.shape[2], x.shape[3], 1, 0)
         if ctx.has_ds:
-            dad, dgd, dbd = ext.bn_bwd(g, ad, ad, gd, md, vd, False)
+            # mask from the join output (the downsample BN itself has no
+            # ReLU): dy_ds = dout·(out>0), applied inside bn_bwd
+            dad, dgd, dbd = ext.bn_bwd(dout, ad, out, gd, md, vd, True)
             dwd = ext.conv2d_wgrad(x, dad, 1, 1, st, 0)
             skip = ext.conv2d_dgrad(dad, wd, x.shape[2], x.shape[3], st, 0)
+            dxt = ext.add_bf16(skip, dx0)
         else:
             dwd = dgd = dbd = None
-            skip = g
-        # residual-join gradient sum: skip grad + conv1's input grad (one
-        # vectorized kernel; an in-epilogue accumulate measured slower —
-        # the RMW on one tensor serializes the epilogue on aliasing)
-        dx0 = ext.conv2d_dgrad(dx1, w1, x.shape[2], x.shape[3], 1, 0)
-        dxt = ext.add_bf16(skip, dx0)
+            # one-pass join: dxt = dx0 + dout·(out>0)
+            dxt = ext.add_relu_bwd_add(dout, out, dx0)
 
         f32 = torch.float32
         return (dxt, None, None, None,

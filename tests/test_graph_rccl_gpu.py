@@ -40,8 +40,9 @@ def test_graph_capture_with_rccl_allreduce():
         for _ in range(3):
             g.replay()
         torch.cuda.synchronize()
-        # 1 warm allreduce ×2 (sum, world 1 = identity), then 3 replays of ×2
-        assert torch.allclose(t, torch.full_like(t, 2.0 ** 4))
+        # capture records without executing; world-1 all_reduce is identity:
+        # 3 replays of ×2 → 8
+        assert torch.allclose(t, torch.full_like(t, 8.0))
     finally:
         dist.destroy_process_group()
 
