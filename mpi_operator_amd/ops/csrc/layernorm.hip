@@ -23,9 +23,14 @@ __global__ void ln_fwd_k(const ushort8 *__restrict__ x,
        row += (long)gridDim.x * waves) {
     const ushort8 *xr = x + row * C8;
     float s = 0.f, sq = 0.f;
+    ushort8 cx[2]; // row cached in registers (2 octets/lane at N<=1024):
+                   // the stats pass otherwise re-reads the whole row
     for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
+      ushort8 v = xr[c];
+      if (ci < 2) cx[ci] = v;
       float f[8];
-      bf8_to_f8(xr[c], f);
+      bf8_to_f8(v, f);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         s += f[j];
@@ -43,8 +48,9 @@ __global__ void ln_fwd_k(const ushort8 *__restrict__ x,
     }
     ushort8 *yr = y + row * C8;
     for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
       float f[8];
-      bf8_to_f8(xr[c], f);
+      bf8_to_f8(ci < 2 ? cx[ci] : xr[c], f);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         f[j] = (f[j] - mu) * rs * gamma[c * 8 + j] + beta[c * 8 + j];
@@ -75,11 +81,19 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
     const ushort8 *dr = dy + row * C8;
     float mu = mean[row], rs = rstd[row];
     float s1 = 0.f, s2 = 0.f;
+    ushort8 cx[2], cd[2]; // row cached (2 octets/lane at N<=1024): the dx
+                          // pass was re-reading x AND dy — the measured gap
+                          // that kept fused LN behind torch native
     for (int c = lane; c < C8; c += 64) {
-      float fx[8], fd[8];
-      bf8_to_f8(xr[c], fx);
-      bf8_to_f8(dr[c], fd);
       int ci = (c - lane) >> 6;
+      ushort8 vx = xr[c], vd = dr[c];
+      if (ci < 2) {
+        cx[ci] = vx;
+        cd[ci] = vd;
+      }
+      float fx[8], fd[8];
+      bf8_to_f8(vx, fx);
+      bf8_to_f8(vd, fd);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xh = (fx[j] - mu) * rs;
@@ -96,9 +110,10 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
     s2 = wave_sum(s2) * inv_n;
     ushort8 *dxr = dx + row * C8;
     for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
       float fx[8], fd[8];
-      bf8_to_f8(xr[c], fx);
-      bf8_to_f8(dr[c], fd);
+      bf8_to_f8(ci < 2 ? cx[ci] : xr[c], fx);
+      bf8_to_f8(ci < 2 ? cd[ci] : dr[c], fd);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xh = (fx[j] - mu) * rs;
