@@ -127,8 +127,15 @@ class BertLayer(nn.Module):
         self.fc2 = BertLinear(cfg.intermediate, cfg.hidden)
         self.ln2 = LayerNorm(cfg.hidden, eps=cfg.eps)
 
+    def _join_ln(self, ln, a, b):
+        """residual + LN; fused single-pass kernel when enabled."""
+        if _FUSED_LN and a.is_cuda and a.dtype == torch.bfloat16:
+            return Fx.layer_norm_add(a, b, ln.weight.float(), ln.bias.float(),
+                                     ln.eps)
+        return ln(a + b)
+
     def forward(self, x, attn_mask=None):
-        x = self.ln1(x + self.attn(x, attn_mask))
+        x = self._join_ln(self.ln1, x, self.attn(x, attn_mask))
         if x.is_cuda and x.dtype == torch.bfloat16:
             # fused FFN: GELU lives in the GEMM epilogues (fwd emits
             # h_pre + gelu(h); bwd's fc2-dx multiplies by gelu'(h_pre))
@@ -138,7 +145,7 @@ class BertLayer(nn.Module):
                        self.fc2.weight, self.fc2.bias).view(b, s, h)
         else:
             y = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
-        return self.ln2(x + y)
+        return self._join_ln(self.ln2, x, y)
 
 
 class BertEmbeddings(nn.Module):

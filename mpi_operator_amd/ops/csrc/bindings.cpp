@@ -145,6 +145,9 @@ hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
 hipError_t mfma_probe(const void *, const void *, float *, hipStream_t);
 hipError_t ln_fwd(const void *, const float *, const float *, void *, float *,
                   float *, long, int, float, hipStream_t);
+hipError_t ln_fwd_add(const void *, const void *, const float *,
+                      const float *, void *, void *, float *, float *, long,
+                      int, float, hipStream_t);
 hipError_t ln_bwd(const void *, const void *, const float *, const float *,
                   const float *, void *, float *, float *, long, int, int *,
                   hipStream_t);
@@ -452,6 +455,27 @@ static std::vector<Tensor> layernorm_fwd(const Tensor &x, const Tensor &gamma,
              y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), M,
              N, (float)eps, cur_stream()));
   return {y, mean, rstd};
+}
+
+// y = LN(a + b) with the sum materialized for backward (one pass instead
+// of a separate residual add)
+static std::vector<Tensor> layernorm_add_fwd(const Tensor &a, const Tensor &b,
+                                             const Tensor &gamma,
+                                             const Tensor &beta, double eps) {
+  const HIPDeviceGuard guard(a.device());
+  Tensor ac = a.contiguous(), bc = b.contiguous();
+  TORCH_CHECK(ac.sizes() == bc.sizes());
+  long M = ac.numel() / ac.size(-1);
+  int N = ac.size(-1);
+  auto f32 = a.options().dtype(at::kFloat);
+  Tensor sum = at::empty_like(ac);
+  Tensor y = at::empty_like(ac);
+  Tensor mean = at::empty({M}, f32), rstd = at::empty({M}, f32);
+  CHK(ln_fwd_add(ac.data_ptr(), bc.data_ptr(), gamma.data_ptr<float>(),
+                 beta.data_ptr<float>(), sum.data_ptr(), y.data_ptr(),
+                 mean.data_ptr<float>(), rstd.data_ptr<float>(), M, N,
+                 (float)eps, cur_stream()));
+  return {y, sum, mean, rstd};
 }
 
 static std::vector<Tensor> layernorm_bwd(const Tensor &dy, const Tensor &x,
@@ -763,6 +787,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gap_bwd", &gap_bwd_b);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_add_fwd", &layernorm_add_fwd);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);

@@ -161,6 +161,70 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
   }
 }
 
+// Residual-join + LayerNorm in one pass: s = a + b (written — backward
+// reads it as the LN input), y = LN(s). Replaces the separate residual
+// add's extra read+write of the [M][N] activation (BERT runs 2 joins per
+// layer per direction).
+__global__ void ln_fwd_add_k(const ushort8 *__restrict__ a,
+                             const ushort8 *__restrict__ b,
+                             const float *__restrict__ gamma,
+                             const float *__restrict__ beta,
+                             ushort8 *__restrict__ sum_out,
+                             ushort8 *__restrict__ y, float *__restrict__ mean,
+                             float *__restrict__ rstd, long M, int C8,
+                             float eps) {
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  int waves = blockDim.x >> 6;
+  int N = C8 * 8;
+  float inv_n = 1.f / (float)N;
+  for (long row = (long)blockIdx.x * waves + wave; row < M;
+       row += (long)gridDim.x * waves) {
+    const ushort8 *ar = a + row * C8;
+    const ushort8 *br = b + row * C8;
+    ushort8 *sr = sum_out + row * C8;
+    float s = 0.f, sq = 0.f;
+    ushort8 cs[2];
+    for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
+      float fa[8], fb[8];
+      bf8_to_f8(ar[c], fa);
+      bf8_to_f8(br[c], fb);
+      ushort8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        // the SAVED sum is bf16-rounded; stats must match what backward
+        // will read, so accumulate on the rounded value
+        v[j] = f2bf(fa[j] + fb[j]);
+        float f = bf2f(v[j]);
+        s += f;
+        sq += f * f;
+      }
+      sr[c] = v;
+      if (ci < 2) cs[ci] = v;
+    }
+    s = wave_sum(s);
+    sq = wave_sum(sq);
+    float mu = s * inv_n;
+    float var = fmaxf(sq * inv_n - mu * mu, 0.f);
+    float rs = rsqrtf(var + eps);
+    if (lane == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    ushort8 *yr = y + row * C8;
+    for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
+      float f[8];
+      bf8_to_f8(ci < 2 ? cs[ci] : sr[c], f);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        f[j] = (f[j] - mu) * rs * gamma[c * 8 + j] + beta[c * 8 + j];
+      yr[c] = f8_to_bf8(f);
+    }
+  }
+}
+
 static int ln_grid(long M, int waves) {
   long g = (M + waves - 1) / waves;
   if (g > 1024) g = 1024;
@@ -176,6 +240,19 @@ extern "C" hipError_t ln_fwd(const void *x, const float *gamma,
   ln_fwd_k<<<ln_grid(M, 4), 256, 0, s>>>((const ushort8 *)x, gamma, beta,
                                          (ushort8 *)y, mean, rstd, M, N / 8,
                                          eps);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
+extern "C" hipError_t ln_fwd_add(const void *a, const void *b,
+                                 const float *gamma, const float *beta,
+                                 void *sum_out, void *y, float *mean,
+                                 float *rstd, long M, int N, float eps,
+                                 hipStream_t s) {
+  if (N % 8) return hipErrorInvalidValue;
+  ln_fwd_add_k<<<ln_grid(M, 4), 256, 0, s>>>(
+      (const ushort8 *)a, (const ushort8 *)b, gamma, beta,
+      (ushort8 *)sum_out, (ushort8 *)y, mean, rstd, M, N / 8, eps);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

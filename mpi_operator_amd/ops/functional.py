@@ -402,3 +402,25 @@ class AttentionFn(torch.autograd.Function):
 
 def attention(qkv, heads, scale, mask=None):
     return AttentionFn.apply(qkv, heads, scale, mask)
+
+
+class LayerNormAddFn(torch.autograd.Function):
+    """y = LN(a + b): the transformer residual join fused into the LN
+    forward (the separate add was an extra full read+write of the [M, H]
+    activation per join). Backward: dLN flows identically to both addends."""
+
+    @staticmethod
+    def forward(ctx, a, b, weight, bias, eps: float):
+        y, s, mean, rstd = hip_ext().layernorm_add_fwd(a, b, weight, bias, eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        s, weight, mean, rstd = ctx.saved_tensors
+        dx, dg, db = hip_ext().layernorm_bwd(dy, s, weight, mean, rstd)
+        return dx, dx, dg, db, None
+
+
+def layer_norm_add(a, b, weight, bias, eps: float = 1e-12):
+    return LayerNormAddFn.apply(a, b, weight, bias, eps)

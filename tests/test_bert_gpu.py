@@ -219,3 +219,32 @@ def test_fused_attention_ragged_seq():
     p = torch.softmax(torch.matmul(q, k.transpose(-1, -2)) * 0.125, dim=-1)
     ref = torch.matmul(p, v).transpose(1, 2).reshape(B, S, H * D)
     assert (out.float() - ref).abs().max().item() < 0.02
+
+
+def test_layer_norm_add_matches_fp32():
+    """Fused residual+LN vs torch fp32, fwd + grads to both addends."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(17)
+    M, N = 384, 1024
+    a = (torch.randn(M, N, device="cuda") * 0.7).to(torch.bfloat16)
+    b = (torch.randn(M, N, device="cuda") * 0.7).to(torch.bfloat16)
+    w = torch.rand(N, device="cuda") + 0.5
+    bb = torch.randn(N, device="cuda") * 0.1
+    aa, ab = a.clone().requires_grad_(True), b.clone().requires_grad_(True)
+    wa, wb = w.clone().requires_grad_(True), bb.clone().requires_grad_(True)
+    y = Fx.layer_norm_add(aa, ab, wa, wb, 1e-12)
+    y.float().square().mean().backward()
+
+    # reference computes on the bf16-rounded sum (what the kernel stats see)
+    ra = (a.float() + b.float()).to(torch.bfloat16).float().requires_grad_(True)
+    rw = w.clone().requires_grad_(True)
+    rb = bb.clone().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(ra, (N,), rw, rb, 1e-12)
+    yr.square().mean().backward()
+
+    assert (y.float() - yr).abs().max().item() < 0.02
+    for g, r, name in ((aa.grad, ra.grad, "da"), (ab.grad, ra.grad, "db"),
+                       (wa.grad, rw.grad, "dw"), (wb.grad, rb.grad, "dbias")):
+        num = (g.float() - r).abs().max().item()
+        den = r.abs().max().item() + 1e-6
+        assert num < 0.08 * den + 1e-4, (name, num, den)
