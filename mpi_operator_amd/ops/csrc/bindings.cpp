@@ -487,9 +487,8 @@ static std::vector<Tensor> layernorm_bwd(const Tensor &dy, const Tensor &x,
   long M = xc.numel() / xc.size(-1);
   int N = xc.size(-1);
   auto f32 = x.options().dtype(at::kFloat);
-  long grid_max = (M + 3) / 4;
-  if (grid_max > 1024) grid_max = 1024;
-  Tensor partial = at::empty({grid_max, 2L * N}, f32);
+  // slab rows must cover ln_bwd's stats grid (≤512 by construction)
+  Tensor partial = at::empty({512, 2L * N}, f32);
   Tensor dgb = at::empty({2L * N}, f32);
   Tensor dx = at::empty_like(xc);
   CHK(ln_bwd(dyc.data_ptr(), xc.data_ptr(), gamma.data_ptr<float>(),

@@ -59,31 +59,29 @@ __global__ void ln_fwd_k(const ushort8 *__restrict__ x,
   }
 }
 
-// dx = rstd * (g*dy - mean(g*dy) - xhat * mean(g*dy*xhat));
-// partial[block][0][N] += dy*xhat (dgamma), partial[block][1][N] += dy (dbeta)
-__global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
-                         const ushort8 *__restrict__ x,
-                         const float *__restrict__ gamma,
-                         const float *__restrict__ mean,
-                         const float *__restrict__ rstd,
-                         ushort8 *__restrict__ dx,
-                         float *__restrict__ partial, long M, int C8) {
+// dx = rstd * (g*dy - mean(g*dy) - xhat * mean(g*dy*xhat)).
+// SLAB-FREE: the round-1 kernel folded dgamma/dbeta slabs at block end,
+// which capped the grid at 32 blocks (512 waves) and left the dx pass
+// latency-bound — the measured reason fused LN lost to torch native.
+// dgamma/dbeta moved to the column-parallel kernel below.
+__global__ void ln_bwd_dx_k(const ushort8 *__restrict__ dy,
+                            const ushort8 *__restrict__ x,
+                            const float *__restrict__ gamma,
+                            const float *__restrict__ mean,
+                            const float *__restrict__ rstd,
+                            ushort8 *__restrict__ dx, long M, int C8) {
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
   int waves = blockDim.x >> 6;
   int N = C8 * 8;
   float inv_n = 1.f / (float)N;
-  // per-thread fp32 col partials (up to 4 octets per lane at N<=2048)
-  float pg[4][8] = {}, pb[4][8] = {};
   for (long row = (long)blockIdx.x * waves + wave; row < M;
        row += (long)gridDim.x * waves) {
     const ushort8 *xr = x + row * C8;
     const ushort8 *dr = dy + row * C8;
     float mu = mean[row], rs = rstd[row];
     float s1 = 0.f, s2 = 0.f;
-    ushort8 cx[2], cd[2]; // row cached (2 octets/lane at N<=1024): the dx
-                          // pass was re-reading x AND dy — the measured gap
-                          // that kept fused LN behind torch native
+    ushort8 cx[2], cd[2]; // row cached in registers at N<=1024
     for (int c = lane; c < C8; c += 64) {
       int ci = (c - lane) >> 6;
       ushort8 vx = xr[c], vd = dr[c];
@@ -100,10 +98,6 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
         float gd = gamma[c * 8 + j] * fd[j];
         s1 += gd;
         s2 += gd * xh;
-        if (ci < 4) {
-          pg[ci][j] += fd[j] * xh;
-          pb[ci][j] += fd[j];
-        }
       }
     }
     s1 = wave_sum(s1) * inv_n;
@@ -122,42 +116,58 @@ __global__ void ln_bwd_k(const ushort8 *__restrict__ dy,
       dxr[c] = f8_to_bf8(fd);
     }
   }
-  // fold this block's per-thread partials into its slab rows via LDS
-  __shared__ float lds[1024 * 8]; // sized for the 16-wave backward block
-  float *slab_g = partial + (long)blockIdx.x * 2 * N;
-  float *slab_b = slab_g + N;
+}
+
+// dgamma/dbeta partials, bn_partials-style: thread owns a channel octet,
+// strides rows (coalesced 16 B/lane), per-row mean/rstd scalar loads;
+// [grid][2][N] fp32 slabs reduced by splitk_reduce.
+__global__ void ln_gb_partials_k(const ushort8 *__restrict__ dy,
+                                 const ushort8 *__restrict__ x,
+                                 const float *__restrict__ mean,
+                                 const float *__restrict__ rstd,
+                                 float *__restrict__ partial, long M, int C8) {
+  int C = C8 * 8;
+  int cb = threadIdx.x % C8;
+  int row_lane = threadIdx.x / C8;
+  int rows_per_block = blockDim.x / C8;
+  float a0[8] = {0}, a1[8] = {0}; // dbeta, dgamma
+  if (row_lane < rows_per_block) {
+    long stride = (long)gridDim.x * rows_per_block;
+    for (long row = (long)blockIdx.x * rows_per_block + row_lane; row < M;
+         row += stride) {
+      long off = row * C8 + cb;
+      float mu = mean[row], rs = rstd[row];
+      float fx[8], fd[8];
+      bf8_to_f8(x[off], fx);
+      bf8_to_f8(dy[off], fd);
 #pragma unroll
-  for (int ci = 0; ci < 4; ++ci) {
-    int c = ci * 64 + lane;
-    // reduce across the block's waves one octet-bank at a time
-#pragma unroll
-    for (int j = 0; j < 8; ++j) lds[threadIdx.x * 8 + j] = pg[ci][j];
-    __syncthreads();
-    if (wave == 0 && c < C8) {
-      float acc[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] = lds[lane * 8 + j];
-      for (int w = 1; w < waves; ++w)
-#pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] += lds[(w * 64 + lane) * 8 + j];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) slab_g[c * 8 + j] = acc[j];
+      for (int j = 0; j < 8; ++j) {
+        a0[j] += fd[j];
+        a1[j] += fd[j] * ((fx[j] - mu) * rs);
+      }
     }
-    __syncthreads();
+  }
+  __shared__ float lds[2][256 * 8];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) lds[threadIdx.x * 8 + j] = pb[ci][j];
-    __syncthreads();
-    if (wave == 0 && c < C8) {
-      float acc[8];
+  for (int j = 0; j < 8; ++j) {
+    lds[0][threadIdx.x * 8 + j] = a1[j]; // dgamma first (slab order)
+    lds[1][threadIdx.x * 8 + j] = a0[j];
+  }
+  __syncthreads();
+  if (row_lane == 0) {
+    for (int rl = 1; rl < rows_per_block; ++rl)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] = lds[lane * 8 + j];
-      for (int w = 1; w < waves; ++w)
+      for (int j = 0; j < 8; ++j) {
+        a1[j] += lds[0][(rl * C8 + cb) * 8 + j];
+        a0[j] += lds[1][(rl * C8 + cb) * 8 + j];
+      }
+    float *pg = partial + (long)blockIdx.x * 2 * C + cb * 8;
+    float *pb = pg + C;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] += lds[(w * 64 + lane) * 8 + j];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) slab_b[c * 8 + j] = acc[j];
+    for (int j = 0; j < 8; ++j) {
+      pg[j] = a1[j];
+      pb[j] = a0[j];
     }
-    __syncthreads();
   }
 }
 
@@ -260,24 +270,26 @@ extern "C" hipError_t ln_fwd_add(const void *a, const void *b,
 extern "C" hipError_t splitk_reduce(const float *, int, long, void *, int,
                                     hipStream_t);
 
-// partial must hold [grid][2][N] fp32; dgamma/dbeta are fp32 [N] each,
-// reduced here via splitk_reduce over the slab
+// partial must hold [grid][2][N] fp32 (grid from *grid_out contract:
+// caller passes the allocation's row count); dgamma/dbeta are fp32 [N]
+// each, reduced via splitk_reduce over the slab
 extern "C" hipError_t ln_bwd(const void *dy, const void *x, const float *gamma,
                              const float *mean, const float *rstd, void *dx,
                              float *partial, float *dgamma_dbeta, long M,
                              int N, int *grid_out, hipStream_t s) {
   if (N % 8 || N > 2048) return hipErrorInvalidValue;
-  // ≤32 slab rows (16-wave blocks keep the dx pass at ~512 waves): the
-  // dgamma/dbeta splitk_reduce over [grid][2N] runs 2 blocks at this len —
-  // with the old 1024-slab grid it serialized 512-deep per element and was
-  // 23% of a fused-LN BERT step (profiled; fused LN lost to torch because
-  // of THIS kernel, not the LN passes themselves)
-  int grid = ln_grid(M, 16);
-  if (grid > 32) grid = 32;
+  int C8 = N / 8;
+  ln_bwd_dx_k<<<ln_grid(M, 4), 256, 0, s>>>(
+      (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
+      (ushort8 *)dx, M, C8);
+  HIP_KERNEL_CHECK();
+  int rpb = 256 / C8;
+  long g = (M + rpb - 1) / rpb;
+  int grid = (int)(g > 512 ? 512 : (g < 1 ? 1 : g));
   if (grid_out) *grid_out = grid;
-  ln_bwd_k<<<grid, 1024, 0, s>>>((const ushort8 *)dy, (const ushort8 *)x,
-                                 gamma, mean, rstd, (ushort8 *)dx, partial, M,
-                                 N / 8);
+  ln_gb_partials_k<<<grid, 256, 0, s>>>((const ushort8 *)dy,
+                                        (const ushort8 *)x, mean, rstd,
+                                        partial, M, C8);
   HIP_KERNEL_CHECK();
   // dgamma_dbeta[0..N) = dgamma, [N..2N) = dbeta
   return splitk_reduce(partial, grid, 2L * N, dgamma_dbeta, 0, s);
