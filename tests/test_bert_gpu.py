@@ -248,3 +248,22 @@ def test_layer_norm_add_matches_fp32():
         num = (g.float() - r).abs().max().item()
         den = r.abs().max().item() + 1e-6
         assert num < 0.08 * den + 1e-4, (name, num, den)
+
+
+def test_ffn_fused_gelu_pipe256_route():
+    """Shape that routes fc1 through the 256²-tile writer kernel
+    (M%256==0, I%256==0, nwg>=128): numerics vs fp32 torch."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(19)
+    M, H, I = 4096, 256, 4096
+    x = (torch.randn(M, H, device="cuda") * 0.5).to(torch.bfloat16)
+    w1 = (torch.randn(I, H, device="cuda") * 0.05).to(torch.bfloat16)
+    b1 = torch.randn(I, device="cuda").to(torch.bfloat16) * 0.1
+    w2 = (torch.randn(H, I, device="cuda") * 0.05).to(torch.bfloat16)
+    b2 = torch.randn(H, device="cuda").to(torch.bfloat16) * 0.1
+    y = Fx.ffn(x, w1, b1, w2, b2)
+    h = torch.nn.functional.linear(x.float(), w1.float(), b1.float())
+    g = torch.nn.functional.gelu(h, approximate="tanh")
+    yr = torch.nn.functional.linear(g, w2.float(), b2.float())
+    assert torch.allclose(y.float(), yr, rtol=0.05, atol=0.05), \
+        (y.float() - yr).abs().max()
