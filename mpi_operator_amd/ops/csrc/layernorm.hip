@@ -235,6 +235,24 @@ __global__ void ln_fwd_add_k(const ushort8 *__restrict__ a,
   }
 }
 
+// Per-channel slab reduce: one 256-thread block per channel strides the
+// grid entries (splitk_reduce at len 2·N with 512 slabs ran 4 blocks ×
+// 512-deep serial — 35 µs/call and 17% of a fused-LN BERT step).
+__global__ void ln_gb_reduce_k(const float *__restrict__ partial,
+                               float *__restrict__ out, int grid, int N2) {
+  int c = blockIdx.x;
+  if (c >= N2) return;
+  float a = 0.f;
+  for (int g = threadIdx.x; g < grid; g += 256)
+    a += partial[(long)g * N2 + c];
+  __shared__ float red[256 / WAVE];
+  a = wave_sum(a);
+  if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = a;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    out[c] = red[0] + red[1] + red[2] + red[3];
+}
+
 static int ln_grid(long M, int waves) {
   long g = (M + waves - 1) / waves;
   if (g > 1024) g = 1024;
@@ -292,5 +310,7 @@ extern "C" hipError_t ln_bwd(const void *dy, const void *x, const float *gamma,
                                         partial, M, C8);
   HIP_KERNEL_CHECK();
   // dgamma_dbeta[0..N) = dgamma, [N..2N) = dbeta
-  return splitk_reduce(partial, grid, 2L * N, dgamma_dbeta, 0, s);
+  ln_gb_reduce_k<<<2 * N, 256, 0, s>>>(partial, dgamma_dbeta, grid, 2 * N);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
 }
