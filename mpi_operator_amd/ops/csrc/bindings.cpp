@@ -101,6 +101,8 @@ hipError_t masked_xent_bwd_launch(const void *, const long *, const float *,
                                   int, long, hipStream_t);
 hipError_t attn_fwd_launch(const void *, void *, void *, const float *, int,
                            int, int, float, hipStream_t);
+hipError_t attn_bwd_launch(const void *, const void *, const void *, void *,
+                           int, int, int, float, hipStream_t);
 hipError_t softmax_xent_bwd_launch(const float *, const long *, void *, int,
                                    int, const float *, hipStream_t);
 hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
@@ -612,6 +614,20 @@ static std::vector<Tensor> attn_fwd(const Tensor &qkv, int64_t heads,
   return {ctx, probs};
 }
 
+static Tensor attn_bwd(const Tensor &qkv, const Tensor &dctx,
+                       const Tensor &probs, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == at::kBFloat16);
+  const HIPDeviceGuard guard(qkv.device());
+  int B = qkv.size(0), S = qkv.size(1), H = qkv.size(3);
+  TORCH_CHECK(S == 128 && qkv.size(4) == 64, "fused attn bwd: S=128, D=64");
+  TORCH_CHECK(qkv.is_contiguous() && probs.is_contiguous());
+  Tensor dc = dctx.contiguous();
+  Tensor dqkv = at::empty_like(qkv);
+  CHK(attn_bwd_launch(qkv.data_ptr(), dc.data_ptr(), probs.data_ptr(),
+                      dqkv.data_ptr(), B, S, H, (float)scale, cur_stream()));
+  return dqkv;
+}
+
 // ------------------------- fused FFN (GELU) -------------------------
 // fc1 forward with the bias+GELU in the epilogue: returns (g, h_pre)
 static std::vector<Tensor> linear_gelu_fwd(const Tensor &x, const Tensor &w,
@@ -793,6 +809,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("masked_xent_fwd", &masked_xent_fwd);
   m.def("linear_gelu_fwd", &linear_gelu_fwd);
   m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.def("linear_wgrad_only", &linear_wgrad_only);
   m.def("linear_gelu_dgrad", &linear_gelu_dgrad);
   m.def("masked_xent_bwd", &masked_xent_bwd);

@@ -385,6 +385,11 @@ class AttentionFn(torch.autograd.Function):
     def backward(ctx, dout):
         qkv, probs = ctx.saved_tensors
         B, S, _, H, D = qkv.shape
+        if S == 128 and D == 64:
+            # fused backward: dP/dS/dQ/dK/dV in one kernel per (b, h),
+            # written straight into the dqkv layout
+            return (hip_ext().attn_bwd(qkv, dout, probs, ctx.scale),
+                    None, None, None)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # B,H,S,D
         do = dout.view(B, S, H, D).transpose(1, 2)
         p = probs.float()

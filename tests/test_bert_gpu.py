@@ -267,3 +267,25 @@ def test_ffn_fused_gelu_pipe256_route():
     yr = torch.nn.functional.linear(g, w2.float(), b2.float())
     assert torch.allclose(y.float(), yr, rtol=0.05, atol=0.05), \
         (y.float() - yr).abs().max()
+
+
+def test_fused_attention_backward_matches_torch():
+    """The fused S=128 attention backward vs the fp32 torch chain."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(23)
+    B, S, H, D = 3, 128, 4, 64
+    qkv = (torch.randn(B, S, 3, H, D, device="cuda") * 0.5).to(torch.bfloat16)
+    a = qkv.clone().requires_grad_(True)
+    out = Fx.attention(a, H, 1.0 / D ** 0.5)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    r = qkv.float().clone().requires_grad_(True)
+    q, k, v = (r[:, :, i].transpose(1, 2) for i in range(3))
+    p = torch.softmax(torch.matmul(q, k.transpose(-1, -2)) / D ** 0.5, dim=-1)
+    ctx_t = torch.matmul(p, v).transpose(1, 2).reshape(B, S, H * D)
+    ctx_t.backward(gout.float())
+
+    err = (a.grad.float() - r.grad).abs().max().item()
+    scale = r.grad.abs().max().item() + 1e-6
+    assert err < 0.08 * scale + 2e-3, (err, scale)
