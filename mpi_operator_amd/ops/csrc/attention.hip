@@ -264,10 +264,11 @@ __global__ __launch_bounds__(256) void attn_bwd_k(
   int b = bh / H, h = bh % H;
   int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
 
-  // tr-subtile images: [4 k][16 col] blocks, cq-fastest
-  __shared__ __align__(128) ushort8 p_tr[S * 8];   // k=sq, col=sk  (32 KB)
-  __shared__ __align__(128) ushort8 ds_tr[S * 8];  // k=sq, col=sk  (32 KB)
-  __shared__ __align__(128) ushort8 do_tr[S * 8];  // k=sq, col=d   (16 KB used)
+  // tr-subtile images: [4 k][16 col] blocks, cq-fastest.
+  // S×S images (P, dS) hold S*S/8 = S*16 slots; S×D images S*D/8 = S*8.
+  __shared__ __align__(128) ushort8 p_tr[S * 16];  // k=sq, col=sk  (32 KB)
+  __shared__ __align__(128) ushort8 ds_tr[S * 16]; // k=sq, col=sk  (32 KB)
+  __shared__ __align__(128) ushort8 do_tr[S * 8];  // k=sq, col=d   (16 KB)
   __shared__ __align__(128) ushort8 k_tr[S * 8];   // k=sk, col=d
   __shared__ __align__(128) ushort8 q_tr[S * 8];   // k=sq, col=d
   __shared__ __align__(128) ushort8 v_nt[S * 8];   // [sk][8 d-octets], swizzled
@@ -288,7 +289,7 @@ __global__ __launch_bounds__(256) void attn_bwd_k(
     v_nt[row * 8 + att_swz(q, row)] =
         *(const ushort8 *)(vp + (long)row * qrow + q * 8);
   }
-  for (int idx = tid; idx < S * 4; idx += 256) {
+  for (int idx = tid; idx < S * 8; idx += 256) {
     int st = idx >> 3, kq = st >> 2, cq = st & 3;
     int kl = (idx & 7) >> 1, ch = idx & 1;
     int r = kq * 4 + kl, d0 = cq * 16 + ch * 8;
@@ -296,7 +297,7 @@ __global__ __launch_bounds__(256) void attn_bwd_k(
     k_tr[idx] = *(const ushort8 *)(kp + (long)r * qrow + d0);
     do_tr[idx] = *(const ushort8 *)(dop + (long)r * crow + d0);
   }
-  for (int idx = tid; idx < S * 8; idx += 256) {
+  for (int idx = tid; idx < S * 16; idx += 256) {
     int st = idx >> 3, kq = st >> 3, cq = st & 7;
     int kl = (idx & 7) >> 1, ch = idx & 1;
     int sq = kq * 4 + kl, sk0 = cq * 16 + ch * 8;
