@@ -639,6 +639,13 @@ int main(int argc, char **argv) {
     // wgrad-class: dw[256][2304] over M=200k pixels (ResNet stage3 3x3
     // runs as gather; this is the same GEMM shape with plain operands);
     // 1x1 wgrad dw[512][2048]; linear dw/dx (BERT-Large bs32)
+    // BERT dw split sweep: prologue share vs chip fill at kt = nk/splits
+    bench_tn(0, 1024, 1024, 4096, 2, 25);
+    bench_tn(0, 1024, 1024, 4096, 4, 25);
+    bench_tn(0, 1024, 1024, 4096, 8, 25);
+    bench_tn(0, 1024, 1024, 4096, 16, 25);
+    bench_tn(0, 1024, 4096, 4096, 2, 25);
+    bench_tn(0, 1024, 4096, 4096, 4, 25);
     bench_tn(0, 256, 2304, 200704, 8, 10);
     bench_tn(0, 512, 2048, 50176, 8, 10);
     bench_tn(0, 1024, 1024, 4096, 1, 25);
