@@ -124,7 +124,7 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
   // row softmax in fp32. value (row, col): reg r of frag ni belongs to
   // row = qrow0 + (r&3) + 8*(r>>2) + 4*(lane>>5), col = ni*32 + (lane&31):
   // a row's values live in one half-wave at fixed r across frags.
-  float mrow[16], srow[16];
+  float srow[16];
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     float m = -3.4e38f;
@@ -153,7 +153,6 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
 #pragma unroll
     for (int off = 16; off > 0; off >>= 1)
       s += __shfl_xor(s, off, 64);
-    mrow[r] = m;
     srow[r] = 1.f / s;
   }
 

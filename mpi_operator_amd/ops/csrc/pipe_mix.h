@@ -46,6 +46,8 @@
 // dgrad and linear dw/dx shapes — as MI355X-native code.
 #pragma once
 
+#include <mutex>
+
 // included from mix_gemm.h after mfma_tile.h; needs LinearWriter etc.
 
 typedef __attribute__((ext_vector_type(2))) unsigned int uint2v_pm;
@@ -513,14 +515,20 @@ __global__ __launch_bounds__(PM_THREADS) void pipe_mix_k(
 // 16-B zeros page for padding/ragged lanes (per-device, lazily allocated;
 // glds must read SOMETHING for the counted-vmcnt contract to hold).
 inline const uint16_t *pm_zeros_page() {
+  // process-lifetime allocation (256 B/device, never freed by design);
+  // guarded: concurrent first calls from two host threads must not leak
+  static std::mutex mu;
   static uint16_t *pages[64] = {};
   int dev = 0;
   hipGetDevice(&dev);
   if (!pages[dev]) {
-    void *p = nullptr;
-    if (hipMalloc(&p, 256) != hipSuccess) return nullptr;
-    hipMemset(p, 0, 256);
-    pages[dev] = (uint16_t *)p;
+    std::lock_guard<std::mutex> lk(mu);
+    if (!pages[dev]) {
+      void *p = nullptr;
+      if (hipMalloc(&p, 256) != hipSuccess) return nullptr;
+      hipMemset(p, 0, 256);
+      pages[dev] = (uint16_t *)p;
+    }
   }
   return pages[dev];
 }
