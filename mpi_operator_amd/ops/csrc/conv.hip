@@ -478,16 +478,21 @@ extern "C" hipError_t conv_dgrad(const void *dy, const void *w, void *dx,
   return splitk_reduce(partial, splits, M * C, dx, 1, strm);
 }
 
-// float4 lanes + two split-accumulators: the scalar version issued one
-// dependent 4 B load per split per element and sat 93% wave-parked.
-template <bool OUT_BF16>
+// float4 lanes; SPLITS is a template parameter for the common counts so
+// the accumulation loop FULLY UNROLLS — the runtime-splits version issued
+// its slab loads as a 2-chain serial loop and sat 73% wave-parked (PMC),
+// 4.7x off bandwidth at the BERT dw shapes.
+template <bool OUT_BF16, int SPLITS = 0>
 __global__ void splitk_reduce_k(const float4v *__restrict__ partial,
-                                int splits, long len4, void *__restrict__ out) {
+                                int splits_rt, long len4,
+                                void *__restrict__ out) {
+  const int splits = SPLITS ? SPLITS : splits_rt;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < len4;
        i += (long)gridDim.x * blockDim.x) {
     float4v a = partial[i];
     float4v b = {0.f, 0.f, 0.f, 0.f};
     int s = 1;
+#pragma unroll
     for (; s + 1 < splits; s += 2) {
       a += partial[(long)s * len4 + i];
       b += partial[(long)(s + 1) * len4 + i];
@@ -509,12 +514,14 @@ extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
   long len4 = len / 4; // len = Kout*RSC, both %8 ⇒ %4
   long blocks = (len4 + 255) / 256;
   if (blocks > 2048) blocks = 2048;
+  const float4v *p4 = (const float4v *)partial;
+  int nb = (int)blocks;
+#define SKR_DISPATCH(BF)                                                        do {                                                                            switch (splits) {                                                             case 2: splitk_reduce_k<BF, 2><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 3: splitk_reduce_k<BF, 3><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 4: splitk_reduce_k<BF, 4><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 5: splitk_reduce_k<BF, 5><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 6: splitk_reduce_k<BF, 6><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 8: splitk_reduce_k<BF, 8><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     case 16: splitk_reduce_k<BF, 16><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     default: splitk_reduce_k<BF, 0><<<nb, 256, 0, s>>>(p4, splits, len4, out); break;     }                                                                           } while (0)
   if (out_bf16)
-    splitk_reduce_k<true><<<(int)blocks, 256, 0, s>>>((const float4v *)partial,
-                                                      splits, len4, out);
+    SKR_DISPATCH(true);
   else
-    splitk_reduce_k<false><<<(int)blocks, 256, 0, s>>>((const float4v *)partial,
-                                                       splits, len4, out);
+    SKR_DISPATCH(false);
+#undef SKR_DISPATCH
   return hipGetLastError();
 }
 

@@ -160,7 +160,11 @@ extern "C" int gemm_tn_tn_splits(int M, int N, int K) {
   // nk>=32: at bs=8-class shapes (nk=16) the slab+reduce overhead beats
   // the fill gain (same-box 312/290 vs 325 seq/s); the bs=32 attn-out dw
   // (nk=64) is the stable +5% winner this path exists for
-  if (off || tiles >= 256 || nk < 32) return 1;
+  if (off || nk < 32) return 1;
+  // fc2-dw-class (tiles 256..511, deep K): 2-way split measured 672 vs
+  // 545-617 TF at sk4/sk1 on (1024,4096,4096) — 512 blocks = 2/CU
+  if (tiles >= 512) return 1;
+  if (tiles >= 256) return 2;
   long s = 512 / tiles;
   if (s > nk / 8) s = nk / 8;
   if (s > 64) s = 64;
