@@ -28,9 +28,11 @@ def _gemm_case(M, N, K, c_f32, seed=0):
     (64, 1000, 2048),   # classifier head shape (edge N)
     (200, 72, 136),     # every dim ragged (M,N edge; K%8==0 only)
     (512, 384, 576),    # conv-like K=9*64
-    (512, 256, 64),     # nt256 route: full tiles, 2-deep K pipeline
-    (768, 512, 96),     # nt256 route: odd K-tile count
-    (2048, 1024, 128),  # nt256 route: XCD-swizzled grid (nwg=32)
+    (512, 256, 64),     # full tiles below the 256-route occupancy floor
+    (768, 512, 96),     # odd K-tile count (128² pipeline path)
+    (2048, 1024, 128),  # XCD-swizzled 128² grid
+    (8192, 1024, 128),  # TRUE 256² pipe route: nwg=128 ≥ occupancy floor
+    (8192, 1024, 192),  # 256² route, odd K-tile count (K%64==0)
 ])
 @pytest.mark.parametrize("c_f32", [False, True])
 def test_gemm_nt_matches_matmul(shape, c_f32):
