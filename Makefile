@@ -18,7 +18,8 @@ test_gpu:         ## GPU tiers (numerics vs fp32 reference, models, bench path)
 bench:            ## 1-GPU flagship benchmark
 	$(PYTHON) bench.py --gpus 1 --steps 30 --warmup 10
 
-manifest:         ## regenerate deploy/v2beta1/mpi-operator.yaml
+manifest:         ## regenerate the CRD + deploy/v2beta1/mpi-operator.yaml
+	$(PYTHON) hack/gen_crd.py
 	hack/generate-manifest.sh
 
 wheel:            ## build the installable wheel (amdrun image ingredient)
