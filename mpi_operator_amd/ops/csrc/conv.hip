@@ -492,6 +492,22 @@ __global__ void splitk_reduce_k(const float4v *__restrict__ partial,
     float4v a = partial[i];
     float4v b = {0.f, 0.f, 0.f, 0.f};
     int s = 1;
+    if (SPLITS == 0) {
+      // runtime split count (wgrad runs 24..256): fixed 8-deep inner
+      // chunks keep 8 independent loads in flight — the plain runtime
+      // loop issued a 2-chain serial walk and sat 73% wave-parked
+      for (; s + 8 <= splits; s += 8) {
+        float4v c0 = {0.f, 0.f, 0.f, 0.f};
+        float4v c1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int j = 0; j < 8; j += 2) {
+          c0 += partial[(long)(s + j) * len4 + i];
+          c1 += partial[(long)(s + j + 1) * len4 + i];
+        }
+        a += c0;
+        b += c1;
+      }
+    }
 #pragma unroll
     for (; s + 1 < splits; s += 2) {
       a += partial[(long)s * len4 + i];
@@ -507,6 +523,31 @@ __global__ void splitk_reduce_k(const float4v *__restrict__ partial,
       ((float4v *)out)[i] = a;
     }
   }
+}
+
+// Per-column slab reduce for SHORT outputs with MANY chunks (colsum db:
+// len = N ≤ 4k, chunks up to 256 — the float4 kernel collapsed to one
+// block there): one 256-thread block per column, threads stride chunks.
+__global__ void slab_colreduce_k(const float *__restrict__ partial,
+                                 float *__restrict__ out, int chunks,
+                                 long N) {
+  long c = blockIdx.x;
+  if (c >= N) return;
+  float a = 0.f;
+  for (int g = threadIdx.x; g < chunks; g += 256)
+    a += partial[(long)g * N + c];
+  __shared__ float red[256 / WAVE];
+  a = wave_sum(a);
+  if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = a;
+  __syncthreads();
+  if (threadIdx.x == 0) out[c] = red[0] + red[1] + red[2] + red[3];
+}
+
+extern "C" hipError_t slab_colreduce(const float *partial, float *out,
+                                     int chunks, long N, hipStream_t s) {
+  slab_colreduce_k<<<(int)N, 256, 0, s>>>(partial, out, chunks, N);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
 }
 
 extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,

@@ -412,6 +412,8 @@ extern "C" int colsum_chunks(long M, int N) {
 extern "C" hipError_t splitk_reduce(const float *partial, int splits, long len,
                                     void *out, int out_bf16,
                                     hipStream_t s); // conv.hip
+extern "C" hipError_t slab_colreduce(const float *, float *, int, long,
+                                     hipStream_t); // conv.hip
 
 extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
                                   long M, int N, hipStream_t s) {
@@ -421,7 +423,9 @@ extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
     int gx = colsum_chunks(M, N);
     colsum8_k<<<dim3(gx, gy), 256, 0, s>>>((const ushort8 *)dy, partial, M, C8);
     HIP_KERNEL_CHECK();
-    return splitk_reduce(partial, gx, (long)N, db, 0, s);
+    // short output + many chunks: per-column reduce (the float4 kernel
+    // collapsed to one block at len=N)
+    return slab_colreduce(partial, db, gx, (long)N, s);
   }
   int gx = colsum_chunks(M, N);
   if (gx > 0) { // M-parallel ragged path (vocab-scale db)
@@ -430,9 +434,7 @@ extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
     colsum_ragged_k<<<grid, 256, 0, s>>>((const uint16_t *)dy, partial, M, N,
                                          rchunk);
     HIP_KERNEL_CHECK();
-    colsum_ragged_reduce_k<<<(N + 255) / 256, 256, 0, s>>>(partial, db, gx, N);
-    HIP_KERNEL_CHECK();
-    return hipSuccess;
+    return slab_colreduce(partial, db, gx, (long)N, s);
   }
   colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N);
   HIP_KERNEL_CHECK();
