@@ -541,13 +541,15 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
                     cur_stream()));
   // dw = dy^T @ x: both operands k-strided (k = batch row m) → fp32.
   // split-K when the [N][K] tile grid underfills the chip (K_reduce = M).
+  // The PADDED dyp (lda = Np) keeps the vocab-ragged head on the pipe
+  // route (tn_cols_ok's padded-stride contract).
   Tensor dw = at::empty({N, K}, f32);
   int dw_splits = gemm_tn_tn_splits(N, K, M);
   Tensor dwp = dw_splits > 1
                    ? at::empty({dw_splits, (long)N * K}, f32)
                    : dw;
-  CHK(gemm_tn_tn_sk(dyc.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
-                    dw.data_ptr(), N, K, M, N, K, K, dw_splits,
+  CHK(gemm_tn_tn_sk(dyp.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
+                    dw.data_ptr(), N, K, M, Np, K, K, dw_splits,
                     cur_stream()));
   Tensor db = at::zeros({N}, f32); // zeros: ragged-N fallback writes directly
   int chunks = colsum_chunks(M, N);

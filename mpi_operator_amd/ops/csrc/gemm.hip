@@ -7,6 +7,14 @@
 extern "C" hipError_t splitk_reduce(const float *, int, long, void *, int,
                                     hipStream_t); // conv.hip
 
+// A TN operand's 16-B column granules may overread up to 7 elements past
+// its `dim` columns; safe when the dim is %8, or when the ROW STRIDE is %8
+// and covers the rounded dim (the padded-allocation contract linear_bwd
+// uses for the ragged vocab head).
+static inline bool tn_cols_ok(int dim, long ld) {
+  return dim % 8 == 0 || (ld % 8 == 0 && ld >= ((dim + 7) & ~7));
+}
+
 extern "C" hipError_t gemm_nt(const void *a, const void *b, void *c, int M,
                               int N, int K, long lda, long ldb, long ldc,
                               int c_f32, hipStream_t s) {
@@ -73,7 +81,7 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
   // pipe TN staging loads 16-B column granules: a cols%8!=0 operand would
   // read past the allocation on its last k-row (BERT vocab 30522 class) —
   // those shapes keep the elementwise-tail mix path
-  if (use_pipemix() && N % 8 == 0) {
+  if (use_pipemix() && tn_cols_ok(N, ldb)) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
@@ -106,7 +114,7 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
-  if (use_pipemix() && N % 8 == 0) {
+  if (use_pipemix() && tn_cols_ok(N, ldb)) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)
@@ -134,7 +142,7 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
 extern "C" hipError_t gemm_tn_tn(const void *a, const void *b, void *c, int M,
                                  int N, int K, long lda, long ldb, long ldc,
                                  int c_f32, hipStream_t s) {
-  if (use_pipemix() && M % 8 == 0 && N % 8 == 0) {
+  if (use_pipemix() && tn_cols_ok(M, lda) && tn_cols_ok(N, ldb)) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
@@ -176,7 +184,7 @@ extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
-  if (use_pipemix() && M % 8 == 0 && N % 8 == 0) {
+  if (use_pipemix() && tn_cols_ok(M, lda) && tn_cols_ok(N, ldb)) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)
