@@ -280,13 +280,13 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
   }
 }
 
-// Partials grid is capped by the slab ([512][2][C]); the elementwise
-// apply kernels have no slab and get a bandwidth-sized grid (256 CUs want
-// >=2 blocks/CU in flight to cover HBM latency).
+// Partials grid is capped by the slab ([1024][2][C]); PMC showed the
+// 512-block cap left the stats kernels 75-79% wave-parked (HBM latency,
+// not bandwidth) — 1024 blocks doubles the outstanding-load pool.
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
-  grid = (int)(g > 512 ? 512 : (g < 1 ? 1 : g));
+  grid = (int)(g > 1024 ? 1024 : (g < 1 ? 1 : g));
 }
 
 static int bn_apply_grid(long M, int C8) {

@@ -340,7 +340,7 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
   Tensor y = empty_cl_bf16(N, C, H, W, x);
   Tensor mean = at::empty({C}, f32), invstd = at::empty({C}, f32);
   Tensor scale = at::empty({C}, f32), shift = at::empty({C}, f32);
-  Tensor partial = at::empty({512L * 2 * C}, f32);
+  Tensor partial = at::empty({1024L * 2 * C}, f32);
   float *rm = running_mean.defined() && running_mean.numel() == C
                   ? running_mean.data_ptr<float>() : nullptr;
   float *rv = rm ? running_var.data_ptr<float>() : nullptr;
@@ -383,7 +383,7 @@ static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
   Tensor dx = empty_cl_bf16(N, C, H, W, x);
   Tensor dgamma = at::empty({C}, f32), dbeta = at::empty({C}, f32);
   Tensor k1 = at::empty({C}, f32), k2 = at::empty({C}, f32), k3 = at::empty({C}, f32);
-  Tensor partial = at::empty({512L * 2 * C}, f32);
+  Tensor partial = at::empty({1024L * 2 * C}, f32);
   CHK(bn_bwd_launch(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
                     invstd.data_ptr<float>(), relu ? 1 : 0, dx.data_ptr(),
