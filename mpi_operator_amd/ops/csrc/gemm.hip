@@ -98,14 +98,14 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
 // `partial` reduced into bf16 C. Same splits heuristic as the dw path.
 extern "C" int gemm_tn_tn_splits(int M, int N, int K); // defined below
 
-// dx split-K measured NET-NEGATIVE on BERT bs=8 same-box (314/322 vs
-// 321/330 with all linear split-K off, while dw-only splits win at bs=32):
-// the bf16 dx is cheap to produce once, and the fp32 slab pass costs more
-// than the fill gain at these sizes — default OFF, MPIAMD_DX_SK=1 enables.
+// dx split-K: measured net-negative in round 1 (the runtime-loop
+// splitk_reduce was 73% wave-parked then), but AFTER the templated/unrolled
+// reduce rewrite it wins same-box at BERT-Large bs32: 1257 -> 1274 seq/s,
+// repeatable. Default ON; MPIAMD_DX_SK=0 disables.
 extern "C" int gemm_nt_tn_splits(int M, int N, int K) {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_DX_SK");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   return on ? gemm_tn_tn_splits(M, N, K) : 1;
 }
