@@ -284,9 +284,14 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
 // 512-block cap left the stats kernels 75-79% wave-parked (HBM latency,
 // not bandwidth) — 1024 blocks doubles the outstanding-load pool.
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
+  static const long cap = [] { // MPIAMD_BN_GRID: same-box A/B of the cap
+    const char *e = getenv("MPIAMD_BN_GRID");
+    long v = e ? atol(e) : 0;
+    return (v >= 1 && v <= 1024) ? v : 1024L;
+  }();
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
-  grid = (int)(g > 1024 ? 1024 : (g < 1 ? 1 : g));
+  grid = (int)(g > cap ? cap : (g < 1 ? 1 : g));
 }
 
 static int bn_apply_grid(long M, int C8) {
