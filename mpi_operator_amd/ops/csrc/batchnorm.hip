@@ -280,14 +280,15 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
   }
 }
 
-// Partials grid is capped by the slab ([1024][2][C]); PMC showed the
-// 512-block cap left the stats kernels 75-79% wave-parked (HBM latency,
-// not bandwidth) — 1024 blocks doubles the outstanding-load pool.
+// Partials grid cap: PMC showed 75-79% wave-parked at 512 blocks, but the
+// 1024-block experiment REGRESSED ResNet101 same-box (3212 vs 3440 img/s,
+// repeatable) — the doubled slab write+finalize read costs more than the
+// extra latency hiding buys. 512 stays; MPIAMD_BN_GRID for future A/Bs.
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
-  static const long cap = [] { // MPIAMD_BN_GRID: same-box A/B of the cap
+  static const long cap = [] {
     const char *e = getenv("MPIAMD_BN_GRID");
     long v = e ? atol(e) : 0;
-    return (v >= 1 && v <= 1024) ? v : 1024L;
+    return (v >= 1 && v <= 1024) ? v : 512L;
   }();
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
