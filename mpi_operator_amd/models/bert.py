@@ -230,6 +230,17 @@ class BertForPreTraining(nn.Module):
 
 
 def to_mi355x_bert(model: BertForPreTraining, device) -> BertForPreTraining:
-    """bf16 working weights on the GPU; LayerNorm/embedding math stays
-    numerically safe (LN computes in fp32 internally via torch)."""
-    return model.to(device=device, dtype=torch.bfloat16)
+    """bf16 working weights on the GPU, with LayerNorm params and linear
+    biases kept fp32: the HIP kernels consume them as fp32 operands anyway
+    (epilogue bias add, LN affine), so fp32 residency removes ~200 per-step
+    dtype-cast kernels (.float() on entry + grad .to(bf16) on exit —
+    ~2.6 ms/step on BERT-Large bs32, prof8 trace)."""
+    model = model.to(device=device, dtype=torch.bfloat16)
+    for mod in model.modules():
+        if isinstance(mod, LayerNorm):
+            mod.weight.data = mod.weight.data.float()
+            mod.bias.data = mod.bias.data.float()
+        elif isinstance(mod, BertLinear):
+            mod.bias.data = mod.bias.data.float()
+    model.mlm_bias.data = model.mlm_bias.data.float()
+    return model

@@ -121,7 +121,7 @@ hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 int gemm_tn_tn_splits(int, int, int);
 hipError_t gemm_tn_tn_sk(const void *, const void *, float *, void *, int,
-                         int, int, long, long, long, int, hipStream_t);
+                         int, int, long, long, long, int, int, hipStream_t);
 hipError_t gemm_nt_tn_sk(const void *, const void *, float *, void *, int,
                          int, int, long, long, long, int, hipStream_t);
 int gemm_nt_tn_splits(int, int, int);
@@ -543,15 +543,22 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   // split-K when the [N][K] tile grid underfills the chip (K_reduce = M).
   // The PADDED dyp (lda = Np) keeps the vocab-ragged head on the pipe
   // route (tn_cols_ok's padded-stride contract).
-  Tensor dw = at::empty({N, K}, f32);
+  // dw in the weight's own dtype: bf16 params take the rounding inside the
+  // GEMM epilogue / split-K reduce instead of an extra fp32 tensor + torch
+  // cast kernel per weight per step
+  bool wbf = wc.scalar_type() == at::kBFloat16;
+  Tensor dw = at::empty({N, K}, wbf ? wc.options() : f32);
   int dw_splits = gemm_tn_tn_splits(N, K, M);
   Tensor dwp = dw_splits > 1
                    ? at::empty({dw_splits, (long)N * K}, f32)
                    : dw;
-  CHK(gemm_tn_tn_sk(dyp.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
-                    dw.data_ptr(), N, K, M, Np, K, K, dw_splits,
+  CHK(gemm_tn_tn_sk(dyp.data_ptr(),
+                    xc.data_ptr(),
+                    dw_splits > 1 ? dwp.data_ptr<float>() : nullptr,
+                    dw.data_ptr(), N, K, M, Np, K, K, dw_splits, wbf ? 1 : 0,
                     cur_stream()));
-  Tensor db = at::zeros({N}, f32); // zeros: ragged-N fallback writes directly
+  // every colsum path writes all N entries — no zero-init needed
+  Tensor db = at::empty({N}, f32);
   int chunks = colsum_chunks(M, N);
   Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
   CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),
@@ -567,13 +574,16 @@ static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
   Tensor dyc = dy.contiguous(), xc = x.contiguous();
   int M = xc.size(0), K = xc.size(1), N = dyc.size(1);
   auto f32 = x.options().dtype(at::kFloat);
-  Tensor dw = at::empty({N, K}, f32);
+  bool wbf = xc.scalar_type() == at::kBFloat16; // dw matches the weight dtype
+  Tensor dw = at::empty({N, K}, wbf ? xc.options() : f32);
   int dw_splits = gemm_tn_tn_splits(N, K, M);
-  Tensor dwp = dw_splits > 1 ? at::empty({dw_splits, (long)N * K}, f32) : dw;
-  CHK(gemm_tn_tn_sk(dyc.data_ptr(), xc.data_ptr(), dwp.data_ptr<float>(),
-                    dw.data_ptr(), N, K, M, N, K, K, dw_splits,
+  Tensor dwp =
+      dw_splits > 1 ? at::empty({dw_splits, (long)N * K}, f32) : dw;
+  CHK(gemm_tn_tn_sk(dyc.data_ptr(), xc.data_ptr(),
+                    dw_splits > 1 ? dwp.data_ptr<float>() : nullptr,
+                    dw.data_ptr(), N, K, M, N, K, K, dw_splits, wbf ? 1 : 0,
                     cur_stream()));
-  Tensor db = at::zeros({N}, f32);
+  Tensor db = at::empty({N}, f32);
   int chunks = colsum_chunks(M, N);
   Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
   CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),

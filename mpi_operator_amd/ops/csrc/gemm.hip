@@ -183,28 +183,31 @@ extern "C" int gemm_tn_tn_splits(int M, int N, int K) {
 extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
                                     float *partial, void *c, int M, int N,
                                     int K, long lda, long ldb, long ldc,
-                                    int splits, hipStream_t s) {
+                                    int splits, int out_bf16, hipStream_t s) {
+  // out_bf16: write C in bf16 directly (weight-grad path — the fp32 C +
+  // torch .to(bf16) cast afterwards was ~200 cast kernels / 1.3 ms per
+  // BERT-Large step, prof8)
   if (use_pipemix() && tn_cols_ok(M, lda) && tn_cols_ok(N, ldb)) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)
       return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearWriter{ldc}, ldc,
-                                true, s);
+                                out_bf16 == 0, s);
     hipError_t e = launch_pipe_mix_wr(sa, sb, partial, M, N, K,
                                       LinearWriter{ldc}, ldc, true, s, splits);
     if (e != hipSuccess) return e;
-    return splitk_reduce(partial, splits, (long)M * ldc, c, 0, s);
+    return splitk_reduce(partial, splits, (long)M * ldc, c, out_bf16, s);
   }
   TnRowMajor la{(const uint16_t *)a, lda, K, M};
   TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
   if (splits <= 1)
     return launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
-                           c, M, N, K, ldc, true, s);
+                           c, M, N, K, ldc, out_bf16 == 0, s);
   hipError_t e = launch_mix_gemm(TnStage<TnRowMajor>{la},
                                  TnStage<TnRowMajor>{lb}, partial, M, N, K,
                                  ldc, true, s, splits);
   if (e != hipSuccess) return e;
-  return splitk_reduce(partial, splits, (long)M * ldc, c, 0, s);
+  return splitk_reduce(partial, splits, (long)M * ldc, c, out_bf16, s);
 }
 
 // bf16 2-D transpose: out[j][i] = in[i][j], output leading dim ldo >= R
