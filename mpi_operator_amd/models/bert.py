@@ -199,10 +199,31 @@ class BertForPreTraining(nn.Module):
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.nsp = BertLinear(cfg.hidden, 2)
 
-    def forward(self, ids, type_ids=None, attn_mask=None):
+    def forward(self, ids, type_ids=None, attn_mask=None, mlm_labels=None,
+                nsp_labels=None):
         x = self.bert(ids, type_ids, attn_mask)
         h = self.mlm_ln(F.gelu(self.mlm_transform(x), approximate="tanh"))
         tok_w = self.bert.embeddings.tok.weight
+        if mlm_labels is not None:
+            # labels-in-forward (HF-style) training path: the fused MLM head
+            # never materializes the unpadded [b·s, V] logits — decoder GEMM,
+            # masked CE and its backward all work on one padded-vocab buffer
+            if h.is_cuda and h.dtype == torch.bfloat16:
+                b, s, hd = h.shape
+                l_mlm = Fx.mlm_head_loss(h.reshape(-1, hd).contiguous(),
+                                         tok_w, self.mlm_bias.float(),
+                                         mlm_labels.view(-1))
+            else:
+                logits = torch.matmul(h, tok_w.t().to(h.dtype)) \
+                    + self.mlm_bias.to(h.dtype)
+                l_mlm = F.cross_entropy(
+                    logits.float().view(-1, self.cfg.vocab_size),
+                    mlm_labels.view(-1), ignore_index=-100)
+            loss = l_mlm
+            if nsp_labels is not None:
+                nsp_logits = self.nsp(x[:, 0])
+                loss = loss + F.cross_entropy(nsp_logits.float(), nsp_labels)
+            return loss
         if h.is_cuda and h.dtype == torch.bfloat16:
             # decoder GEMM on the in-tree NT kernel with the bias folded
             # into the epilogue (weight tying: grads flow to tok.weight)

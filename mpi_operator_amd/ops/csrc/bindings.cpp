@@ -67,7 +67,7 @@ hipError_t add_relu_bwd_add(const void *, const void *, const void *, void *,
 hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t bias_add(void *, const float *, long, int, hipStream_t);
-hipError_t colsum_bf16(const void *, float *, float *, long, int, hipStream_t);
+hipError_t colsum_bf16(const void *, float *, float *, long, int, long, hipStream_t);
 int colsum_chunks(long, int);
 struct SgdDesc {
   const void *grad;
@@ -95,10 +95,10 @@ hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
 hipError_t softmax_xent_fwd_launch(const void *, const long *, float *, float *,
                                    int, int, hipStream_t);
 hipError_t masked_xent_fwd_launch(const void *, const long *, float *, float *,
-                                  int, int, long, hipStream_t);
+                                  int, int, long, long, hipStream_t);
 hipError_t masked_xent_bwd_launch(const void *, const long *, const float *,
                                   const float *, const float *, void *, int,
-                                  int, long, hipStream_t);
+                                  int, long, long, hipStream_t);
 hipError_t attn_fwd_launch(const void *, void *, void *, const float *, int,
                            int, int, float, hipStream_t);
 hipError_t attn_bwd_launch(const void *, const void *, const void *, void *,
@@ -562,7 +562,7 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   int chunks = colsum_chunks(M, N);
   Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
   CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),
-                  db.data_ptr<float>(), M, N, cur_stream()));
+                  db.data_ptr<float>(), M, N, N, cur_stream()));
   return {dx, dw, db};
 }
 
@@ -587,7 +587,7 @@ static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
   int chunks = colsum_chunks(M, N);
   Tensor dbp = chunks > 0 ? at::empty({chunks, (long)N}, f32) : db;
   CHK(colsum_bf16(dyc.data_ptr(), dbp.data_ptr<float>(),
-                  db.data_ptr<float>(), M, N, cur_stream()));
+                  db.data_ptr<float>(), M, N, N, cur_stream()));
   return {dw, db};
 }
 
@@ -715,7 +715,7 @@ static std::vector<Tensor> masked_xent_fwd(const Tensor &logits,
   Tensor out = at::empty({2}, f32); // loss_sum, valid_count
   CHK(masked_xent_fwd_launch(logits.data_ptr(), target.data_ptr<long>(),
                              stats.data_ptr<float>(), out.data_ptr<float>(),
-                             B, V, ignore_index, cur_stream()));
+                             B, V, V, ignore_index, cur_stream()));
   return {out, stats};
 }
 
@@ -729,9 +729,81 @@ static Tensor masked_xent_bwd(const Tensor &logits, const Tensor &target,
   Tensor d = at::empty({B, V}, logits.options());
   CHK(masked_xent_bwd_launch(logits.data_ptr(), target.data_ptr<long>(),
                              stats.data_ptr<float>(), out.data_ptr<float>(),
-                             dloss.data_ptr<float>(), d.data_ptr(), B, V,
+                             dloss.data_ptr<float>(), d.data_ptr(), B, V, V,
                              ignore_index, cur_stream()));
   return d;
+}
+
+// --------------- fused MLM head: decoder GEMM + masked CE ---------------
+// The decoder writes logits into a [M][Vp] padded-vocab buffer (Vp =
+// roundup(V, 8)); CE reads rows with stride Vp; backward re-derives
+// probabilities into a padded dlogits whose pad columns the kernel zeroes,
+// so the dx/dw GEMMs and db colsum consume it DIRECTLY (lda = Vp) — no
+// per-step [M][Vp] zero-fill + strided pad copy (prof8: ~180 us/step), and
+// no unpadded fp32/bf16 logits round-trips. Reference role: the MLM
+// pretraining head of the out-of-tree BERT images.
+static std::vector<Tensor> mlm_head_fwd(const Tensor &h, const Tensor &w,
+                                        const Tensor &b, const Tensor &target,
+                                        int64_t ignore_index) {
+  TORCH_CHECK(h.is_cuda() && h.scalar_type() == at::kBFloat16 && h.dim() == 2);
+  TORCH_CHECK(h.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(b.scalar_type() == at::kFloat, "mlm bias must be fp32");
+  const HIPDeviceGuard guard(h.device());
+  int M = h.size(0), K = h.size(1), V = w.size(0);
+  TORCH_CHECK(w.size(1) == K && target.numel() == M);
+  long Vp = ((long)V + 7) / 8 * 8;
+  auto f32 = h.options().dtype(at::kFloat);
+  Tensor logits = at::empty({M, Vp}, h.options());
+  Tensor bc = b.contiguous();
+  CHK(gemm_nt_bias(h.data_ptr(), w.data_ptr(), bc.data_ptr<float>(),
+                   logits.data_ptr(), M, V, K, K, K, Vp, cur_stream()));
+  Tensor stats = at::empty({M, 2}, f32);
+  Tensor out = at::empty({2}, f32); // loss_sum, valid_count
+  CHK(masked_xent_fwd_launch(logits.data_ptr(), target.data_ptr<long>(),
+                             stats.data_ptr<float>(), out.data_ptr<float>(),
+                             M, V, Vp, ignore_index, cur_stream()));
+  return {out, logits, stats};
+}
+
+static std::vector<Tensor> mlm_head_bwd(const Tensor &logits, int64_t V64,
+                                        const Tensor &target,
+                                        const Tensor &stats, const Tensor &out,
+                                        const Tensor &dloss, const Tensor &h,
+                                        const Tensor &w, int64_t ignore_index) {
+  const HIPDeviceGuard guard(h.device());
+  int M = logits.size(0), V = (int)V64, K = h.size(1);
+  long Vp = logits.size(1);
+  TORCH_CHECK(dloss.is_cuda() && dloss.scalar_type() == at::kFloat &&
+              dloss.numel() == 1, "dloss must be a device fp32 scalar");
+  auto f32 = h.options().dtype(at::kFloat);
+  Tensor dlog = at::empty_like(logits);
+  CHK(masked_xent_bwd_launch(logits.data_ptr(), target.data_ptr<long>(),
+                             stats.data_ptr<float>(), out.data_ptr<float>(),
+                             dloss.data_ptr<float>(), dlog.data_ptr(), M, V,
+                             Vp, ignore_index, cur_stream()));
+  // dh = dlog . w (reduce over padded vocab; pads are zero)
+  Tensor dh = at::empty({M, K}, h.options());
+  int dx_splits = gemm_nt_tn_splits(M, K, V);
+  Tensor dhp = dx_splits > 1 ? at::empty({dx_splits, (long)M * K}, f32) : dh;
+  CHK(gemm_nt_tn_sk(dlog.data_ptr(), w.data_ptr(),
+                    dx_splits > 1 ? dhp.data_ptr<float>() : nullptr,
+                    dh.data_ptr(), M, K, V, Vp, K, K, dx_splits,
+                    cur_stream()));
+  // dw (tied embedding grad) in the weight dtype straight from the GEMM
+  bool wbf = w.scalar_type() == at::kBFloat16;
+  Tensor dw = at::empty({(long)V, K}, wbf ? w.options() : f32);
+  int dw_splits = gemm_tn_tn_splits(V, K, M);
+  Tensor dwp = dw_splits > 1 ? at::empty({dw_splits, (long)V * K}, f32) : dw;
+  CHK(gemm_tn_tn_sk(dlog.data_ptr(), h.data_ptr(),
+                    dw_splits > 1 ? dwp.data_ptr<float>() : nullptr,
+                    dw.data_ptr(), V, K, M, Vp, K, K, dw_splits, wbf ? 1 : 0,
+                    cur_stream()));
+  Tensor db = at::empty({V}, f32);
+  int chunks = colsum_chunks(M, V);
+  Tensor dbp = chunks > 0 ? at::empty({chunks, (long)V}, f32) : db;
+  CHK(colsum_bf16(dlog.data_ptr(), dbp.data_ptr<float>(),
+                  db.data_ptr<float>(), M, V, Vp, cur_stream()));
+  return {dh, dw, db};
 }
 
 // ------------------------- add-relu -------------------------
@@ -819,6 +891,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_bwd", &linear_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("masked_xent_fwd", &masked_xent_fwd);
+  m.def("mlm_head_fwd", &mlm_head_fwd);
+  m.def("mlm_head_bwd", &mlm_head_bwd);
   m.def("linear_gelu_fwd", &linear_gelu_fwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);

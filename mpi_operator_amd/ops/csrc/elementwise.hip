@@ -375,14 +375,15 @@ extern "C" hipError_t bias_add(void *y, const float *b, long M, int N,
 // -column serial version before THAT ran 16 blocks and was 48% of a
 // BERT-Large step.
 __global__ void colsum8_k(const ushort8 *__restrict__ dy,
-                          float *__restrict__ partial, long M, int C8) {
+                          float *__restrict__ partial, long M, int C8,
+                          long ld8) {
   int cb = blockIdx.y * 32 + (threadIdx.x & 31);
   int rl = threadIdx.x >> 5; // 8 row lanes per block
   float a[8] = {0};
   if (cb < C8) {
     for (long m = (long)blockIdx.x * 8 + rl; m < M; m += (long)gridDim.x * 8) {
       float f[8];
-      bf8_to_f8(dy[m * C8 + cb], f);
+      bf8_to_f8(dy[m * ld8 + cb], f);
 #pragma unroll
       for (int j = 0; j < 8; ++j) a[j] += f[j];
     }
@@ -404,11 +405,11 @@ __global__ void colsum8_k(const ushort8 *__restrict__ dy,
 
 // fallback for ragged N (e.g. the 2-way NSP head)
 __global__ void colsum_k(const uint16_t *__restrict__ dy, float *__restrict__ db,
-                         long M, int N) {
+                         long M, int N, long ld) {
   for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
        n += gridDim.x * blockDim.x) {
     float a = 0;
-    for (long m = 0; m < M; ++m) a += bf2f(dy[m * N + n]);
+    for (long m = 0; m < M; ++m) a += bf2f(dy[m * ld + n]);
     db[n] = a;
   }
 }
@@ -418,12 +419,12 @@ __global__ void colsum_k(const uint16_t *__restrict__ dy, float *__restrict__ db
 // row-chunks × col-chunks into fp32 partials, then a per-column reduce.
 __global__ void colsum_ragged_k(const uint16_t *__restrict__ dy,
                                 float *__restrict__ partial, long M, int N,
-                                long rchunk) {
+                                long ld, long rchunk) {
   int n = blockIdx.y * 256 + threadIdx.x;
   if (n >= N) return;
   long r0 = blockIdx.x * rchunk, r1 = min(M, r0 + rchunk);
   float a = 0;
-  for (long m = r0; m < r1; ++m) a += bf2f(dy[m * N + n]);
+  for (long m = r0; m < r1; ++m) a += bf2f(dy[m * ld + n]);
   partial[(long)blockIdx.x * N + n] = a;
 }
 
@@ -464,12 +465,13 @@ extern "C" hipError_t slab_colreduce(const float *, float *, int, long,
                                      hipStream_t); // conv.hip
 
 extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
-                                  long M, int N, hipStream_t s) {
-  if (N % 8 == 0) {
+                                  long M, int N, long ld, hipStream_t s) {
+  if (N % 8 == 0 && ld % 8 == 0) {
     int C8 = N / 8;
     int gy = (C8 + 31) / 32;
     int gx = colsum_chunks(M, N);
-    colsum8_k<<<dim3(gx, gy), 256, 0, s>>>((const ushort8 *)dy, partial, M, C8);
+    colsum8_k<<<dim3(gx, gy), 256, 0, s>>>((const ushort8 *)dy, partial, M, C8,
+                                           ld / 8);
     HIP_KERNEL_CHECK();
     // short output + many chunks: per-column reduce (the float4 kernel
     // collapsed to one block at len=N)
@@ -480,11 +482,11 @@ extern "C" hipError_t colsum_bf16(const void *dy, float *partial, float *db,
     long rchunk = (M + gx - 1) / gx;
     dim3 grid(gx, (N + 255) / 256);
     colsum_ragged_k<<<grid, 256, 0, s>>>((const uint16_t *)dy, partial, M, N,
-                                         rchunk);
+                                         ld, rchunk);
     HIP_KERNEL_CHECK();
     return slab_colreduce(partial, db, gx, (long)N, s);
   }
-  colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N);
+  colsum_k<<<cdiv_h(N, 256), 256, 0, s>>>((const uint16_t *)dy, db, M, N, ld);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

@@ -17,10 +17,10 @@ __global__ void masked_xent_fwd_k(const uint16_t *__restrict__ logits,
                                   const long *__restrict__ target,
                                   float *__restrict__ stats, // [B][2]
                                   float *__restrict__ out,   // [2]: sum, count
-                                  int B, int V, long ignore_index) {
+                                  int B, int V, long ld, long ignore_index) {
   int b = blockIdx.x;
   long t = target[b];
-  const uint16_t *row = logits + (long)b * V;
+  const uint16_t *row = logits + (long)b * ld; // ld >= V (padded-vocab rows)
   // 16 B/lane vectorized row passes (scalar u16 strided loads measured
   // 254 µs/call at vocab scale — 8x off bandwidth). Ragged V means row
   // bases are only 4-B aligned: scalar head up to the first 16-B
@@ -76,11 +76,15 @@ __global__ void masked_xent_bwd_k(const uint16_t *__restrict__ logits,
                                   const float *__restrict__ out, // [2]
                                   const float *__restrict__ dscale,
                                   uint16_t *__restrict__ dlogits, int B,
-                                  int V, long ignore_index) {
+                                  int V, long ld, long ignore_index) {
   int b = blockIdx.x;
   long t = target[b];
-  const uint16_t *row = logits + (long)b * V;
-  uint16_t *drow = dlogits + (long)b * V;
+  const uint16_t *row = logits + (long)b * ld;
+  uint16_t *drow = dlogits + (long)b * ld;
+  // pad columns [V, ld) are written ZERO: the downstream dx GEMM reduces
+  // over the padded vocab dim and relies on zero pads to mask its
+  // zeros-page-granule edge (gemm.hip tn_cols_ok contract)
+  for (int v = V + (int)threadIdx.x; v < ld; v += blockDim.x) drow[v] = 0;
   if (t == ignore_index) {
     for (int v = threadIdx.x; v < V; v += blockDim.x) drow[v] = 0;
     return;
@@ -123,12 +127,12 @@ __global__ void masked_xent_bwd_k(const uint16_t *__restrict__ logits,
 
 extern "C" hipError_t masked_xent_fwd_launch(const void *logits,
                                              const long *target, float *stats,
-                                             float *out, int B, int V,
+                                             float *out, int B, int V, long ld,
                                              long ignore_index,
                                              hipStream_t s) {
   hipMemsetAsync(out, 0, 8, s);
   masked_xent_fwd_k<<<B, 256, 0, s>>>((const uint16_t *)logits, target, stats,
-                                      out, B, V, ignore_index);
+                                      out, B, V, ld, ignore_index);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -138,11 +142,11 @@ extern "C" hipError_t masked_xent_bwd_launch(const void *logits,
                                              const float *stats,
                                              const float *out,
                                              const float *dscale, void *dlogits,
-                                             int B, int V, long ignore_index,
-                                             hipStream_t s) {
+                                             int B, int V, long ld,
+                                             long ignore_index, hipStream_t s) {
   masked_xent_bwd_k<<<B, 256, 0, s>>>((const uint16_t *)logits, target, stats,
                                       out, dscale, (uint16_t *)dlogits, B, V,
-                                      ignore_index);
+                                      ld, ignore_index);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

@@ -320,6 +320,36 @@ def masked_softmax_cross_entropy(logits, target, ignore_index=-100):
     return MaskedSoftmaxCrossEntropyFn.apply(logits, target, ignore_index)
 
 
+class MlmHeadLossFn(torch.autograd.Function):
+    """Fused MLM decoder + masked CE: the vocab-scale logits live ONLY in a
+    [M, roundup(V,8)] padded buffer — the decoder GEMM writes it (bias in
+    the epilogue), CE reads it strided, and the backward kernel writes the
+    padded dlogits (zeroed pads) that the dx/dw GEMMs and db colsum consume
+    in place. Removes the per-step pad zero-fill + strided copy and the
+    unpadded logits materialization of the linear+CE composition."""
+
+    @staticmethod
+    def forward(ctx, h, w, b, target, ignore_index):
+        out, logits_pad, stats = hip_ext().mlm_head_fwd(h, w, b, target,
+                                                        ignore_index)
+        ctx.save_for_backward(h, w, logits_pad, stats, out, target)
+        ctx.ignore_index = ignore_index
+        ctx.V = w.shape[0]
+        return out[0] / out[1].clamp(min=1.0)
+
+    @staticmethod
+    def backward(ctx, dloss):
+        h, w, logits_pad, stats, out, target = ctx.saved_tensors
+        dh, dw, db = hip_ext().mlm_head_bwd(
+            logits_pad, ctx.V, target, stats, out,
+            dloss.reshape(1).float().contiguous(), h, w, ctx.ignore_index)
+        return dh, dw, db, None, None
+
+
+def mlm_head_loss(h, w, b, target, ignore_index=-100):
+    return MlmHeadLossFn.apply(h, w, b, target, ignore_index)
+
+
 class FfnFn(torch.autograd.Function):
     """BERT FFN pair fc2(gelu(fc1(x))) with the GELU fused into the GEMM
     epilogues on both sides: fc1's epilogue emits (h_pre, gelu(h)); the

@@ -82,3 +82,19 @@ def test_bert_all_params_receive_grads():
     missing = [n for n, p in m.named_parameters()
                if p.grad is None and "pos" not in n and "typ" not in n]
     assert not missing, missing
+
+
+def test_labels_in_forward_matches_loss_composition():
+    """m(ids, mlm_labels=..., nsp_labels=...) (HF-style) equals the
+    logits + m.loss(...) composition on the CPU path."""
+    torch.manual_seed(5)
+    cfg = tiny_cfg()
+    m = BertForPreTraining(cfg)
+    ids = torch.randint(0, cfg.vocab_size, (3, 12))
+    mlm_labels = torch.full_like(ids, -100)
+    mlm_labels[:, ::3] = ids[:, ::3]
+    nsp = torch.randint(0, 2, (3,))
+    loss_fwd = m(ids, mlm_labels=mlm_labels, nsp_labels=nsp)
+    mlm_logits, nsp_logits = m(ids)
+    loss_ref = m.loss(mlm_logits, nsp_logits, mlm_labels, nsp)
+    assert abs(loss_fwd.item() - loss_ref.item()) < 1e-5 * abs(loss_ref.item()) + 1e-6

@@ -61,6 +61,38 @@ def test_bert_base_train_step_gpu():
     assert losses[-1] < losses[0], losses
 
 
+def test_mlm_head_loss_matches_composed_path():
+    """Fused decoder+CE head (padded-vocab buffer) vs the logits+loss
+    composition: same loss and same input/weight/bias grads."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(7)
+    M, K, V = 256, 768, 30522  # ragged vocab exercises the pad contract
+    h = ((torch.rand(M, K, device="cuda") * 2 - 1) * 0.5).to(torch.bfloat16)
+    w = ((torch.rand(V, K, device="cuda") * 2 - 1) * 0.05).to(torch.bfloat16)
+    b = torch.randn(V, device="cuda") * 0.01
+    tgt = torch.randint(0, V, (M,), device="cuda")
+    tgt[::3] = -100  # masked-out rows
+
+    h1 = h.clone().requires_grad_()
+    w1 = w.clone().requires_grad_()
+    b1 = b.clone().requires_grad_()
+    loss1 = Fx.mlm_head_loss(h1, w1, b1, tgt)
+    loss1.backward()
+
+    h2 = h.clone().requires_grad_()
+    w2 = w.clone().requires_grad_()
+    b2 = b.clone().requires_grad_()
+    logits = Fx.linear(h2, w2, b2)
+    loss2 = Fx.masked_softmax_cross_entropy(logits, tgt)
+    loss2.backward()
+
+    assert abs(loss1.item() - loss2.item()) < 2e-3 * abs(loss2.item()) + 1e-4
+    for a, c in ((h1.grad, h2.grad), (w1.grad, w2.grad), (b1.grad, b2.grad)):
+        num = (a.float() - c.float()).norm().item()
+        den = c.float().norm().item() + 1e-30
+        assert num / den < 0.02, (num / den, a.shape)
+
+
 @pytest.mark.parametrize("M,N", [(4096, 1024), (100, 768), (7, 2048)])
 def test_layernorm_fwd_bwd_matches_fp32(M, N):
     from mpi_operator_amd.ops import functional as Fx

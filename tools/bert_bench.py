@@ -38,8 +38,8 @@ def main():
 
     def step():
         opt.zero_grad()
-        mlm_logits, nsp_logits = m(ids)
-        loss = m.loss(mlm_logits, nsp_logits, mlm_labels, nsp)
+        # labels-in-forward: fused MLM head (no unpadded vocab logits)
+        loss = m(ids, mlm_labels=mlm_labels, nsp_labels=nsp)
         loss.backward()
         opt.step()
         return loss
