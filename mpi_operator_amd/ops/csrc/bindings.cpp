@@ -566,6 +566,29 @@ static std::vector<Tensor> linear_bwd(const Tensor &dy, const Tensor &x,
   return {dx, dw, db};
 }
 
+// dx only (side-stream wgrad mode: the main stream runs the dgrad chain
+// while linear_wgrad_only runs concurrently on a second stream)
+static Tensor linear_dgrad(const Tensor &dy, const Tensor &w) {
+  const HIPDeviceGuard guard(dy.device());
+  Tensor dyc = dy.contiguous(), wc = w.contiguous();
+  int M = dyc.size(0), N = wc.size(0), K = wc.size(1);
+  auto f32 = dy.options().dtype(at::kFloat);
+  long Np = (N + 7) / 8 * 8;
+  Tensor dyp = dyc;
+  if (Np != N) {
+    dyp = at::zeros({M, Np}, dyc.options());
+    dyp.narrow(1, 0, N).copy_(dyc);
+  }
+  Tensor dx = at::empty({M, K}, dyc.options());
+  int dx_splits = gemm_nt_tn_splits(M, K, N);
+  Tensor dxp = dx_splits > 1 ? at::empty({dx_splits, (long)M * K}, f32) : dx;
+  CHK(gemm_nt_tn_sk(dyp.data_ptr(), wc.data_ptr(),
+                    dx_splits > 1 ? dxp.data_ptr<float>() : nullptr,
+                    dx.data_ptr(), M, K, N, Np, K, K, dx_splits,
+                    cur_stream()));
+  return dx;
+}
+
 // dw/db only (FFN backward computes fc2's dx separately with the fused
 // dgelu epilogue — re-running the full linear_bwd would pay that GEMM twice)
 static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
@@ -798,12 +821,15 @@ static std::vector<Tensor> mlm_head_bwd(const Tensor &logits, int64_t V64,
                     dw_splits > 1 ? dwp.data_ptr<float>() : nullptr,
                     dw.data_ptr(), V, K, M, Vp, K, K, dw_splits, wbf ? 1 : 0,
                     cur_stream()));
-  Tensor db = at::empty({V}, f32);
-  int chunks = colsum_chunks(M, V);
-  Tensor dbp = chunks > 0 ? at::empty({chunks, (long)V}, f32) : db;
+  // db over the PADDED width: pads are zero, so summing Vp columns keeps
+  // the %8-vectorized colsum8 slab path (the ragged V route ran 2-B scalar
+  // loads at ~4x off bandwidth); the bias grad is the leading V entries
+  Tensor db_pad = at::empty({Vp}, f32);
+  int chunks = colsum_chunks(M, (int)Vp);
+  Tensor dbp = chunks > 0 ? at::empty({chunks, Vp}, f32) : db_pad;
   CHK(colsum_bf16(dlog.data_ptr(), dbp.data_ptr<float>(),
-                  db.data_ptr<float>(), M, V, Vp, cur_stream()));
-  return {dh, dw, db};
+                  db_pad.data_ptr<float>(), M, (int)Vp, Vp, cur_stream()));
+  return {dh, dw, db_pad.narrow(0, 0, V)};
 }
 
 // ------------------------- add-relu -------------------------
@@ -889,6 +915,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_add_fwd", &layernorm_add_fwd);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
+  m.def("linear_dgrad", &linear_dgrad);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("masked_xent_fwd", &masked_xent_fwd);
   m.def("mlm_head_fwd", &mlm_head_fwd);

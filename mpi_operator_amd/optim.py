@@ -10,6 +10,7 @@ from __future__ import annotations
 import torch
 
 from .ops import sgd_momentum_step
+from .ops import functional as _Fx
 
 
 class FusedSGD(torch.optim.Optimizer):
@@ -20,6 +21,9 @@ class FusedSGD(torch.optim.Optimizer):
 
     @torch.no_grad()
     def step(self, closure=None):
+        # side-stream wgrad mode: grads may still be in flight on the side
+        # stream — order them before the fused update (no-op otherwise)
+        _Fx.join_wgrad_stream()
         loss = None
         if closure is not None:
             with torch.enable_grad():
