@@ -14,8 +14,6 @@ reference delegates to external CUDA images.
 """
 from __future__ import annotations
 
-import os
-
 import torch
 
 from . import reference as ref
@@ -26,55 +24,17 @@ def _cl(t):  # channels_last view check for 4-D activations
     return t.is_contiguous(memory_format=torch.channels_last)
 
 
-# ---- side-stream weight gradients (MPIAMD_SIDE_WGRAD=1) ----
-# The backward's dw/db GEMMs have no consumer until the optimizer step, so
-# they can run on a second HIP stream concurrently with the serial dgrad
-# chain (wgrad tiles fill CUs the dgrad tail leaves idle). Event edges make
-# this capture-safe: inside a hipGraph the record/wait pairs become graph
-# dependencies. FusedSGD.step() calls join_wgrad_stream() before touching
-# grads; single-process only (the DDP bucket hooks assume main-stream
-# grads), so the trainer disables it when world_size > 1.
-_SIDE_WGRAD = os.environ.get("MPIAMD_SIDE_WGRAD", "0") == "1"
-_side_stream = None
-_side_pending: list = []
-
-
-def side_wgrad_enabled() -> bool:
-    return _SIDE_WGRAD and torch.cuda.is_available() and not (
-        torch.distributed.is_available()
-        and torch.distributed.is_initialized()
-        and torch.distributed.get_world_size() > 1)
-
-
-def _run_side(fn):
-    """Run fn() on the side stream, ordered after all current main-stream
-    work; outputs joined back in join_wgrad_stream()."""
-    global _side_stream
-    if _side_stream is None:
-        _side_stream = torch.cuda.Stream()
-    ev = torch.cuda.Event()
-    ev.record()
-    _side_stream.wait_event(ev)
-    with torch.cuda.stream(_side_stream):
-        outs = fn()
-        done = torch.cuda.Event()
-        done.record()
-    _side_pending.append((done, outs))
-    return outs
+# Side-stream weight-gradient overlap was tried and REJECTED (round 2):
+# running the dw/db GEMMs on a second HIP stream concurrently with the
+# dgrad chain measured 1363 -> 1310 seq/s on BERT-Large bs32 — the wgrad
+# and dgrad pipelined GEMMs contend for CUs and both fall off their tuned
+# occupancy; correctness also requires record_stream on every cross-stream
+# input. Kernels stay single-stream; join_wgrad_stream() remains as a
+# no-op contract point for FusedSGD.
 
 
 def join_wgrad_stream():
-    """Make the current stream wait for all pending side-stream wgrads.
-    No-op when nothing is pending (safe to call unconditionally)."""
-    if not _side_pending:
-        return
-    cur = torch.cuda.current_stream()
-    for done, outs in _side_pending:
-        cur.wait_event(done)
-        for t in outs:
-            if torch.is_tensor(t):
-                t.record_stream(cur)
-    _side_pending.clear()
+    return None
 
 
 class Conv2dFn(torch.autograd.Function):
@@ -232,11 +192,7 @@ class LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         if dy.is_cuda:
             dy = dy.contiguous()
-            if side_wgrad_enabled():
-                dw, db = _run_side(lambda: hip_ext().linear_wgrad_only(dy, x))
-                dx = hip_ext().linear_dgrad(dy, w)
-            else:
-                dx, dw, db = hip_ext().linear_bwd(dy, x, w)
+            dx, dw, db = hip_ext().linear_bwd(dy, x, w)
             return dx, dw.to(w.dtype), db.to(ctx.b_dtype)
         dx = dy @ w
         dw = dy.transpose(0, 1).float() @ x.float()
@@ -433,18 +389,10 @@ class FfnFn(torch.autograd.Function):
         bd1, bd2 = ctx.b_dtypes
         if dy.is_cuda:
             dy = dy.contiguous()
-            if side_wgrad_enabled():
-                dw2, db2 = _run_side(
-                    lambda: hip_ext().linear_wgrad_only(dy, g))
-                dh = hip_ext().linear_gelu_dgrad(dy, w2, pre)
-                dw1, db1 = _run_side(
-                    lambda: hip_ext().linear_wgrad_only(dh, x))
-                dx = hip_ext().linear_dgrad(dh, w1)
-            else:
-                # fc2 weight grads; its dx is produced by the fused-dgelu GEMM
-                dw2, db2 = hip_ext().linear_wgrad_only(dy, g)
-                dh = hip_ext().linear_gelu_dgrad(dy, w2, pre)
-                dx, dw1, db1 = hip_ext().linear_bwd(dh, x, w1)
+            # fc2 weight grads; its dx is produced by the fused-dgelu GEMM
+            dw2, db2 = hip_ext().linear_wgrad_only(dy, g)
+            dh = hip_ext().linear_gelu_dgrad(dy, w2, pre)
+            dx, dw1, db1 = hip_ext().linear_bwd(dh, x, w1)
             return (dx, dw1.to(w1.dtype), db1.to(bd1),
                     dw2.to(w2.dtype), db2.to(bd2))
         dg = dy @ w2
