@@ -674,11 +674,11 @@ static std::vector<Tensor> linear_gelu_fwd(const Tensor &x, const Tensor &w,
   TORCH_CHECK(w.size(1) == K);
   Tensor bias = b.to(at::kFloat).contiguous();
   Tensor g = at::empty({M, N}, x.options());
-  Tensor pre = at::empty({M, N}, x.options());
+  Tensor deriv = at::empty({M, N}, x.options()); // gelu'(h), for backward
   CHK(gemm_nt_gelu_bias(x.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
-                        pre.data_ptr(), g.data_ptr(), M, N, K, K, K, N,
+                        deriv.data_ptr(), g.data_ptr(), M, N, K, K, K, N,
                         cur_stream()));
-  return {g, pre};
+  return {g, deriv};
 }
 
 // fc2-dx with dgelu fused: dh = (dy·w2) ⊙ gelu'(h_pre)

@@ -34,13 +34,14 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, false, s, 1, bias);
 }
 
-// FFN fc1 fused forward: g = gelu(x·w1ᵀ + b), pre-activation h saved for
-// backward (GeluBiasWriter). Both operands NT k-contiguous.
+// FFN fc1 fused forward: g = gelu(x·w1ᵀ + b); the epilogue also saves
+// gelu'(h) (bf16, layout of C) so the backward epilogue is exp-free
+// (GeluBiasWriter). Both operands NT k-contiguous.
 extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
-                                        const float *bias, void *pre, void *g,
+                                        const float *bias, void *deriv, void *g,
                                         int M, int N, int K, long lda,
                                         long ldb, long ldc, hipStream_t s) {
-  GeluBiasWriter wrt{ldc, bias, (uint16_t *)pre};
+  GeluBiasWriter wrt{ldc, bias, (uint16_t *)deriv};
   GemmLoader la{(const uint16_t *)x, M, lda, K};
   GemmLoader lb{(const uint16_t *)w, N, ldb, K};
   if (M % 256 == 0 && N % 256 == 0 && K % 64 == 0 && ldc == N &&
@@ -56,13 +57,13 @@ extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
   return launch_mix_gemm_wr(ga, gb, g, M, N, K, wrt, ldc, false, s);
 }
 
-// FFN backward: dh = (dy·w2) ⊙ gelu'(h)  — the fc2-dx GEMM with the
-// dgelu applied in the epilogue (GeluBwdWriter); A NT, B (w2) TN.
+// FFN backward: dh = (dy·w2) ⊙ gelu'(h) — the fc2-dx GEMM multiplying by
+// the SAVED derivative in the epilogue (GeluBwdWriter); A NT, B (w2) TN.
 extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
-                                         const void *pre, void *dh, int M,
+                                         const void *deriv, void *dh, int M,
                                          int N, int K, long lda, long ldb,
                                          long ldc, hipStream_t s) {
-  GeluBwdWriter wrt{ldc, (const uint16_t *)pre};
+  GeluBwdWriter wrt{ldc, (const uint16_t *)deriv};
   if (use_pipemix() && N % 8 == 0) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, ldb, K, N}};

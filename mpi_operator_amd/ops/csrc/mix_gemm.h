@@ -187,6 +187,18 @@ DEV_INLINE float dgelu_f(float x) {
          0.5f * x * (1.f - t * t) * 0.7978845608028654f *
              (1.f + 3.f * 0.044715f * x * x);
 }
+// gelu and its derivative off ONE tanh evaluation (forward epilogue saves
+// the derivative so the backward epilogue is exp-free — recomputing
+// gelu'(h) there cost ~27 us of v_exp per fc2-dx GEMM, 16.7M elements at
+// the transcendental issue rate, serialized after the MFMA work)
+DEV_INLINE float gelu_pair_f(float x, float &dg) {
+  float u = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+  float t = tanh_fast(u);
+  dg = 0.5f * (1.f + t) +
+       0.5f * x * (1.f - t * t) * 0.7978845608028654f *
+           (1.f + 3.f * 0.044715f * x * x);
+  return 0.5f * x * (1.f + t);
+}
 
 // FFN fc1 epilogue: writes BOTH the pre-activation h = xW₁ᵀ+b (saved for
 // backward) and g = gelu(h) — the separate torch GELU pass re-read the
@@ -196,18 +208,18 @@ struct GeluBiasWriter {
   static constexpr bool STATS = false;
   long ldc;
   const float *bias;
-  uint16_t *pre; // h (pre-activation), bf16, same layout as C
+  uint16_t *deriv; // gelu'(h), bf16, same layout as C (saved for backward)
   typedef long RowCtx;
   DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
   DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
-    float h = v + bias[col];
-    pre[b + col] = f2bf(h);
-    p[b + col] = gelu_f(h);
+    float h = v + bias[col], dg;
+    p[b + col] = gelu_pair_f(h, dg);
+    deriv[b + col] = f2bf(dg);
   }
   DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
-    float h = v + bias[col];
-    pre[b + col] = f2bf(h);
-    p[b + col] = f2bf(gelu_f(h));
+    float h = v + bias[col], dg;
+    p[b + col] = f2bf(gelu_pair_f(h, dg));
+    deriv[b + col] = f2bf(dg);
   }
 };
 
@@ -218,14 +230,14 @@ struct GeluBwdWriter {
   static constexpr bool ACC = false;
   static constexpr bool STATS = false;
   long ldc;
-  const uint16_t *pre; // h (pre-activation) saved by GeluBiasWriter
+  const uint16_t *deriv; // gelu'(h) saved by GeluBiasWriter — exp-free
   typedef long RowCtx;
   DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
   DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
-    p[b + col] = v * dgelu_f(bf2f(pre[b + col]));
+    p[b + col] = v * bf2f(deriv[b + col]);
   }
   DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
-    p[b + col] = f2bf(v * dgelu_f(bf2f(pre[b + col])));
+    p[b + col] = f2bf(v * bf2f(deriv[b + col]));
   }
 };
 

@@ -365,38 +365,41 @@ def mlm_head_loss(h, w, b, target, ignore_index=-100):
 
 class FfnFn(torch.autograd.Function):
     """BERT FFN pair fc2(gelu(fc1(x))) with the GELU fused into the GEMM
-    epilogues on both sides: fc1's epilogue emits (h_pre, gelu(h)); the
-    backward fc2-dx GEMM multiplies by gelu'(h_pre) in its epilogue — the
-    two standalone GELU elementwise passes over the [M, intermediate]
-    activation disappear. x: [M, H]; w1: [I, H]; w2: [H, I]."""
+    epilogues on both sides: fc1's epilogue emits (gelu(h), gelu'(h)) off a
+    single tanh; the backward fc2-dx GEMM multiplies by the SAVED gelu'(h)
+    in its epilogue (exp-free — recomputing the tanh there serialized ~27 us
+    of v_exp per layer after the MFMA work). The two standalone GELU
+    elementwise passes over the [M, intermediate] activation disappear.
+    x: [M, H]; w1: [I, H]; w2: [H, I]. On CPU `aux` holds the
+    pre-activation instead and the derivative is recomputed in fp32."""
 
     @staticmethod
     def forward(ctx, x, w1, b1, w2, b2):
         if x.is_cuda:
-            g, pre = hip_ext().linear_gelu_fwd(x, w1, b1)
+            g, aux = hip_ext().linear_gelu_fwd(x, w1, b1)  # aux = gelu'(h)
             y = hip_ext().linear_fwd(g, w2, b2)
         else:
-            pre = ref.linear_fwd(x, w1, b1.float()).float()
-            g = torch.nn.functional.gelu(pre, approximate="tanh").to(x.dtype)
+            aux = ref.linear_fwd(x, w1, b1.float()).float()  # aux = h_pre
+            g = torch.nn.functional.gelu(aux, approximate="tanh").to(x.dtype)
             y = ref.linear_fwd(g, w2, b2)
-        ctx.save_for_backward(x, w1, w2, g, pre)
+        ctx.save_for_backward(x, w1, w2, g, aux)
         ctx.b_dtypes = (b1.dtype, b2.dtype)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w1, w2, g, pre = ctx.saved_tensors
+        x, w1, w2, g, aux = ctx.saved_tensors
         bd1, bd2 = ctx.b_dtypes
         if dy.is_cuda:
             dy = dy.contiguous()
             # fc2 weight grads; its dx is produced by the fused-dgelu GEMM
             dw2, db2 = hip_ext().linear_wgrad_only(dy, g)
-            dh = hip_ext().linear_gelu_dgrad(dy, w2, pre)
+            dh = hip_ext().linear_gelu_dgrad(dy, w2, aux)
             dx, dw1, db1 = hip_ext().linear_bwd(dh, x, w1)
             return (dx, dw1.to(w1.dtype), db1.to(bd1),
                     dw2.to(w2.dtype), db2.to(bd2))
         dg = dy @ w2
-        dh = (dg.float() * ref.dgelu(pre)).to(dy.dtype)
+        dh = (dg.float() * ref.dgelu(aux)).to(dy.dtype)
         dx = dh @ w1
         dw1 = dh.transpose(0, 1).float() @ x.float()
         db1 = dh.float().sum(0)
