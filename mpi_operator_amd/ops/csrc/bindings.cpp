@@ -109,6 +109,10 @@ hipError_t gemm_nt(const void *, const void *, void *, int, int, int, long,
                    long, long, int, hipStream_t);
 hipError_t gemm_nt_bias(const void *, const void *, const float *, void *,
                         int, int, int, long, long, long, hipStream_t);
+hipError_t gemm_nt_bias_sk(const void *, const void *, const float *, float *,
+                           void *, int, int, int, long, long, long, int,
+                           hipStream_t);
+int gemm_fwd_splits(int, int, int);
 hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 hipError_t gemm_nt_gelu_bias(const void *, const void *, const float *,
@@ -512,8 +516,15 @@ static Tensor linear_fwd(const Tensor &x, const Tensor &w, const Tensor &b) {
   // fp32 — the classifier keeps an fp32 bias for exactness)
   Tensor bf = b.scalar_type() == at::kFloat ? b.contiguous()
                                             : b.to(at::kFloat).contiguous();
-  CHK(gemm_nt_bias(xc.data_ptr(), wc.data_ptr(), bf.data_ptr<float>(),
-                   y.data_ptr(), M, N, K, K, K, N, cur_stream()));
+  int fsp = gemm_fwd_splits(M, N, K);
+  Tensor part;
+  float *pp = nullptr;
+  if (fsp > 1) {
+    part = at::empty({fsp, (long)M * N}, x.options().dtype(at::kFloat));
+    pp = part.data_ptr<float>();
+  }
+  CHK(gemm_nt_bias_sk(xc.data_ptr(), wc.data_ptr(), bf.data_ptr<float>(), pp,
+                      y.data_ptr(), M, N, K, K, K, N, fsp, cur_stream()));
   return y;
 }
 

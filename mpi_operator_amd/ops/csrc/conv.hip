@@ -525,6 +525,60 @@ __global__ void splitk_reduce_k(const float4v *__restrict__ partial,
   }
 }
 
+// Split-K reduce + bias epilogue for FORWARD activations (bf16 out): the
+// fc2-class shapes (M=4096, N=1024) give 256 pipe-mix workgroups = 1 per
+// CU; split-K x2 fills the second block slot, and the bias lands here
+// instead of in the (now fp32-slab) GEMM epilogue. col = flat % N.
+template <int SPLITS>
+__global__ void splitk_bias_reduce_k(const float4v *__restrict__ partial,
+                                     int splits_rt, long len4, int N4,
+                                     const float4v *__restrict__ bias,
+                                     uint16_t *__restrict__ out) {
+  const int splits = SPLITS ? SPLITS : splits_rt;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < len4;
+       i += (long)gridDim.x * blockDim.x) {
+    float4v a = partial[i];
+    float4v b = {0.f, 0.f, 0.f, 0.f};
+    int s = 1;
+#pragma unroll
+    for (; s + 1 < splits; s += 2) {
+      a += partial[(long)s * len4 + i];
+      b += partial[(long)(s + 1) * len4 + i];
+    }
+    if (s < splits) b += partial[(long)s * len4 + i];
+    a += b + bias[(int)(i % N4)];
+    uint16_t *o = out + i * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = f2bf(a[j]);
+  }
+}
+
+extern "C" hipError_t splitk_bias_reduce(const float *partial, int splits,
+                                         long len, int N, const float *bias,
+                                         void *out, hipStream_t s) {
+  long len4 = len / 4;
+  int grid = (int)((len4 + 255) / 256);
+  if (grid > 4096) grid = 4096;
+  switch (splits) {
+  case 2:
+    splitk_bias_reduce_k<2><<<grid, 256, 0, s>>>(
+        (const float4v *)partial, splits, len4, N / 4, (const float4v *)bias,
+        (uint16_t *)out);
+    break;
+  case 4:
+    splitk_bias_reduce_k<4><<<grid, 256, 0, s>>>(
+        (const float4v *)partial, splits, len4, N / 4, (const float4v *)bias,
+        (uint16_t *)out);
+    break;
+  default:
+    splitk_bias_reduce_k<0><<<grid, 256, 0, s>>>(
+        (const float4v *)partial, splits, len4, N / 4, (const float4v *)bias,
+        (uint16_t *)out);
+  }
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
 // Per-column slab reduce for SHORT outputs with MANY chunks (colsum db:
 // len = N ≤ 4k, chunks up to 256 — the float4 kernel collapsed to one
 // block there): one 256-thread block per column, threads stride chunks.

@@ -6,6 +6,9 @@
 
 extern "C" hipError_t splitk_reduce(const float *, int, long, void *, int,
                                     hipStream_t); // conv.hip
+extern "C" hipError_t splitk_bias_reduce(const float *, int, long, int,
+                                         const float *, void *,
+                                         hipStream_t); // conv.hip
 
 // A TN operand's 16-B column granules may overread up to 7 elements past
 // its `dim` columns; safe when the dim is %8, or when the ROW STRIDE is %8
@@ -34,14 +37,53 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
   return launch_nt_gemm(la, lb, c, M, N, K, ldc, false, s, 1, bias);
 }
 
+// Forward split-K (fc2-class M=4096/N=1024 shapes give 256 pipe-mix
+// workgroups = exactly 1/CU, half the block slots idle; deep K amortizes
+// the fp32 slab + bias-reduce pass). Gate MPIAMD_FWD_SK; heuristic shared
+// with the dw path.
+extern "C" int gemm_tn_tn_splits(int M, int N, int K); // below
+extern "C" int gemm_fwd_splits(int M, int N, int K) {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_FWD_SK");
+    return e && e[0] == '1';
+  }();
+  if (!on || N % 8) return 1;
+  int sp = gemm_tn_tn_splits(M, N, K);
+  return sp > 4 ? 4 : sp;
+}
+
+extern "C" hipError_t gemm_nt_bias_sk(const void *a, const void *b,
+                                      const float *bias, float *partial,
+                                      void *c, int M, int N, int K, long lda,
+                                      long ldb, long ldc, int splits,
+                                      hipStream_t s) {
+  if (splits <= 1 || !use_pipemix() || ldc != N)
+    return gemm_nt_bias(a, b, bias, c, M, N, K, lda, ldb, ldc, s);
+  NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
+  NtPipe<PlainNtSrc> sb{{(const uint16_t *)b, ldb, N, K}};
+  hipError_t e = launch_pipe_mix_wr(sa, sb, partial, M, N, K,
+                                    LinearWriter{ldc}, ldc, true, s, splits);
+  if (e != hipSuccess) return e;
+  return splitk_bias_reduce(partial, splits, (long)M * N, N, bias, c, s);
+}
+
 // FFN fc1 fused forward: g = gelu(x·w1ᵀ + b); the epilogue also saves
 // gelu'(h) (bf16, layout of C) so the backward epilogue is exp-free
 // (GeluBiasWriter). Both operands NT k-contiguous.
-extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
-                                        const float *bias, void *deriv, void *g,
-                                        int M, int N, int K, long lda,
-                                        long ldb, long ldc, hipStream_t s) {
-  GeluBiasWriter wrt{ldc, bias, (uint16_t *)deriv};
+// MPIAMD_GELU_DERIV (default 1): fwd saves gelu'(h) so bwd is exp-free;
+// =0 restores the save-pre/recompute-tanh pair for A/B.
+static bool gelu_deriv_mode() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_GELU_DERIV");
+    return !(e && e[0] == '0');
+  }();
+  return on;
+}
+
+template <class WR>
+static hipError_t gelu_fwd_route(WR wrt, const void *x, const void *w, void *g,
+                                 int M, int N, int K, long lda, long ldb,
+                                 long ldc, hipStream_t s) {
   GemmLoader la{(const uint16_t *)x, M, lda, K};
   GemmLoader lb{(const uint16_t *)w, N, ldb, K};
   if (M % 256 == 0 && N % 256 == 0 && K % 64 == 0 && ldc == N &&
@@ -57,13 +99,23 @@ extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
   return launch_mix_gemm_wr(ga, gb, g, M, N, K, wrt, ldc, false, s);
 }
 
+extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
+                                        const float *bias, void *deriv, void *g,
+                                        int M, int N, int K, long lda,
+                                        long ldb, long ldc, hipStream_t s) {
+  if (gelu_deriv_mode())
+    return gelu_fwd_route(GeluBiasWriter{ldc, bias, (uint16_t *)deriv}, x, w,
+                          g, M, N, K, lda, ldb, ldc, s);
+  return gelu_fwd_route(GeluBiasPreWriter{ldc, bias, (uint16_t *)deriv}, x, w,
+                        g, M, N, K, lda, ldb, ldc, s);
+}
+
 // FFN backward: dh = (dy·w2) ⊙ gelu'(h) — the fc2-dx GEMM multiplying by
 // the SAVED derivative in the epilogue (GeluBwdWriter); A NT, B (w2) TN.
-extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
-                                         const void *deriv, void *dh, int M,
-                                         int N, int K, long lda, long ldb,
-                                         long ldc, hipStream_t s) {
-  GeluBwdWriter wrt{ldc, (const uint16_t *)deriv};
+template <class WR>
+static hipError_t gelubwd_route(WR wrt, const void *dy, const void *w,
+                                void *dh, int M, int N, int K, long lda,
+                                long ldb, long ldc, hipStream_t s) {
   if (use_pipemix() && N % 8 == 0) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, ldb, K, N}};
@@ -73,6 +125,17 @@ extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
   TnRowMajor lb{(const uint16_t *)w, ldb, K, N};
   return launch_mix_gemm_wr(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
                             dh, M, N, K, wrt, ldc, false, s);
+}
+
+extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
+                                         const void *deriv, void *dh, int M,
+                                         int N, int K, long lda, long ldb,
+                                         long ldc, hipStream_t s) {
+  if (gelu_deriv_mode())
+    return gelubwd_route(GeluBwdWriter{ldc, (const uint16_t *)deriv}, dy, w,
+                         dh, M, N, K, lda, ldb, ldc, s);
+  return gelubwd_route(GeluBwdPreWriter{ldc, (const uint16_t *)deriv}, dy, w,
+                       dh, M, N, K, lda, ldb, ldc, s);
 }
 
 // C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx

@@ -241,6 +241,43 @@ struct GeluBwdWriter {
   }
 };
 
+// legacy pair (MPIAMD_GELU_DERIV=0): forward saves the PRE-ACTIVATION and
+// backward recomputes gelu' (one tanh per element in the bwd epilogue)
+struct GeluBiasPreWriter {
+  static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
+  long ldc;
+  const float *bias;
+  uint16_t *pre; // h (pre-activation), bf16
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    float h = v + bias[col];
+    pre[b + col] = f2bf(h);
+    p[b + col] = gelu_f(h);
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    float h = v + bias[col];
+    pre[b + col] = f2bf(h);
+    p[b + col] = f2bf(gelu_f(h));
+  }
+};
+
+struct GeluBwdPreWriter {
+  static constexpr bool ACC = false;
+  static constexpr bool STATS = false;
+  long ldc;
+  const uint16_t *pre;
+  typedef long RowCtx;
+  DEV_INLINE RowCtx row_ctx(int row) const { return (long)row * ldc; }
+  DEV_INLINE void store_f32(float *p, RowCtx b, int col, float v) const {
+    p[b + col] = v * dgelu_f(bf2f(pre[b + col]));
+  }
+  DEV_INLINE void store_bf16(uint16_t *p, RowCtx b, int col, float v) const {
+    p[b + col] = f2bf(v * dgelu_f(bf2f(pre[b + col])));
+  }
+};
+
 // Conv-forward epilogue that ALSO produces the BatchNorm per-channel
 // partial statistics (Σy, Σy²) — removes bn_partials' full re-read of the
 // activation it just wrote (~9% of a ResNet step at 8 TB/s).
