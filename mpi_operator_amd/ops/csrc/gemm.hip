@@ -39,13 +39,14 @@ extern "C" hipError_t gemm_nt_bias(const void *a, const void *b,
 
 // Forward split-K (fc2-class M=4096/N=1024 shapes give 256 pipe-mix
 // workgroups = exactly 1/CU, half the block slots idle; deep K amortizes
-// the fp32 slab + bias-reduce pass). Gate MPIAMD_FWD_SK; heuristic shared
-// with the dw path.
+// the fp32 slab + bias-reduce pass). Measured same-box +0.7% on BERT-Large
+// bs32 (1372 -> 1381): default ON, MPIAMD_FWD_SK=0 disables. Heuristic
+// shared with the dw path.
 extern "C" int gemm_tn_tn_splits(int M, int N, int K); // below
 extern "C" int gemm_fwd_splits(int M, int N, int K) {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_FWD_SK");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   if (!on || N % 8) return 1;
   int sp = gemm_tn_tn_splits(M, N, K);
