@@ -175,6 +175,41 @@ __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
   k3[c] = g_is * s1 * inv_m;
 }
 
+// bwd finalize, bands layout: one 256-thread block per channel striding
+// the slab rows — the lane8 version gave each channel only 32 lanes and
+// just C/8 blocks of parallelism (8 blocks at C=64), and its strided
+// cross-XCD slab reads ran ~8 us/call x 104 BN layers x both directions.
+__global__ void bn_finalize_bwd_bands_k(
+    const float *__restrict__ partial, int grid, int C,
+    const float *__restrict__ gamma, const float *__restrict__ invstd,
+    float inv_m, float *__restrict__ dbeta, float *__restrict__ dgamma,
+    float *__restrict__ k1, float *__restrict__ k2, float *__restrict__ k3) {
+  int c = blockIdx.x;
+  if (c >= C) return;
+  float s0 = 0.f, s1 = 0.f;
+  for (int g = threadIdx.x; g < grid; g += 256) {
+    s0 += partial[(long)g * 2 * C + c];
+    s1 += partial[(long)g * 2 * C + C + c];
+  }
+  __shared__ float red[2][256 / WAVE];
+  s0 = wave_sum(s0);
+  s1 = wave_sum(s1);
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    red[0][threadIdx.x / WAVE] = s0;
+    red[1][threadIdx.x / WAVE] = s1;
+  }
+  __syncthreads();
+  if (threadIdx.x != 0) return;
+  s0 = red[0][0] + red[0][1] + red[0][2] + red[0][3];
+  s1 = red[1][0] + red[1][1] + red[1][2] + red[1][3];
+  dbeta[c] = s0;
+  dgamma[c] = s1;
+  float g_is = gamma[c] * invstd[c];
+  k1[c] = g_is;
+  k2[c] = g_is * s0 * inv_m;
+  k3[c] = g_is * s1 * inv_m;
+}
+
 // apply scale/shift (+optional residual add) (+ReLU): fwd-train, fwd-eval
 // and the bottleneck-join fusion (res != nullptr folds the skip connection
 // into this pass — one fewer full activation read+write per block).
@@ -280,18 +315,30 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
   }
 }
 
-// Partials grid cap: PMC showed 75-79% wave-parked at 512 blocks, but the
-// 1024-block experiment REGRESSED ResNet101 same-box (3212 vs 3440 img/s,
-// repeatable) — the doubled slab write+finalize read costs more than the
-// extra latency hiding buys. 512 stays; MPIAMD_BN_GRID for future A/Bs.
-static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
-  static const long cap = [] {
+// Partials grid cap, CHANNEL-AWARE: the flat 512->1024 experiment lost
+// 6.6% because for C=2048 layers the fp32 slab ([g][2][C]) outgrew the
+// activation being reduced; for C=64 layers 512 blocks left the stats
+// kernels 75-79% wave-parked. Scale the cap so the slab stays ~2 MB-class:
+// C8=8 -> 4096 blocks, C8=32 -> 2048, C8=64 -> 1024, C8>=128 -> 512.
+// MPIAMD_BN_GRID overrides for A/Bs. Bindings size the slab with
+// bn_grid_cap — keep them in sync.
+extern "C" int bn_grid_cap(int C8) {
+  static const long ovr = [] {
     const char *e = getenv("MPIAMD_BN_GRID");
     long v = e ? atol(e) : 0;
-    return (v >= 1 && v <= 1024) ? v : 512L;
+    return (v >= 1 && v <= 8192) ? v : 0;
   }();
+  if (ovr) return (int)ovr;
+  long cap = 65536 / (C8 < 1 ? 1 : C8);
+  if (cap < 512) cap = 512;
+  if (cap > 4096) cap = 4096;
+  return (int)cap;
+}
+
+static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
   rows_per_block = 256 / C8;
   long g = (M + rows_per_block - 1) / rows_per_block;
+  long cap = bn_grid_cap(C8);
   grid = (int)(g > cap ? cap : (g < 1 ? 1 : g));
 }
 
@@ -314,7 +361,7 @@ extern "C" hipError_t bn_fwd_train_launch(
                                         nullptr, nullptr, partial, M, C8, 0);
   HIP_KERNEL_CHECK();
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
-  bn_finalize_fwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+  bn_finalize_fwd_bands_k<<<C, 256, 0, s>>>(
       partial, grid, C, gamma, beta, 1.f / (float)M, eps, mean, invstd, scale,
       shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
@@ -418,7 +465,7 @@ extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
                                         (const ushort8 *)y, mean, invstd,
                                         partial, M, C8, relu);
   HIP_KERNEL_CHECK();
-  bn_finalize_bwd_k<<<cdiv_h((long)C * 32, 256), 256, 0, s>>>(
+  bn_finalize_bwd_bands_k<<<C, 256, 0, s>>>(
       partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
   HIP_KERNEL_CHECK();
   bn_bwd_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(

@@ -78,6 +78,7 @@ struct SgdDesc {
 };
 hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
                            hipStream_t);
+int bn_grid_cap(int);
 hipError_t bn_fwd_train_launch(const void *, const void *, const float *,
                                const float *, float, int, void *, float *,
                                float *, float *, float *, float *, float *,
@@ -344,7 +345,7 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
   Tensor y = empty_cl_bf16(N, C, H, W, x);
   Tensor mean = at::empty({C}, f32), invstd = at::empty({C}, f32);
   Tensor scale = at::empty({C}, f32), shift = at::empty({C}, f32);
-  Tensor partial = at::empty({1024L * 2 * C}, f32);
+  Tensor partial = at::empty({(long)bn_grid_cap(C / 8) * 2 * C}, f32);
   float *rm = running_mean.defined() && running_mean.numel() == C
                   ? running_mean.data_ptr<float>() : nullptr;
   float *rv = rm ? running_var.data_ptr<float>() : nullptr;
@@ -387,7 +388,7 @@ static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
   Tensor dx = empty_cl_bf16(N, C, H, W, x);
   Tensor dgamma = at::empty({C}, f32), dbeta = at::empty({C}, f32);
   Tensor k1 = at::empty({C}, f32), k2 = at::empty({C}, f32), k3 = at::empty({C}, f32);
-  Tensor partial = at::empty({1024L * 2 * C}, f32);
+  Tensor partial = at::empty({(long)bn_grid_cap(C / 8) * 2 * C}, f32);
   CHK(bn_bwd_launch(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
                     invstd.data_ptr<float>(), relu ? 1 : 0, dx.data_ptr(),
