@@ -175,6 +175,15 @@ __global__ void bn_finalize_bwd_k(const float *__restrict__ partial, int grid,
   k3[c] = g_is * s1 * inv_m;
 }
 
+__global__ void bn_finalize_fwd_bands_k(
+    const float *__restrict__ partial, int grid, int C,
+    const float *__restrict__ gamma, const float *__restrict__ beta,
+    float inv_m, float eps, float *__restrict__ mean,
+    float *__restrict__ invstd, float *__restrict__ scale,
+    float *__restrict__ shift, float *__restrict__ running_mean,
+    float *__restrict__ running_var, float momentum,
+    float unbias); // defined below (shared with the conv-epilogue pre path)
+
 // bwd finalize, bands layout: one 256-thread block per channel striding
 // the slab rows — the lane8 version gave each channel only 32 lanes and
 // just C/8 blocks of parallelism (8 blocks at C=64), and its strided
