@@ -324,24 +324,20 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
   }
 }
 
-// Partials grid cap, CHANNEL-AWARE: the flat 512->1024 experiment lost
-// 6.6% because for C=2048 layers the fp32 slab ([g][2][C]) outgrew the
-// activation being reduced; for C=64 layers 512 blocks left the stats
-// kernels 75-79% wave-parked. Scale the cap so the slab stays ~2 MB-class:
-// C8=8 -> 4096 blocks, C8=32 -> 2048, C8=64 -> 1024, C8>=128 -> 512.
-// MPIAMD_BN_GRID overrides for A/Bs. Bindings size the slab with
-// bn_grid_cap — keep them in sync.
+// Partials grid cap: 512 measured best THREE times now — flat 1024 lost
+// 6.6%, and the channel-aware cap (up to 4096 bands for narrow layers)
+// lost 4.4% (3363 vs 3517 same-box) even with the bands finalize: the
+// deeper slab costs more in finalize reads + L2 pressure than the extra
+// partials parallelism buys. MPIAMD_BN_GRID overrides for A/Bs; bindings
+// size the slab with bn_grid_cap — keep them in sync.
 extern "C" int bn_grid_cap(int C8) {
   static const long ovr = [] {
     const char *e = getenv("MPIAMD_BN_GRID");
     long v = e ? atol(e) : 0;
     return (v >= 1 && v <= 8192) ? v : 0;
   }();
-  if (ovr) return (int)ovr;
-  long cap = 65536 / (C8 < 1 ? 1 : C8);
-  if (cap < 512) cap = 512;
-  if (cap > 4096) cap = 4096;
-  return (int)cap;
+  (void)C8;
+  return ovr ? (int)ovr : 512;
 }
 
 static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {

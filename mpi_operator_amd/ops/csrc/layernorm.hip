@@ -8,12 +8,18 @@
 // existing splitk_reduce kernel (atomics on N addresses would serialize).
 #include "common.h"
 
-// y = (x - mean) * rstd * gamma + beta;  saves mean/rstd per row
+// y = (x - mean) * rstd * gamma + beta;  saves mean/rstd per row.
+// C8T: compile-time C8 specialization (0 = runtime). The runtime-bounded
+// per-lane octet loops don't unroll (same class of loss as the splitk /
+// attention / SGD kernels), leaving only 2 loads in flight per wave.
+template <int C8T>
 __global__ void ln_fwd_k(const ushort8 *__restrict__ x,
                          const float *__restrict__ gamma,
                          const float *__restrict__ beta,
                          ushort8 *__restrict__ y, float *__restrict__ mean,
-                         float *__restrict__ rstd, long M, int C8, float eps) {
+                         float *__restrict__ rstd, long M, int C8rt,
+                         float eps) {
+  const int C8 = C8T ? C8T : C8rt;
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
   int waves = blockDim.x >> 6;
@@ -64,12 +70,14 @@ __global__ void ln_fwd_k(const ushort8 *__restrict__ x,
 // which capped the grid at 32 blocks (512 waves) and left the dx pass
 // latency-bound — the measured reason fused LN lost to torch native.
 // dgamma/dbeta moved to the column-parallel kernel below.
+template <int C8T>
 __global__ void ln_bwd_dx_k(const ushort8 *__restrict__ dy,
                             const ushort8 *__restrict__ x,
                             const float *__restrict__ gamma,
                             const float *__restrict__ mean,
                             const float *__restrict__ rstd,
-                            ushort8 *__restrict__ dx, long M, int C8) {
+                            ushort8 *__restrict__ dx, long M, int C8rt) {
+  const int C8 = C8T ? C8T : C8rt;
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
   int waves = blockDim.x >> 6;
@@ -175,14 +183,16 @@ __global__ void ln_gb_partials_k(const ushort8 *__restrict__ dy,
 // reads it as the LN input), y = LN(s). Replaces the separate residual
 // add's extra read+write of the [M][N] activation (BERT runs 2 joins per
 // layer per direction).
+template <int C8T>
 __global__ void ln_fwd_add_k(const ushort8 *__restrict__ a,
                              const ushort8 *__restrict__ b,
                              const float *__restrict__ gamma,
                              const float *__restrict__ beta,
                              ushort8 *__restrict__ sum_out,
                              ushort8 *__restrict__ y, float *__restrict__ mean,
-                             float *__restrict__ rstd, long M, int C8,
+                             float *__restrict__ rstd, long M, int C8rt,
                              float eps) {
+  const int C8 = C8T ? C8T : C8rt;
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
   int waves = blockDim.x >> 6;
@@ -265,9 +275,18 @@ extern "C" hipError_t ln_fwd(const void *x, const float *gamma,
                              float *rstd, long M, int N, float eps,
                              hipStream_t s) {
   if (N % 8) return hipErrorInvalidValue;
-  ln_fwd_k<<<ln_grid(M, 4), 256, 0, s>>>((const ushort8 *)x, gamma, beta,
-                                         (ushort8 *)y, mean, rstd, M, N / 8,
-                                         eps);
+  if (N == 1024)
+    ln_fwd_k<128><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)x, gamma, beta, (ushort8 *)y, mean, rstd, M, N / 8,
+        eps);
+  else if (N == 768)
+    ln_fwd_k<96><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)x, gamma, beta, (ushort8 *)y, mean, rstd, M, N / 8,
+        eps);
+  else
+    ln_fwd_k<0><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)x, gamma, beta, (ushort8 *)y, mean, rstd, M, N / 8,
+        eps);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -278,9 +297,18 @@ extern "C" hipError_t ln_fwd_add(const void *a, const void *b,
                                  float *rstd, long M, int N, float eps,
                                  hipStream_t s) {
   if (N % 8) return hipErrorInvalidValue;
-  ln_fwd_add_k<<<ln_grid(M, 4), 256, 0, s>>>(
-      (const ushort8 *)a, (const ushort8 *)b, gamma, beta,
-      (ushort8 *)sum_out, (ushort8 *)y, mean, rstd, M, N / 8, eps);
+  if (N == 1024)
+    ln_fwd_add_k<128><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)a, (const ushort8 *)b, gamma, beta,
+        (ushort8 *)sum_out, (ushort8 *)y, mean, rstd, M, N / 8, eps);
+  else if (N == 768)
+    ln_fwd_add_k<96><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)a, (const ushort8 *)b, gamma, beta,
+        (ushort8 *)sum_out, (ushort8 *)y, mean, rstd, M, N / 8, eps);
+  else
+    ln_fwd_add_k<0><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)a, (const ushort8 *)b, gamma, beta,
+        (ushort8 *)sum_out, (ushort8 *)y, mean, rstd, M, N / 8, eps);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -297,9 +325,18 @@ extern "C" hipError_t ln_bwd(const void *dy, const void *x, const float *gamma,
                              int N, int *grid_out, hipStream_t s) {
   if (N % 8 || N > 2048) return hipErrorInvalidValue;
   int C8 = N / 8;
-  ln_bwd_dx_k<<<ln_grid(M, 4), 256, 0, s>>>(
-      (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
-      (ushort8 *)dx, M, C8);
+  if (N == 1024)
+    ln_bwd_dx_k<128><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
+        (ushort8 *)dx, M, C8);
+  else if (N == 768)
+    ln_bwd_dx_k<96><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
+        (ushort8 *)dx, M, C8);
+  else
+    ln_bwd_dx_k<0><<<ln_grid(M, 4), 256, 0, s>>>(
+        (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
+        (ushort8 *)dx, M, C8);
   HIP_KERNEL_CHECK();
   int rpb = 256 / C8;
   long g = (M + rpb - 1) / rpb;
