@@ -15,6 +15,7 @@ template <int WHAT> // 0: fwd stats (sum, sumsq); 1: bwd stats (dy*m, dy*m*xhat)
 __global__ void bn_partials_k(const ushort8 *__restrict__ x,
                               const ushort8 *__restrict__ dy,
                               const ushort8 *__restrict__ y,
+                              const uint8_t *__restrict__ mask,
                               const float *__restrict__ mean,
                               const float *__restrict__ invstd,
                               float *__restrict__ partial, // [grid][2][C]
@@ -59,12 +60,21 @@ __global__ void bn_partials_k(const ushort8 *__restrict__ x,
         bf8_to_f8(vdy, fdy);
         bf8_to_f8(vdy2, fdy2);
         if (relu) {
-          ushort8 vy = y[off];
-          ushort8 vy2 = has2 ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (mask) {
+            int m1 = mask[off], m2 = has2 ? mask[off2] : 0;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
-            if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+            for (int j = 0; j < 8; ++j) {
+              if (!(m1 & (1 << j))) fdy[j] = 0.f;
+              if (!(m2 & (1 << j))) fdy2[j] = 0.f;
+            }
+          } else {
+            ushort8 vy = y[off];
+            ushort8 vy2 = has2 ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+              if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+            }
           }
         }
 #pragma unroll
@@ -226,7 +236,9 @@ __global__ void bn_apply_k(const ushort8 *__restrict__ x,
                            const ushort8 *__restrict__ res,
                            const float *__restrict__ scale,
                            const float *__restrict__ shift,
-                           ushort8 *__restrict__ y, long M, int C8, int relu) {
+                           ushort8 *__restrict__ y,
+                           uint8_t *__restrict__ mask, // y>0 bits, [M][C8]
+                           long M, int C8, int relu) {
   int cb = threadIdx.x % C8;
   int row_lane = threadIdx.x / C8;
   int rows_per_block = blockDim.x / C8;
@@ -253,17 +265,27 @@ __global__ void bn_apply_k(const ushort8 *__restrict__ x,
       bf8_to_f8(w, g);
       bf8_to_f8(w2, g2);
     }
+    int m1 = 0, m2 = 0;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       f[j] = f[j] * sc[j] + sh[j] + g[j];
       f2[j] = f2[j] * sc[j] + sh[j] + g2[j];
       if (relu) {
+        // mask records the PRE-clamp sign so backward's dy gating matches
+        // the y>0 test it replaces (bf16-rounded y>0 iff f>0 here: rounding
+        // keeps sign and relu output is never negative)
+        if (f[j] > 0.f) m1 |= 1 << j;
+        if (f2[j] > 0.f) m2 |= 1 << j;
         f[j] = fmaxf(f[j], 0.f);
         f2[j] = fmaxf(f2[j], 0.f);
       }
     }
     y[off] = f8_to_bf8(f);
     if (r2 < M) y[off2] = f8_to_bf8(f2);
+    if (relu && mask) {
+      mask[off] = (uint8_t)m1;
+      if (r2 < M) mask[off2] = (uint8_t)m2;
+    }
   }
 }
 
@@ -271,6 +293,7 @@ __global__ void bn_apply_k(const ushort8 *__restrict__ x,
 __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
                                const ushort8 *__restrict__ x,
                                const ushort8 *__restrict__ y,
+                               const uint8_t *__restrict__ mask,
                                const float *__restrict__ mean,
                                const float *__restrict__ invstd,
                                const float *__restrict__ k1,
@@ -306,12 +329,21 @@ __global__ void bn_bwd_apply_k(const ushort8 *__restrict__ dy,
     bf8_to_f8(vdy2, fdy2);
     bf8_to_f8(vx2, fx2);
     if (relu) {
-      ushort8 vy = y[off];
-      ushort8 vy2 = r2 < M ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (mask) { // 1-byte relu mask replaces the 16-byte y re-read
+        int m1 = mask[off], m2 = r2 < M ? mask[off2] : 0;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
-        if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+        for (int j = 0; j < 8; ++j) {
+          if (!(m1 & (1 << j))) fdy[j] = 0.f;
+          if (!(m2 & (1 << j))) fdy2[j] = 0.f;
+        }
+      } else {
+        ushort8 vy = y[off];
+        ushort8 vy2 = r2 < M ? y[off2] : ushort8{0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          if (!(bf2f(vy[j]) > 0.f)) fdy[j] = 0.f;
+          if (!(bf2f(vy2[j]) > 0.f)) fdy2[j] = 0.f;
+        }
       }
     }
 #pragma unroll
@@ -355,15 +387,17 @@ static int bn_apply_grid(long M, int C8) {
 
 extern "C" hipError_t bn_fwd_train_launch(
     const void *x, const void *res, const float *gamma, const float *beta,
-    float eps, int relu, void *y, float *mean, float *invstd, float *scale,
-    float *shift, float *partial, float *running_mean, float *running_var,
-    float momentum, long M, int C, hipStream_t s) {
+    float eps, int relu, void *y, uint8_t *relu_mask, float *mean,
+    float *invstd, float *scale, float *shift, float *partial,
+    float *running_mean, float *running_var, float momentum, long M, int C,
+    hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
   bn_partials_k<0><<<grid, 256, 0, s>>>((const ushort8 *)x, nullptr, nullptr,
-                                        nullptr, nullptr, partial, M, C8, 0);
+                                        nullptr, nullptr, nullptr, partial, M,
+                                        C8, 0);
   HIP_KERNEL_CHECK();
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
   bn_finalize_fwd_bands_k<<<C, 256, 0, s>>>(
@@ -371,8 +405,8 @@ extern "C" hipError_t bn_fwd_train_launch(
       shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y, M,
-      C8, relu);
+      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y,
+      relu_mask, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -425,9 +459,9 @@ __global__ void bn_finalize_fwd_bands_k(
 extern "C" hipError_t bn_fwd_train_pre_launch(
     const float *pre_partial, int pre_grid, const void *x, const void *res,
     const float *gamma, const float *beta, float eps, int relu, void *y,
-    float *mean, float *invstd, float *scale, float *shift,
-    float *running_mean, float *running_var, float momentum, long M, int C,
-    hipStream_t s) {
+    uint8_t *relu_mask, float *mean, float *invstd, float *scale,
+    float *shift, float *running_mean, float *running_var, float momentum,
+    long M, int C, hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
@@ -436,8 +470,8 @@ extern "C" hipError_t bn_fwd_train_pre_launch(
       invstd, scale, shift, running_mean, running_var, momentum, unbias);
   HIP_KERNEL_CHECK();
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y, M,
-      C8, relu);
+      (const ushort8 *)x, (const ushort8 *)res, scale, shift, (ushort8 *)y,
+      relu_mask, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
@@ -450,32 +484,33 @@ extern "C" hipError_t bn_fwd_eval_launch(const void *x, const float *scale,
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
   bn_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)x, nullptr, scale, shift, (ushort8 *)y, M, C8, relu);
+      (const ushort8 *)x, nullptr, scale, shift, (ushort8 *)y, nullptr, M, C8,
+      relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }
 
 extern "C" hipError_t bn_bwd_launch(const void *dy, const void *x,
-                                    const void *y, const float *gamma,
-                                    const float *mean, const float *invstd,
-                                    int relu, void *dx, float *dgamma,
-                                    float *dbeta, float *k1, float *k2,
-                                    float *k3, float *partial, long M, int C,
-                                    hipStream_t s) {
+                                    const void *y, const uint8_t *relu_mask,
+                                    const float *gamma, const float *mean,
+                                    const float *invstd, int relu, void *dx,
+                                    float *dgamma, float *dbeta, float *k1,
+                                    float *k2, float *k3, float *partial,
+                                    long M, int C, hipStream_t s) {
   int C8 = C / 8;
   if (C8 < 1 || C8 > 256) return hipErrorInvalidValue;
   int grid, rpb;
   bn_geom(M, C8, grid, rpb);
   bn_partials_k<1><<<grid, 256, 0, s>>>((const ushort8 *)x, (const ushort8 *)dy,
-                                        (const ushort8 *)y, mean, invstd,
-                                        partial, M, C8, relu);
+                                        (const ushort8 *)y, relu_mask, mean,
+                                        invstd, partial, M, C8, relu);
   HIP_KERNEL_CHECK();
   bn_finalize_bwd_bands_k<<<C, 256, 0, s>>>(
       partial, grid, C, gamma, invstd, 1.f / (float)M, dbeta, dgamma, k1, k2, k3);
   HIP_KERNEL_CHECK();
   bn_bwd_apply_k<<<bn_apply_grid(M, C8), 256, 0, s>>>(
-      (const ushort8 *)dy, (const ushort8 *)x, (const ushort8 *)y, mean,
-      invstd, k1, k2, k3, (ushort8 *)dx, M, C8, relu);
+      (const ushort8 *)dy, (const ushort8 *)x, (const ushort8 *)y, relu_mask,
+      mean, invstd, k1, k2, k3, (ushort8 *)dx, M, C8, relu);
   HIP_KERNEL_CHECK();
   return hipSuccess;
 }

@@ -64,6 +64,8 @@ hipError_t add_bf16(const void *, const void *, void *, long, hipStream_t);
 hipError_t add_relu_bwd(const void *, const void *, void *, long, hipStream_t);
 hipError_t add_relu_bwd_add(const void *, const void *, const void *, void *,
                             long, hipStream_t);
+hipError_t add_relu_bwd_add_mask(const void *, const void *, const void *,
+                                 void *, long, hipStream_t);
 hipError_t gap_fwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t gap_bwd(const void *, void *, int, int, int, hipStream_t);
 hipError_t bias_add(void *, const float *, long, int, hipStream_t);
@@ -80,15 +82,17 @@ hipError_t sgd_step_launch(const SgdDesc *, int, int, float, float, float, int,
                            hipStream_t);
 int bn_grid_cap(int);
 hipError_t bn_fwd_train_launch(const void *, const void *, const float *,
-                               const float *, float, int, void *, float *,
+                               const float *, float, int, void *, uint8_t *,
                                float *, float *, float *, float *, float *,
-                               float *, float, long, int, hipStream_t);
+                               float *, float *, float, long, int,
+                               hipStream_t);
 hipError_t bn_fwd_eval_launch(const void *, const float *, const float *, int,
                               void *, long, int, hipStream_t);
 hipError_t bn_bwd_launch(const void *, const void *, const void *,
-                         const float *, const float *, const float *, int,
-                         void *, float *, float *, float *, float *, float *,
-                         float *, long, int, hipStream_t);
+                         const uint8_t *, const float *, const float *,
+                         const float *, int, void *, float *, float *,
+                         float *, float *, float *, float *, long, int,
+                         hipStream_t);
 hipError_t maxpool_fwd_launch(const void *, void *, uint8_t *, int, int, int,
                               int, int, int, int, int, int, hipStream_t);
 hipError_t maxpool_bwd_launch(const void *, const uint8_t *, void *, int, int,
@@ -141,9 +145,9 @@ hipError_t conv_fwd_bn(const void *, const void *, void *, int, int, int,
                        hipStream_t);
 hipError_t bn_fwd_train_pre_launch(const float *, int, const void *,
                                    const void *, const float *, const float *,
-                                   float, int, void *, float *, float *,
-                                   float *, float *, float *, float *, float,
-                                   long, int, hipStream_t);
+                                   float, int, void *, uint8_t *, float *,
+                                   float *, float *, float *, float *,
+                                   float *, float, long, int, hipStream_t);
 hipError_t conv_dgrad_1x1_acc(const void *, const void *, void *, long, int,
                               int, hipStream_t);
 hipError_t conv_wgrad_implicit(const void *, const void *, float *, void *,
@@ -323,13 +327,21 @@ static std::vector<Tensor> bn_fwd_train_pre(
     check_cl_bf16(res, "res");
     resp = res.data_ptr();
   }
+  Tensor mask;
+  uint8_t *mp = nullptr;
+  if (relu) { // 1 bit/element relu mask: backward skips the y re-read
+    mask = at::empty({M, (long)C / 8}, x.options().dtype(at::kByte));
+    mp = mask.data_ptr<uint8_t>();
+  } else {
+    mask = at::empty({0}, x.options().dtype(at::kByte));
+  }
   CHK(bn_fwd_train_pre_launch(
       slab.data_ptr<float>(), (int)((M + 63) / 64), x.data_ptr(), resp,
       gamma.data_ptr<float>(), beta.data_ptr<float>(), (float)eps,
-      relu ? 1 : 0, y.data_ptr(), mean.data_ptr<float>(),
+      relu ? 1 : 0, y.data_ptr(), mp, mean.data_ptr<float>(),
       invstd.data_ptr<float>(), scale.data_ptr<float>(),
       shift.data_ptr<float>(), rm, rv, (float)momentum, M, C, cur_stream()));
-  return {y, mean, invstd};
+  return {y, mean, invstd, mask};
 }
 
 static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
@@ -354,13 +366,21 @@ static std::vector<Tensor> bn_fwd_train(const Tensor &x, const Tensor &gamma,
     check_cl_bf16(res, "res");
     resp = res.data_ptr();
   }
+  Tensor mask;
+  uint8_t *mp = nullptr;
+  if (relu) { // 1 bit/element relu mask: backward skips the y re-read
+    mask = at::empty({M, (long)C / 8}, x.options().dtype(at::kByte));
+    mp = mask.data_ptr<uint8_t>();
+  } else {
+    mask = at::empty({0}, x.options().dtype(at::kByte));
+  }
   CHK(bn_fwd_train_launch(x.data_ptr(), resp, gamma.data_ptr<float>(),
                           beta.data_ptr<float>(), (float)eps, relu ? 1 : 0,
-                          y.data_ptr(), mean.data_ptr<float>(),
+                          y.data_ptr(), mp, mean.data_ptr<float>(),
                           invstd.data_ptr<float>(), scale.data_ptr<float>(),
                           shift.data_ptr<float>(), partial.data_ptr<float>(),
                           rm, rv, (float)momentum, M, C, cur_stream()));
-  return {y, mean, invstd};
+  return {y, mean, invstd, mask};
 }
 
 static Tensor bn_fwd_eval(const Tensor &x, const Tensor &scale,
@@ -378,7 +398,7 @@ static Tensor bn_fwd_eval(const Tensor &x, const Tensor &scale,
 static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
                                   const Tensor &y, const Tensor &gamma,
                                   const Tensor &mean, const Tensor &invstd,
-                                  bool relu) {
+                                  bool relu, const Tensor &mask) {
   check_cl_bf16(dy, "dy");
   check_cl_bf16(x, "x");
   const HIPDeviceGuard guard(x.device());
@@ -389,7 +409,9 @@ static std::vector<Tensor> bn_bwd(const Tensor &dy, const Tensor &x,
   Tensor dgamma = at::empty({C}, f32), dbeta = at::empty({C}, f32);
   Tensor k1 = at::empty({C}, f32), k2 = at::empty({C}, f32), k3 = at::empty({C}, f32);
   Tensor partial = at::empty({(long)bn_grid_cap(C / 8) * 2 * C}, f32);
-  CHK(bn_bwd_launch(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+  const uint8_t *mp =
+      mask.defined() && mask.numel() > 0 ? mask.data_ptr<uint8_t>() : nullptr;
+  CHK(bn_bwd_launch(dy.data_ptr(), x.data_ptr(), y.data_ptr(), mp,
                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
                     invstd.data_ptr<float>(), relu ? 1 : 0, dx.data_ptr(),
                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
@@ -871,6 +893,20 @@ static Tensor add_relu_bwd_add_b(const Tensor &dy, const Tensor &y,
   return out;
 }
 
+// join backward off the bn relu-mask (1 byte per 8 channels): dxt =
+// dx0 + dy·mask — drops the full y re-read
+static Tensor add_relu_bwd_add_mask_b(const Tensor &dy, const Tensor &mask,
+                                      const Tensor &dx0) {
+  const HIPDeviceGuard guard(dy.device());
+  TORCH_CHECK(dy.numel() == dx0.numel() && dy.numel() % 8 == 0 &&
+              mask.numel() == dy.numel() / 8 &&
+              mask.scalar_type() == at::kByte);
+  Tensor out = at::empty_like(dx0);
+  CHK(add_relu_bwd_add_mask(dy.data_ptr(), mask.data_ptr(), dx0.data_ptr(),
+                            out.data_ptr(), dy.numel(), cur_stream()));
+  return out;
+}
+
 // ------------------------- sgd -------------------------
 static void sgd_step(std::vector<Tensor> masters, std::vector<Tensor> grads,
                      std::vector<Tensor> momenta, std::vector<Tensor> outs,
@@ -942,6 +978,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_fwd", &add_relu_fwd_b);
   m.def("add_relu_bwd", &add_relu_bwd_b);
   m.def("add_relu_bwd_add", &add_relu_bwd_add_b);
+  m.def("add_relu_bwd_add_mask", &add_relu_bwd_add_mask_b);
   m.def("sgd_step", &sgd_step);
   m.def("gemm_nt", &gemm_nt_b);
   m.def("mfma_probe", [](const Tensor &a, const Tensor &b) {

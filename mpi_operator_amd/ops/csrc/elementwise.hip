@@ -108,6 +108,37 @@ __global__ void add_relu_bwd_add_k(const ushort8 *__restrict__ dy,
   }
 }
 
+// mask form: the 1-byte-per-octet relu mask (bn_apply_k) replaces the
+// 16-byte y re-read — the join backward drops from 3 reads to 2 + 1/16.
+__global__ void add_relu_bwd_add_mask_k(const ushort8 *__restrict__ dy,
+                                        const uint8_t *__restrict__ mask,
+                                        const ushort8 *__restrict__ dx0,
+                                        ushort8 *__restrict__ dxt, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8 vd = dy[i], v0 = dx0[i];
+    int m = mask[i];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = (m & (1 << j)) ? bf2f(vd[j]) : 0.f;
+      o[j] = f2bf(g + bf2f(v0[j]));
+    }
+    dxt[i] = o;
+  }
+}
+
+extern "C" hipError_t add_relu_bwd_add_mask(const void *dy, const void *mask,
+                                            const void *dx0, void *dxt,
+                                            long n, hipStream_t s) {
+  long n8 = n / 8;
+  add_relu_bwd_add_mask_k<<<ew_grid(n8), 256, 0, s>>>(
+      (const ushort8 *)dy, (const uint8_t *)mask, (const ushort8 *)dx0,
+      (ushort8 *)dxt, n8);
+  HIP_KERNEL_CHECK();
+  return hipSuccess;
+}
+
 extern "C" hipError_t add_relu_bwd_add(const void *dy, const void *y,
                                        const void *dx0, void *dxt, long n,
                                        hipStream_t s) {

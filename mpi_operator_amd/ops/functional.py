@@ -82,13 +82,14 @@ class BNReLUFn(torch.autograd.Function):
         if x.is_cuda:
             assert _cl(x), "bn_relu: GPU tensors must be channels_last"
             empty = torch.empty(0)
-            y, mean, invstd = hip_ext().bn_fwd_train(
+            y, mean, invstd, mask = hip_ext().bn_fwd_train(
                 x, gamma, beta, eps, relu,
                 running_mean if running_mean is not None else empty,
                 running_var if running_var is not None else empty, momentum,
                 empty)
         else:
             y, mean, invstd = ref.bn_relu_fwd_train(x, gamma, beta, eps, relu)
+            mask = torch.empty(0, dtype=torch.uint8)
             if running_mean is not None:
                 with torch.no_grad():
                     n = x.numel() / x.shape[1]
@@ -96,7 +97,7 @@ class BNReLUFn(torch.autograd.Function):
                     running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                     running_var.mul_(1 - momentum).add_(var * n / max(n - 1, 1), alpha=momentum)
         ctx.relu = relu
-        ctx.save_for_backward(x, y, gamma, mean, invstd)
+        ctx.save_for_backward(x, y, gamma, mean, invstd, mask)
         ctx.mark_non_differentiable(mean, invstd)
         # without this, autograd materializes a zero tensor for dmean and
         # dinvstd on EVERY backward (2 fill launches × every BN layer × step)
@@ -105,10 +106,11 @@ class BNReLUFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy, _dmean=None, _dinvstd=None):
-        x, y, gamma, mean, invstd = ctx.saved_tensors
+        x, y, gamma, mean, invstd, mask = ctx.saved_tensors
         if dy.is_cuda:
             dy = dy.contiguous(memory_format=torch.channels_last)
-            dx, dgamma, dbeta = hip_ext().bn_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
+            dx, dgamma, dbeta = hip_ext().bn_bwd(dy, x, y, gamma, mean,
+                                                 invstd, ctx.relu, mask)
         else:
             dx, dgamma, dbeta = ref.bn_relu_bwd(dy, x, y, gamma, mean, invstd, ctx.relu)
         return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, None,

@@ -67,7 +67,8 @@ def test_bn_fwd_bwd(C, relu):
     beta = torch.randn(C, device="cuda") * 0.2
     rm = torch.zeros(C, device="cuda")
     rv = torch.ones(C, device="cuda")
-    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, 1e-5, relu, rm, rv, 0.3, torch.empty(0))
+    y, mean, invstd, mask = ext.bn_fwd_train(x, gamma, beta, 1e-5, relu, rm,
+                                             rv, 0.3, torch.empty(0))
     yr, mr, ir = ref.bn_relu_fwd_train(x.float().cpu(), gamma.cpu(), beta.cpu(), 1e-5, relu)
     assert relerr(mean.cpu(), mr) < 1e-3
     assert relerr(invstd.cpu(), ir) < 1e-3
@@ -79,12 +80,16 @@ def test_bn_fwd_bwd(C, relu):
     assert relerr(rv.cpu(), 0.7 + 0.3 * var_b * n / (n - 1)) < 1e-3
 
     dy = rand_cl(N, C, H, W, seed=21)
-    dx, dgamma, dbeta = ext.bn_bwd(dy, x, y, gamma, mean, invstd, relu)
+    dx, dgamma, dbeta = ext.bn_bwd(dy, x, y, gamma, mean, invstd, relu, mask)
     dxr, dgr, dbr = ref.bn_relu_bwd(dy.float().cpu(), x.float().cpu(), yr,
                                     gamma.cpu(), mr, ir, relu)
     assert relerr(dbeta.cpu(), dbr) < 5e-3
     assert relerr(dgamma.cpu(), dgr) < 5e-3
     assert relerr(dx.cpu(), dxr) < 0.05
+    # mask-less backward (legacy y>0 path) must agree exactly
+    dx2, dg2, db2 = ext.bn_bwd(dy, x, y, gamma, mean, invstd, relu,
+                               torch.empty(0, dtype=torch.uint8))
+    assert torch.equal(dx, dx2) and torch.equal(dg2, dgamma)
 
 
 def test_bn_eval():
