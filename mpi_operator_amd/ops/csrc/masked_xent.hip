@@ -30,36 +30,52 @@ __global__ void masked_xent_fwd_k(const uint16_t *__restrict__ logits,
   const ushort8 *row8 = (const ushort8 *)(row + head);
   int V8 = (V - head) / 8;
   int tail0 = head + V8 * 8;
-  __shared__ float red[256 / WAVE];
-  float mx = -3.4e38f;
+  // ONLINE max+sum in one pass over the row (the two-pass form read the
+  // 250 MB vocab-scale logits twice — pure HBM waste): per-thread running
+  // (mx, sum) with a rescale when the max moves, then a wave/block combine
+  // that rescales each partial sum to the global max. Same logsumexp.
+  __shared__ float redm[256 / WAVE], reds[256 / WAVE];
+  float mx = -3.4e38f, sum = 0.f;
   for (int v = threadIdx.x; v < V8; v += blockDim.x) {
     float f[8];
     bf8_to_f8(row8[v], f);
+    float m8 = f[0];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) mx = fmaxf(mx, f[j]);
-  }
-  if (threadIdx.x < head) mx = fmaxf(mx, bf2f(row[threadIdx.x]));
-  for (int v = tail0 + threadIdx.x; v < V; v += blockDim.x)
-    mx = fmaxf(mx, bf2f(row[v]));
-  mx = wave_max(mx);
-  if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = mx;
-  __syncthreads();
-  mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
-  float sum = 0;
-  for (int v = threadIdx.x; v < V8; v += blockDim.x) {
-    float f[8];
-    bf8_to_f8(row8[v], f);
+    for (int j = 1; j < 8; ++j) m8 = fmaxf(m8, f[j]);
+    if (m8 > mx) {
+      sum *= __expf(mx - m8);
+      mx = m8;
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) sum += __expf(f[j] - mx);
   }
-  if (threadIdx.x < head) sum += __expf(bf2f(row[threadIdx.x]) - mx);
-  for (int v = tail0 + threadIdx.x; v < V; v += blockDim.x)
-    sum += __expf(bf2f(row[v]) - mx);
+  if (threadIdx.x < head) {
+    float f = bf2f(row[threadIdx.x]);
+    if (f > mx) {
+      sum *= __expf(mx - f);
+      mx = f;
+    }
+    sum += __expf(f - mx);
+  }
+  for (int v = tail0 + threadIdx.x; v < V; v += blockDim.x) {
+    float f = bf2f(row[v]);
+    if (f > mx) {
+      sum *= __expf(mx - f);
+      mx = f;
+    }
+    sum += __expf(f - mx);
+  }
+  float wm = wave_max(mx);
+  sum *= __expf(mx - wm);
   sum = wave_sum(sum);
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    redm[threadIdx.x / WAVE] = wm;
+    reds[threadIdx.x / WAVE] = sum;
+  }
   __syncthreads();
-  if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = sum;
-  __syncthreads();
-  sum = red[0] + red[1] + red[2] + red[3];
+  mx = fmaxf(fmaxf(redm[0], redm[1]), fmaxf(redm[2], redm[3]));
+  sum = reds[0] * __expf(redm[0] - mx) + reds[1] * __expf(redm[1] - mx) +
+        reds[2] * __expf(redm[2] - mx) + reds[3] * __expf(redm[3] - mx);
   if (threadIdx.x == 0) {
     stats[b * 2] = mx;
     stats[b * 2 + 1] = 1.f / sum;
