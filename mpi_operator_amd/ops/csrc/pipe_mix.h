@@ -546,7 +546,16 @@ static hipError_t launch_pipe_mix_wr(const SA &sa, const SB &sb, void *c,
   int kts = (nk + splits - 1) / splits;
   long split_stride = (long)M * ldc;
   int nwg = tiles_m * tiles_n;
-  int cpx = (nwg % 8 == 0 && nwg >= 32 && splits == 1) ? nwg / 8 : 0;
+  // XCD fold also applies under split-K: the 2-D grid dispatches x-fastest,
+  // so within each split slice consecutive tile ids still round-robin the
+  // XCDs and the fold restores contiguous tile bands per XCD L2.
+  // MPIAMD_SK_CPX=0 restores the old splits==1-only gate for A/B.
+  static const bool sk_cpx = [] {
+    const char *e = getenv("MPIAMD_SK_CPX");
+    return !(e && e[0] == '0');
+  }();
+  int cpx = (nwg % 8 == 0 && nwg >= 32 && (splits == 1 || sk_cpx))
+                ? nwg / 8 : 0;
   dim3 grid(nwg, splits);
   if (c_f32)
     pipe_mix_k<SA, SB, true, WR><<<grid, PM_THREADS, 0, s>>>(
