@@ -125,7 +125,8 @@ hipError_t gemm_nt_gelu_bias(const void *, const void *, const float *,
                              hipStream_t);
 hipError_t gemm_nt_tn_gelubwd(const void *, const void *, const void *,
                               void *, int, int, int, long, long, long,
-                              hipStream_t);
+                              void *, hipStream_t);
+int gemm_gelubwd_wants_wt(int, int, int);
 hipError_t gemm_tn_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
 int gemm_tn_tn_splits(int, int, int);
@@ -724,8 +725,14 @@ static Tensor linear_gelu_dgrad(const Tensor &dy, const Tensor &w,
   int M = dy.size(0), K = dy.size(1), N = w.size(1);
   TORCH_CHECK(pre.size(0) == M && pre.size(1) == N);
   Tensor dh = at::empty({M, N}, dy.options());
+  Tensor wt; // w^T scratch: unlocks the NT pipe256 route on full-tile shapes
+  void *wtp = nullptr;
+  if (gemm_gelubwd_wants_wt(M, N, K)) {
+    wt = at::empty({N, K}, dy.options());
+    wtp = wt.data_ptr();
+  }
   CHK(gemm_nt_tn_gelubwd(dy.data_ptr(), w.data_ptr(), pre.data_ptr(),
-                         dh.data_ptr(), M, N, K, K, N, N, cur_stream()));
+                         dh.data_ptr(), M, N, K, K, N, N, wtp, cur_stream()));
   return dh;
 }
 

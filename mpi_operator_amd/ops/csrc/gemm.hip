@@ -111,12 +111,28 @@ extern "C" hipError_t gemm_nt_gelu_bias(const void *x, const void *w,
                         g, M, N, K, lda, ldb, ldc, s);
 }
 
+extern "C" hipError_t transpose2d_bf16(const void *, void *, int, int, long,
+                                       hipStream_t); // defined below
+
 // FFN backward: dh = (dy·w2) ⊙ gelu'(h) — the fc2-dx GEMM multiplying by
 // the SAVED derivative in the epilogue (GeluBwdWriter); A NT, B (w2) TN.
 template <class WR>
 static hipError_t gelubwd_route(WR wrt, const void *dy, const void *w,
                                 void *dh, int M, int N, int K, long lda,
-                                long ldb, long ldc, hipStream_t s) {
+                                long ldb, long ldc, hipStream_t s,
+                                void *wt_buf = nullptr) {
+  // Full-256-tile shapes (fc2-dx: 4096x4096x1024) run 355 TF on the
+  // TN-staged pipe_mix but 550+ TF on the NT pipe256 — worth materializing
+  // w2^T once per call (~8 us for 33 MB) to take the NT route. wt_buf is
+  // caller-allocated [N][K] bf16.
+  if (wt_buf && M % 256 == 0 && N % 256 == 0 && K % 64 == 0 && ldc == N &&
+      ldb == K && (long)(M / 256) * (N / 256) >= 128) {
+    hipError_t e = transpose2d_bf16(w, wt_buf, K, N, K, s);
+    if (e != hipSuccess) return e;
+    GemmLoader la{(const uint16_t *)dy, M, lda, K};
+    GemmLoader lb{(const uint16_t *)wt_buf, N, K, K};
+    return launch_pipe256_wr(la, lb, dh, M, N, K, ldc, false, wrt, s);
+  }
   if (use_pipemix() && N % 8 == 0) {
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)dy, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)w, ldb, K, N}};
@@ -128,15 +144,26 @@ static hipError_t gelubwd_route(WR wrt, const void *dy, const void *w,
                             dh, M, N, K, wrt, ldc, false, s);
 }
 
+// wt_buf: optional [N][K] bf16 scratch enabling the transpose+pipe256 route
+extern "C" int gemm_gelubwd_wants_wt(int M, int N, int K) {
+  static const bool off = [] {
+    const char *e = getenv("MPIAMD_GELUBWD_WT");
+    return e && e[0] == '0';
+  }();
+  return !off && M % 256 == 0 && N % 256 == 0 && K % 64 == 0 &&
+         (long)(M / 256) * (N / 256) >= 128;
+}
+
 extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
                                          const void *deriv, void *dh, int M,
                                          int N, int K, long lda, long ldb,
-                                         long ldc, hipStream_t s) {
+                                         long ldc, void *wt_buf,
+                                         hipStream_t s) {
   if (gelu_deriv_mode())
     return gelubwd_route(GeluBwdWriter{ldc, (const uint16_t *)deriv}, dy, w,
-                         dh, M, N, K, lda, ldb, ldc, s);
+                         dh, M, N, K, lda, ldb, ldc, s, wt_buf);
   return gelubwd_route(GeluBwdPreWriter{ldc, (const uint16_t *)deriv}, dy, w,
-                       dh, M, N, K, lda, ldb, ldc, s);
+                       dh, M, N, K, lda, ldb, ldc, s, wt_buf);
 }
 
 // C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx

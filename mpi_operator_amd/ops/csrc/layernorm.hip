@@ -263,12 +263,15 @@ __global__ void ln_gb_reduce_k(const float *__restrict__ partial,
     out[c] = red[0] + red[1] + red[2] + red[3];
 }
 
-// MPIAMD_LN_SPEC=0 forces the runtime-C8 kernels (A/B: the compile-time
-// octet loops won bert-base +13% / bs8 +6% but bs32 per-kernel means moved)
+// LN C8 specialization: same-box A/B at BERT-Large bs32 showed the
+// compile-time octet loops LOSE 2.4% (1384 vs 1415; ln_bwd_dx 15.5->24.1
+// us) — the unrolled form costs more than the runtime loop at M=4096
+// despite lower VGPRs. Default OFF; MPIAMD_LN_SPEC=1 enables for
+// small-shape A/Bs.
 static bool ln_spec() {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_LN_SPEC");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   return on;
 }
