@@ -120,6 +120,8 @@ hipError_t gemm_nt_bias_sk(const void *, const void *, const float *, float *,
 int gemm_fwd_splits(int, int, int);
 hipError_t gemm_nt_tn(const void *, const void *, void *, int, int, int, long,
                       long, long, int, hipStream_t);
+hipError_t gemm_nt_tn_acc(const void *, const void *, void *, int, int, int,
+                          long, long, long, hipStream_t);
 hipError_t gemm_nt_gelu_bias(const void *, const void *, const float *,
                              void *, void *, int, int, int, long, long, long,
                              hipStream_t);
@@ -624,6 +626,20 @@ static Tensor linear_dgrad(const Tensor &dy, const Tensor &w) {
   return dx;
 }
 
+// dx accumulated in place into `acc` (acc += dy @ w) — residual-join
+// backward without the separate add pass
+static Tensor linear_dgrad_acc(const Tensor &dy, const Tensor &w,
+                               Tensor acc) {
+  const HIPDeviceGuard guard(dy.device());
+  Tensor dyc = dy.contiguous(), wc = w.contiguous();
+  int M = dyc.size(0), N = wc.size(0), K = wc.size(1);
+  TORCH_CHECK(N % 8 == 0, "linear_dgrad_acc requires out_features %8");
+  TORCH_CHECK(acc.is_contiguous() && acc.size(0) == M && acc.size(1) == K);
+  CHK(gemm_nt_tn_acc(dyc.data_ptr(), wc.data_ptr(), acc.data_ptr(), M, K, N,
+                     N, K, K, cur_stream()));
+  return acc;
+}
+
 // dw/db only (FFN backward computes fc2's dx separately with the fused
 // dgelu epilogue — re-running the full linear_bwd would pay that GEMM twice)
 static std::vector<Tensor> linear_wgrad_only(const Tensor &dy,
@@ -971,6 +987,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("linear_dgrad", &linear_dgrad);
+  m.def("linear_dgrad_acc", &linear_dgrad_acc);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("masked_xent_fwd", &masked_xent_fwd);
   m.def("mlm_head_fwd", &mlm_head_fwd);

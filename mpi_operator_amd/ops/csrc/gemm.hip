@@ -144,13 +144,16 @@ static hipError_t gelubwd_route(WR wrt, const void *dy, const void *w,
                             dh, M, N, K, wrt, ldc, false, s);
 }
 
-// wt_buf: optional [N][K] bf16 scratch enabling the transpose+pipe256 route
+// wt_buf: optional [N][K] bf16 scratch enabling the transpose+pipe256
+// route. Measured a WASH same-box (1275 vs 1274 avg at bs32) despite the
+// per-kernel mix-vs-256 gap in prof10 — pipe256 with the deriv-reading
+// epilogue loses its edge. Default OFF (MPIAMD_GELUBWD_WT=1 enables).
 extern "C" int gemm_gelubwd_wants_wt(int M, int N, int K) {
-  static const bool off = [] {
+  static const bool on = [] {
     const char *e = getenv("MPIAMD_GELUBWD_WT");
-    return e && e[0] == '0';
+    return e && e[0] == '1';
   }();
-  return !off && M % 256 == 0 && N % 256 == 0 && K % 64 == 0 &&
+  return on && M % 256 == 0 && N % 256 == 0 && K % 64 == 0 &&
          (long)(M / 256) * (N / 256) >= 128;
 }
 
@@ -164,6 +167,24 @@ extern "C" hipError_t gemm_nt_tn_gelubwd(const void *dy, const void *w,
                          dh, M, N, K, lda, ldb, ldc, s, wt_buf);
   return gelubwd_route(GeluBwdPreWriter{ldc, (const uint16_t *)deriv}, dy, w,
                        dh, M, N, K, lda, ldb, ldc, s, wt_buf);
+}
+
+// dx ACCUMULATE form: C += A·B (LinearAccWriter) — the transformer
+// residual-join backward folds the dgrad into the LN-dx buffer instead of
+// a separate full-activation add pass (autograd's CUDAFunctor_add join).
+extern "C" hipError_t gemm_nt_tn_acc(const void *a, const void *b, void *c,
+                                     int M, int N, int K, long lda, long ldb,
+                                     long ldc, hipStream_t s) {
+  if (use_pipemix() && tn_cols_ok(N, ldb)) {
+    NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
+    TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
+    return launch_pipe_mix_wr(sa, sb, c, M, N, K, LinearAccWriter{ldc}, ldc,
+                              false, s);
+  }
+  GemmLoader la{(const uint16_t *)a, M, lda, K};
+  TnRowMajor lb{(const uint16_t *)b, ldb, K, N};
+  return launch_mix_gemm_wr(NtStage<GemmLoader>{la}, TnStage<TnRowMajor>{lb},
+                            c, M, N, K, LinearAccWriter{ldc}, ldc, false, s);
 }
 
 // C[M][N] = A[M][K-contig] · B(k-strided [K rows][N cols])  — linear dx
