@@ -137,6 +137,22 @@ class BertLayer(nn.Module):
         return ln(a + b)
 
     def forward(self, x, attn_mask=None):
+        if (x.is_cuda and x.dtype == torch.bfloat16 and attn_mask is None
+                and self.attn.head_dim == 64 and x.shape[1] == 128
+                and _os.environ.get("MPIAMD_LAYER_FUSED", "1") == "1"):
+            # whole-layer composite Function: both residual-join backward
+            # adds fold into dgrad GEMM epilogues (Fx.BertLayerFn)
+            b, s, h = x.shape
+            y = Fx.bert_layer(
+                x.reshape(-1, h).contiguous(), b, s, self.attn.heads,
+                self.attn.scale, self.ln1.eps,
+                self.attn.qkv.weight, self.attn.qkv.bias,
+                self.attn.out.weight, self.attn.out.bias,
+                self.ln1.weight.float(), self.ln1.bias.float(),
+                self.fc1.weight, self.fc1.bias,
+                self.fc2.weight, self.fc2.bias,
+                self.ln2.weight.float(), self.ln2.bias.float())
+            return y.view(b, s, h)
         x = self._join_ln(self.ln1, x, self.attn(x, attn_mask))
         if x.is_cuda and x.dtype == torch.bfloat16:
             # fused FFN: GELU lives in the GEMM epilogues (fwd emits

@@ -457,6 +457,78 @@ def attention(qkv, heads, scale, mask=None):
     return AttentionFn.apply(qkv, heads, scale, mask)
 
 
+class BertLayerFn(torch.autograd.Function):
+    """One whole transformer encoder layer (fused QKV → attention →
+    out-proj → LN-join → FFN(GELU) → LN-join) as a single autograd node,
+    chaining the HIP primitives manually (the ResNet BottleneckFn pattern).
+
+    What the composite buys over per-op autograd, per layer per step:
+      - both residual-join backward adds (autograd CUDAFunctor_add over the
+        [M, H] activation) fold into the dgrad GEMM epilogues
+        (linear_dgrad_acc: dx_join += dy·W in place);
+      - autograd bookkeeping for ~8 interior nodes disappears.
+    GPU-only; requires S == 128 (fused attention backward) and head_dim 64.
+    """
+
+    @staticmethod
+    def forward(ctx, x2d, b, s, heads, scale, eps,
+                qkv_w, qkv_b, out_w, out_b, ln1_w, ln1_b,
+                fc1_w, fc1_b, fc2_w, fc2_b, ln2_w, ln2_b):
+        ext = hip_ext()
+        qkv = ext.linear_fwd(x2d, qkv_w, qkv_b)
+        qkv5 = qkv.view(b, s, 3, heads, 64)
+        att, probs = ext.attn_fwd(qkv5, heads, scale, None, True)
+        att2 = att.view(b * s, heads * 64)
+        o = ext.linear_fwd(att2, out_w, out_b)
+        y1, s1, mean1, rstd1 = ext.layernorm_add_fwd(x2d, o, ln1_w, ln1_b,
+                                                     eps)
+        g, deriv = ext.linear_gelu_fwd(y1, fc1_w, fc1_b)
+        z = ext.linear_fwd(g, fc2_w, fc2_b)
+        y2, s2, mean2, rstd2 = ext.layernorm_add_fwd(y1, z, ln2_w, ln2_b,
+                                                     eps)
+        ctx.save_for_backward(x2d, qkv5, probs, att2, s1, mean1, rstd1,
+                              y1, g, deriv, s2, mean2, rstd2,
+                              qkv_w, out_w, ln1_w, fc1_w, fc2_w, ln2_w)
+        ctx.dims = (b, s, heads, scale)
+        return y2
+
+    @staticmethod
+    def backward(ctx, dy2):
+        (x2d, qkv5, probs, att2, s1, mean1, rstd1, y1, g, deriv,
+         s2, mean2, rstd2, qkv_w, out_w, ln1_w, fc1_w, fc2_w,
+         ln2_w) = ctx.saved_tensors
+        b, s, heads, scale = ctx.dims
+        ext = hip_ext()
+        dy2 = dy2.contiguous()
+        # LN2 join: dxs2 is the grad of (y1 + z) — it feeds BOTH the
+        # residual (y1) and the FFN branch; the FFN dx accumulates into it
+        dxs2, dg_ln2, db_ln2 = ext.layernorm_bwd(dy2, s2, ln2_w, mean2,
+                                                 rstd2)
+        dw2, db2 = ext.linear_wgrad_only(dxs2, g)
+        dh = ext.linear_gelu_dgrad(dxs2, fc2_w, deriv)
+        dw1, db1 = ext.linear_wgrad_only(dh, y1)
+        dy1 = ext.linear_dgrad_acc(dh, fc1_w, dxs2)  # dxs2 += dh @ w1
+        # LN1 join: same pattern with the attention branch
+        dxs1, dg_ln1, db_ln1 = ext.layernorm_bwd(dy1, s1, ln1_w, mean1,
+                                                 rstd1)
+        dow, dob = ext.linear_wgrad_only(dxs1, att2)
+        do_ = ext.linear_dgrad(dxs1, out_w)
+        dqkv = ext.attn_bwd(qkv5, do_, probs, scale)
+        dqkv2 = dqkv.view(b * s, -1)
+        dqw, dqb = ext.linear_wgrad_only(dqkv2, x2d)
+        dx = ext.linear_dgrad_acc(dqkv2, qkv_w, dxs1)  # dxs1 += dqkv @ Wqkv
+        return (dx, None, None, None, None, None,
+                dqw, dqb, dow, dob, dg_ln1, db_ln1,
+                dw1, db1, dw2, db2, dg_ln2, db_ln2)
+
+
+def bert_layer(x2d, b, s, heads, scale, eps, qkv_w, qkv_b, out_w, out_b,
+               ln1_w, ln1_b, fc1_w, fc1_b, fc2_w, fc2_b, ln2_w, ln2_b):
+    return BertLayerFn.apply(x2d, b, s, heads, scale, eps, qkv_w, qkv_b,
+                             out_w, out_b, ln1_w, ln1_b, fc1_w, fc1_b,
+                             fc2_w, fc2_b, ln2_w, ln2_b)
+
+
 class LayerNormAddFn(torch.autograd.Function):
     """y = LN(a + b): the transformer residual join fused into the LN
     forward (the separate add was an extra full read+write of the [M, H]

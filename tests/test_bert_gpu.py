@@ -321,3 +321,42 @@ def test_fused_attention_backward_matches_torch():
     err = (a.grad.float() - r.grad).abs().max().item()
     scale = r.grad.abs().max().item() + 1e-6
     assert err < 0.08 * scale + 2e-3, (err, scale)
+
+
+def test_composite_layer_matches_decomposed():
+    """BertLayerFn (whole-layer composite with accumulate-dgrad joins) vs
+    the per-op Fn path: same output and same grads for every param."""
+    import os
+    from mpi_operator_amd.models.bert import BertConfig, BertLayer, LayerNorm, BertLinear
+    torch.manual_seed(11)
+    cfg = BertConfig(hidden=512, layers=1, heads=8, intermediate=2048)
+    layer = BertLayer(cfg).to(device="cuda", dtype=torch.bfloat16)
+    for mod in layer.modules():
+        if isinstance(mod, LayerNorm):
+            mod.weight.data = mod.weight.data.float()
+            mod.bias.data = mod.bias.data.float()
+        elif isinstance(mod, BertLinear):
+            mod.bias.data = mod.bias.data.float()
+    x = ((torch.rand(2, 128, 512, device="cuda") * 2 - 1) * 0.5).to(torch.bfloat16)
+
+    def run(fused):
+        os.environ["MPIAMD_LAYER_FUSED"] = "1" if fused else "0"
+        for p in layer.parameters():
+            p.grad = None
+        xc = x.clone().requires_grad_()
+        y = layer(xc)
+        y.float().square().mean().backward()
+        grads = {n: p.grad.detach().float().clone()
+                 for n, p in layer.named_parameters()}
+        return y.detach().float(), xc.grad.detach().float().clone(), grads
+
+    y1, dx1, g1 = run(True)
+    y0, dx0, g0 = run(False)
+    os.environ.pop("MPIAMD_LAYER_FUSED", None)
+    assert (y1 - y0).abs().max().item() < 1e-5, "composite fwd differs"
+    rel = (dx1 - dx0).norm().item() / (dx0.norm().item() + 1e-30)
+    assert rel < 0.02, f"dx mismatch {rel}"
+    for n in g0:
+        num = (g1[n] - g0[n]).norm().item()
+        den = g0[n].norm().item() + 1e-30
+        assert num / den < 0.02, (n, num / den)
