@@ -139,9 +139,13 @@ class BertLayer(nn.Module):
     def forward(self, x, attn_mask=None):
         if (x.is_cuda and x.dtype == torch.bfloat16 and attn_mask is None
                 and self.attn.head_dim == 64 and x.shape[1] == 128
-                and _os.environ.get("MPIAMD_LAYER_FUSED", "1") == "1"):
-            # whole-layer composite Function: both residual-join backward
-            # adds fold into dgrad GEMM epilogues (Fx.BertLayerFn)
+                and _os.environ.get("MPIAMD_LAYER_FUSED", "0") == "1"):
+            # whole-layer composite Function (Fx.BertLayerFn): residual-join
+            # backward adds fold into accumulate-dgrad GEMM epilogues.
+            # Measured same-box 1400 vs 1461 seq/s AGAINST the per-op path
+            # at BERT-Large bs32 — the unsplit acc-dgrad GEMM (no split-K,
+            # extra C read in the epilogue) costs more than the two
+            # CUDAFunctor_add joins it removes. Default OFF.
             b, s, h = x.shape
             y = Fx.bert_layer(
                 x.reshape(-1, h).contiguous(), b, s, self.attn.heads,
