@@ -648,7 +648,7 @@ extern "C" hipError_t conv_wgrad_implicit(const void *dy, const void *x,
       e = launch_mix_gemm(TnStage<TnRowMajor>{la}, TnStage<TnRowMajor>{lb},
                           partial, Kout, RSC, (int)M, RSC, true, strm, splits);
     }
-  } else if (use_pipegather()) {
+  } else if (use_pipegather_wgrad()) {
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)dy, (long)Kout, (int)M, Kout}};
     TnPipe<XcolSrc> sb{
         {(const uint16_t *)x, H, W, C, HO, WO, S, stride, pad, RSC, M}};

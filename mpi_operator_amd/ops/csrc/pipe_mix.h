@@ -76,6 +76,18 @@ static inline bool use_pipegather() {
   return use_pipemix() && on;
 }
 
+// wgrad-only override: the -2.9% PIPEGATHER measurement bundled fwd/dgrad
+// and wgrad gathers; this lets the Xcol wgrad pipe be A/B'd alone (it is
+// the largest single conv kernel, 11% of a ResNet101 step, and its mix
+// counterpart shows 15.3% LDS bank conflicts in PMC).
+static inline bool use_pipegather_wgrad() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_PIPEGATHER_WGRAD");
+    return e && e[0] == '1';
+  }();
+  return (use_pipemix() && on) || use_pipegather();
+}
+
 // ---- SrcMaps ----------------------------------------------------------
 // concept (stateful: each thread stages the same 2 (row/col, k-offset)
 // granules every k-tile, so the expensive decomposition hoists into init):
