@@ -77,13 +77,13 @@ static inline bool use_pipegather() {
 }
 
 // wgrad-only override: the -2.9% PIPEGATHER measurement bundled fwd/dgrad
-// and wgrad gathers; this lets the Xcol wgrad pipe be A/B'd alone (it is
-// the largest single conv kernel, 11% of a ResNet101 step, and its mix
-// counterpart shows 15.3% LDS bank conflicts in PMC).
+// and wgrad gathers; A/B'd alone the Xcol wgrad pipe WINS (3566 -> 3573
+// img/s ResNet101, 6129 ResNet50 same-box) — default ON.
+// MPIAMD_PIPEGATHER_WGRAD=0 reverts to the register-staged mix gather.
 static inline bool use_pipegather_wgrad() {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_PIPEGATHER_WGRAD");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   return (use_pipemix() && on) || use_pipegather();
 }
