@@ -577,3 +577,207 @@ static hipError_t launch_pipe_mix_wr(const SA &sa, const SB &sb, void *c,
         sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, 0, zeros);
   return hipGetLastError();
 }
+
+// ======================================================================
+// 8-wave (512-thread) variant: same 128² tile, same LDS images and stage
+// schedule, but 2×4 waves of 64M×32N each — 2 blocks/CU become 16
+// waves/CU instead of 8 (the 4-wave form leaves half the wave slots
+// empty at its 64 KiB LDS footprint; PMC: 43-48% parked, 20-30% stall on
+// the dw/dx pool). Cost: B-frags are read by 2 waves each (LDS
+// read:MFMA 1.5 vs 1.0). A/B gated at the launchers (MPIAMD_PIPE8).
+// Per-thread VM-op counts HALVE (1 granule per stage instead of 2), so
+// every counted vmcnt is half the 4-wave kernel's.
+// ======================================================================
+
+template <class SRC> struct Nt8Pipe {
+  static constexpr bool TR = false;
+  SRC s;
+  DEV_INLINE void init(int tid, int base) {
+    int row = tid >> 2;
+    s.init(0, base + row, pm_swz(tid & 3, row) * 8);
+  }
+  DEV_INLINE void stage(int tid, int kb, ushort8 *img,
+                        const uint16_t *zeros) const {
+    const uint16_t *src = s.ptr16(0, kb);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void *)(src ? src : zeros),
+        (__attribute__((address_space(3))) void *)(img + tid), 16, 0, 0);
+  }
+  DEV_INLINE bf16x8 read(const ushort8 *img, int lane, int frag0, int ks) const {
+    int row = frag0 + (lane & 31);
+    int q = ks * 2 + (lane >> 5);
+    return us8_to_bf8v(img[row * 4 + pm_swz(q, row)]);
+  }
+};
+
+template <class SRC> struct Tn8Pipe {
+  static constexpr bool TR = true;
+  SRC s;
+  DEV_INLINE void init(int tid, int base) {
+    int st = tid >> 3;
+    int kq = st >> 3, cq = st & 7;
+    int kl = (tid & 7) >> 1, ch = tid & 1;
+    s.init(0, base + cq * 16 + ch * 8, kq * 4 + kl);
+  }
+  DEV_INLINE void stage(int tid, int kb, ushort8 *img,
+                        const uint16_t *zeros) const {
+    const uint16_t *src = s.ptr16(0, kb);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void *)(src ? src : zeros),
+        (__attribute__((address_space(3))) void *)(img + tid), 16, 0, 0);
+  }
+  DEV_INLINE bf16x8 read(const ushort8 *img, int lane, int frag0, int ks) const {
+    unsigned base = (unsigned)(unsigned long)(
+        __attribute__((address_space(3))) const void *)img;
+    int kq0 = ks * 4 + ((lane >> 5) & 1) * 2;
+    int cq = (frag0 >> 4) + ((lane >> 4) & 1);
+    unsigned a0 = base + (unsigned)((kq0 * 8 + cq) * 128 + (lane & 15) * 8);
+    uint2v_pm lo, hi;
+    asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                 "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
+                 "s_waitcnt lgkmcnt(0)"
+                 : "=&v"(lo), "=&v"(hi)
+                 : "v"(a0)
+                 : "memory");
+    union { unsigned u[4]; bf16x8 v; } r;
+    r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
+    return r.v;
+  }
+};
+
+template <class SA, class SB, bool C_F32, class WR>
+__global__ __launch_bounds__(512) void pipe_mix8_k(
+    SA sa, SB sb, void *__restrict__ cptr, int M, int N, int K, WR wrt,
+    int tiles_n, int kt_per_split, long split_stride, int xcd_cpx, int swap,
+    const uint16_t *__restrict__ zeros) {
+  int tile = swap ? blockIdx.y : blockIdx.x;
+  int split = swap ? blockIdx.x : blockIdx.y;
+  if (xcd_cpx) tile = (tile & 7) * xcd_cpx + (tile >> 3);
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * PM_BM, col0 = tn * PM_BM;
+  int tid = threadIdx.x;
+  int lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 2, wc = wave & 3; // 2×4: wave C = 64M × 32N
+
+  __shared__ __align__(128) ushort8 lds[8 * PM_HSZ];
+  if constexpr (WR::STATS) wrt.reset();
+#define PM8_IMG(buf, op, kh) (lds + (((buf) * 2 + (op)) * 2 + (kh)) * PM_HSZ)
+
+  float16v acc[2] = {};
+
+  int nk_total = (K + PM_BK - 1) / PM_BK;
+  int t0 = split * kt_per_split;
+  int nk = min(kt_per_split, nk_total - t0);
+  if (nk < 0) nk = 0;
+
+  sa.init(tid, row0);
+  sb.init(tid, col0);
+  auto stage_half = [&](int t, int kh) {
+    int kb = (t0 + t) * PM_BK + kh * 32;
+    sa.stage(tid, kb, PM8_IMG(t & 1, 0, kh), zeros);
+    sb.stage(tid, kb, PM8_IMG(t & 1, 1, kh), zeros);
+  };
+
+  if (nk > 0) {
+    // prologue proves BOTH tile-0 halves at the barrier (same race rule
+    // as the 4-wave kernel); per-thread op counts are HALVED here
+    stage_half(0, 0);
+    stage_half(0, 1);
+    if (nk > 1) {
+      stage_half(1, 0);
+      stage_half(1, 1);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      if (t > 0 || kh > 0) {
+        bool steady = (kh == 0) ? (t + 1 < nk) : (t + 2 < nk);
+        if (steady)
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      bf16x8 af[2][2], bf_[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          af[mi][ks] = sa.read(PM8_IMG(buf, 0, kh), lane, wr * 64 + mi * 32, ks);
+        bf_[ks] = sb.read(PM8_IMG(buf, 1, kh), lane, wc * 32, ks);
+      }
+      if (kh == 0) {
+        if (t > 0 && t + 1 < nk) stage_half(t + 1, 1);
+      } else {
+        if (t + 2 < nk) stage_half(t + 2, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          acc[mi] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[mi][ks], bf_[ks], acc[mi], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  cptr = (void *)((char *)cptr + split * split_stride * (C_F32 ? 4 : 2));
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      if (row >= M) continue;
+      typename WR::RowCtx rc = wrt.row_ctx(row);
+      int col = col0 + wc * 32 + (lane & 31);
+      if (col >= N) continue;
+      if (C_F32)
+        wrt.store_f32((float *)cptr, rc, col, acc[mi][r]);
+      else
+        wrt.store_bf16((uint16_t *)cptr, rc, col, acc[mi][r]);
+    }
+  }
+}
+#undef PM8_IMG
+
+static inline bool use_pipe8() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_PIPE8");
+    return e && e[0] == '1';
+  }();
+  return on;
+}
+
+template <class SA8, class SB8, class WR>
+static hipError_t launch_pipe_mix8_wr(const SA8 &sa, const SB8 &sb, void *c,
+                                      int M, int N, int K, const WR &wrt,
+                                      long ldc, bool c_f32, hipStream_t s,
+                                      int splits = 1) {
+  const uint16_t *zeros = pm_zeros_page();
+  if (!zeros) return hipErrorOutOfMemory;
+  int tiles_m = (M + PM_BM - 1) / PM_BM, tiles_n = (N + PM_BM - 1) / PM_BM;
+  int nk = (K + PM_BK - 1) / PM_BK;
+  if (splits > nk) splits = nk > 0 ? nk : 1;
+  int kts = (nk + splits - 1) / splits;
+  long split_stride = (long)M * ldc;
+  int nwg = tiles_m * tiles_n;
+  int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
+  dim3 grid(nwg, splits);
+  if (c_f32)
+    pipe_mix8_k<SA8, SB8, true, WR><<<grid, 512, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, 0, zeros);
+  else
+    pipe_mix8_k<SA8, SB8, false, WR><<<grid, 512, 0, s>>>(
+        sa, sb, c, M, N, K, wrt, tiles_n, kts, split_stride, cpx, 0, zeros);
+  return hipGetLastError();
+}
