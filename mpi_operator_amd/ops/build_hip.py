@@ -49,7 +49,10 @@ def _torch_flags():
 def _stale(obj: str, src: str) -> bool:
     if not os.path.exists(obj):
         return True
-    dep = [src] + [os.path.join(CSRC, h) for h in ("common.h", "mfma_tile.h", "mix_gemm.h", "gemm256.h")]
+    # EVERY header in csrc counts: conv.o once went stale against a
+    # pipe_mix.h-only edit because this list was hand-maintained
+    import glob as _glob
+    dep = [src] + _glob.glob(os.path.join(CSRC, "*.h"))
     om = os.path.getmtime(obj)
     return any(os.path.getmtime(d) > om for d in dep if os.path.exists(d))
 

@@ -228,18 +228,6 @@ extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
                                     int K, long lda, long ldb, long ldc,
                                     int splits, hipStream_t s) {
   if (use_pipemix() && tn_cols_ok(N, ldb)) {
-    if (use_pipe8()) {
-      Nt8Pipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
-      Tn8Pipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
-      if (splits <= 1)
-        return launch_pipe_mix8_wr(sa, sb, c, M, N, K, LinearWriter{ldc},
-                                   ldc, false, s);
-      hipError_t e = launch_pipe_mix8_wr(sa, sb, partial, M, N, K,
-                                         LinearWriter{ldc}, ldc, true, s,
-                                         splits);
-      if (e != hipSuccess) return e;
-      return splitk_reduce(partial, splits, (long)M * ldc, c, 1, s);
-    }
     NtPipe<PlainNtSrc> sa{{(const uint16_t *)a, lda, M, K}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)
@@ -313,18 +301,6 @@ extern "C" hipError_t gemm_tn_tn_sk(const void *a, const void *b,
   // torch .to(bf16) cast afterwards was ~200 cast kernels / 1.3 ms per
   // BERT-Large step, prof8)
   if (use_pipemix() && tn_cols_ok(M, lda) && tn_cols_ok(N, ldb)) {
-    if (use_pipe8()) {
-      Tn8Pipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
-      Tn8Pipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
-      if (splits <= 1)
-        return launch_pipe_mix8_wr(sa, sb, c, M, N, K, LinearWriter{ldc},
-                                   ldc, out_bf16 == 0, s);
-      hipError_t e = launch_pipe_mix8_wr(sa, sb, partial, M, N, K,
-                                         LinearWriter{ldc}, ldc, true, s,
-                                         splits);
-      if (e != hipSuccess) return e;
-      return splitk_reduce(partial, splits, (long)M * ldc, c, out_bf16, s);
-    }
     TnPipe<PlainTnSrc> sa{{(const uint16_t *)a, lda, K, M}};
     TnPipe<PlainTnSrc> sb{{(const uint16_t *)b, ldb, K, N}};
     if (splits <= 1)

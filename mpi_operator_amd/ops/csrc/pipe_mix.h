@@ -545,11 +545,32 @@ inline const uint16_t *pm_zeros_page() {
   return pages[dev];
 }
 
+// ---- 8-wave variant plumbing (definitions at end of file) ----
+template <class SRC> struct Nt8Pipe;
+template <class SRC> struct Tn8Pipe;
+template <class P> struct Pipe8Of;
+template <class S> struct Pipe8Of<NtPipe<S>> { using type = Nt8Pipe<S>; };
+template <class S> struct Pipe8Of<TnPipe<S>> { using type = Tn8Pipe<S>; };
+static inline bool use_pipe8();
+template <class SA8, class SB8, class WR>
+static hipError_t launch_pipe_mix8_wr(const SA8 &sa, const SB8 &sb, void *c,
+                                      int M, int N, int K, const WR &wrt,
+                                      long ldc, bool c_f32, hipStream_t s,
+                                      int splits);
+
 template <class SA, class SB, class WR>
 static hipError_t launch_pipe_mix_wr(const SA &sa, const SB &sb, void *c,
                                      int M, int N, int K, const WR &wrt,
                                      long ldc, bool c_f32, hipStream_t s,
                                      int splits = 1) {
+  // MPIAMD_PIPE8: route every non-STATS writer through the 8-wave kernel
+  // (the band-slab BnStatsWriter assumes the 2x2 wave structure)
+  if constexpr (!WR::STATS) {
+    if (use_pipe8())
+      return launch_pipe_mix8_wr(typename Pipe8Of<SA>::type{sa.s},
+                                 typename Pipe8Of<SB>::type{sb.s}, c, M, N,
+                                 K, wrt, ldc, c_f32, s, splits);
+  }
   const uint16_t *zeros = pm_zeros_page();
   if (!zeros) return hipErrorOutOfMemory;
   int tiles_m = (M + PM_BM - 1) / PM_BM, tiles_n = (N + PM_BM - 1) / PM_BM;
@@ -762,7 +783,7 @@ template <class SA8, class SB8, class WR>
 static hipError_t launch_pipe_mix8_wr(const SA8 &sa, const SB8 &sb, void *c,
                                       int M, int N, int K, const WR &wrt,
                                       long ldc, bool c_f32, hipStream_t s,
-                                      int splits = 1) {
+                                      int splits) {
   const uint16_t *zeros = pm_zeros_page();
   if (!zeros) return hipErrorOutOfMemory;
   int tiles_m = (M + PM_BM - 1) / PM_BM, tiles_n = (N + PM_BM - 1) / PM_BM;
