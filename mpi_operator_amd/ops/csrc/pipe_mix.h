@@ -771,10 +771,15 @@ __global__ __launch_bounds__(512) void pipe_mix8_k(
 }
 #undef PM8_IMG
 
+// Default ON: same-box A/B with the 8-wave kernel on every pipe route —
+// BERT-Large bs32 1421 -> 1468 seq/s, bs8 607 -> 684, ResNet101
+// 3560 -> 3697 img/s. The extra waves (16/CU vs 8 at the same 64 KiB LDS
+// footprint) buy more than the duplicated B-frag LDS reads cost.
+// MPIAMD_PIPE8=0 reverts to the 4-wave kernel.
 static inline bool use_pipe8() {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_PIPE8");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   return on;
 }
