@@ -213,6 +213,158 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
 }
 #undef P256_IMG
 
+// ======================================================================
+// 16-wave (1024-thread) variant of the 256² pipeline: identical tile,
+// LDS images, 4-phase schedule and swizzle, but 4(M)×4(N) waves of
+// 64×64 each — 16 waves/CU instead of 8 at the same 128 KiB footprint
+// (the 512-thread form leaves half the wave slots empty at 1 block/CU).
+// Per-thread glds counts HALVE (1 granule per stage): every counted
+// vmcnt is half the 8-wave kernel's. A/B gate: MPIAMD_P256X16.
+// ======================================================================
+template <bool C_F32, bool BIAS, int SWZ = 1, int SP = 1,
+          class WR = LinearWriter, bool USE_WR = false>
+__global__ __launch_bounds__(1024) void pipe256x16_gemm_k(
+    const uint16_t *__restrict__ a, long lda, const uint16_t *__restrict__ b,
+    long ldb, void *__restrict__ cptr, int M, int N, int K, long ldc,
+    int tiles_n, int xcd_cpx, const float *__restrict__ bias,
+    WR wrt = WR{}) {
+  int tile = blockIdx.x;
+  if (xcd_cpx) tile = (tile & 7) * xcd_cpx + (tile >> 3);
+  int tm = tile / tiles_n, tn = tile % tiles_n;
+  int row0 = tm * 256, col0 = tn * 256;
+  int tid = threadIdx.x;
+  int lane = tid & 63, wave = tid >> 6;
+  int wr = wave >> 2, wc = wave & 3; // 4x4 wave grid, 64x64 C per wave
+
+  constexpr int HSZ = 256 * 4;
+  __shared__ ushort8 lds[8 * HSZ]; // [buf][op][kh] = 128 KiB
+#define P256X_IMG(buf, op, kh) (lds + (((buf) * 2 + (op)) * 2 + (kh)) * HSZ)
+
+  if (USE_WR) {
+    if constexpr (WR::STATS) wrt.reset();
+  }
+  float16v acc[2][2] = {};
+  int nk = K / 64;
+
+  const uint16_t *ap = a + (long)row0 * lda;
+  const uint16_t *bp = b + (long)col0 * ldb;
+  auto stage = [&](int t, int op, int kh) { // 1 glds per thread
+    const uint16_t *p = op ? bp : ap;
+    long ld = op ? ldb : lda;
+    ushort8 *img = P256X_IMG(t & 1, op, kh);
+    int kb = t * 64 + kh * 32;
+    int r = tid >> 2;
+    int k = kb + p256_swz<SWZ>(tid & 3, r) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void *)(p + (long)r * ld + k),
+        (__attribute__((address_space(3))) void *)(img + tid), 16, 0, 0);
+  };
+
+  auto read_a = [&](bf16x8 af[2], int kh, int mh, int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int r = wr * 64 + mh * 32 + (lane & 31);
+      int q = kk * 2 + (lane >> 5);
+      af[kk] = us8_to_bf8v(P256X_IMG(buf, 0, kh)[r * 4 + p256_swz<SWZ>(q, r)]);
+    }
+  };
+  auto read_b = [&](bf16x8 bf_[2][2], int kh, int buf) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        int r = wc * 64 + ni * 32 + (lane & 31);
+        int q = kk * 2 + (lane >> 5);
+        bf_[ni][kk] =
+            us8_to_bf8v(P256X_IMG(buf, 1, kh)[r * 4 + p256_swz<SWZ>(q, r)]);
+      }
+  };
+
+  if (nk > 0) {
+    stage(0, 0, 0);
+    stage(0, 1, 0);
+    stage(0, 0, 1);
+    stage(0, 1, 1);
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kh0 proven (2 ops)
+    __builtin_amdgcn_s_barrier();
+  }
+
+  if (SP == 1 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 512)
+    __builtin_amdgcn_s_setprio(1);
+
+  bf16x8 af[2], bf_[2][2];
+  for (int t = 0; t < nk; ++t) {
+    int buf = t & 1;
+    bool pre = t + 1 < nk;
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      int kh = ph >> 1, mh = ph & 1;
+      if (ph == 1 || ph == 3) {
+        if (pre) // 1 op per stage now: one fresh stage allowed in flight
+          asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      if (mh == 0) read_b(bf_, kh, buf);
+      read_a(af, kh, mh, buf);
+      if (pre) stage(t + 1, ph & 1, ph >> 1);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mh][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[kk], bf_[ni][kk], acc[mh][ni], 0, 0, 0);
+    }
+  }
+  if (SP == 1) __builtin_amdgcn_s_setprio(0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+  for (int mh = 0; mh < 2; ++mh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      long row = row0 + wr * 64 + mh * 32 + (r & 3) + 8 * (r >> 2) +
+                 4 * (lane >> 5);
+      if (USE_WR) {
+        typename WR::RowCtx rc = wrt.row_ctx((int)row);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+          if (C_F32)
+            wrt.store_f32((float *)cptr, rc, col, acc[mh][ni][r]);
+          else
+            wrt.store_bf16((uint16_t *)cptr, rc, col, acc[mh][ni][r]);
+        }
+        continue;
+      }
+      long base = row * ldc;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int col = col0 + wc * 64 + ni * 32 + (lane & 31);
+        float v = BIAS ? acc[mh][ni][r] + bias[col] : acc[mh][ni][r];
+        if (C_F32)
+          ((float *)cptr)[base + col] = v;
+        else
+          ((uint16_t *)cptr)[base + col] = f2bf(v);
+      }
+    }
+  }
+  if (USE_WR) {
+    if constexpr (WR::STATS) wrt.flush(lane);
+  }
+}
+#undef P256X_IMG
+
+static inline bool use_p256x16() {
+  static const bool on = [] {
+    const char *e = getenv("MPIAMD_P256X16");
+    return e && e[0] == '1';
+  }();
+  return on;
+}
+
 // Writer-parameterized launch (fused epilogues: GELU+bias, BN stats, ...)
 template <class WR, class LA, class LB>
 static hipError_t launch_pipe256_wr(const LA &la, const LB &lb, void *c,
@@ -222,6 +374,17 @@ static hipError_t launch_pipe256_wr(const LA &la, const LB &lb, void *c,
   int nwg = tiles_m * tiles_n;
   int cpx = (nwg % 8 == 0 && nwg >= 32) ? nwg / 8 : 0;
   constexpr int SWZ = 1, NPB = 1, SP = 1;
+  if (use_p256x16()) {
+    if (c_f32)
+      pipe256x16_gemm_k<true, false, SWZ, SP, WR, true><<<nwg, 1024, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, nullptr,
+          wrt);
+    else
+      pipe256x16_gemm_k<false, false, SWZ, SP, WR, true><<<nwg, 1024, 0, s>>>(
+          la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, nullptr,
+          wrt);
+    return hipGetLastError();
+  }
   if (c_f32)
     pipe256_gemm_k<true, false, SWZ, NPB, SP, WR, true><<<nwg, 512, 0, s>>>(
         la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, nullptr, wrt);
@@ -242,6 +405,24 @@ static hipError_t launch_pipe256(const LA &la, const LB &lb, void *c, int M,
   // swz1 + 1 barrier/phase + static setprio: 1146 TF @4k³ vs 789 for the
   // round-1 gemm256 (+44%); +17-42% on the real model shapes
   constexpr int SWZ = 1, NPB = 1, SP = 1;
+  if (use_p256x16()) {
+    if (c_f32) {
+      if (bias)
+        pipe256x16_gemm_k<true, true, SWZ, SP><<<nwg, 1024, 0, s>>>(
+            la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+      else
+        pipe256x16_gemm_k<true, false, SWZ, SP><<<nwg, 1024, 0, s>>>(
+            la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+    } else {
+      if (bias)
+        pipe256x16_gemm_k<false, true, SWZ, SP><<<nwg, 1024, 0, s>>>(
+            la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+      else
+        pipe256x16_gemm_k<false, false, SWZ, SP><<<nwg, 1024, 0, s>>>(
+            la.p, la.ld, lb.p, lb.ld, c, M, N, K, ldc, tiles_n, cpx, bias);
+    }
+    return hipGetLastError();
+  }
   if (c_f32) {
     if (bias)
       pipe256_gemm_k<true, true, SWZ, NPB, SP><<<nwg, 512, 0, s>>>(

@@ -62,16 +62,16 @@ static inline bool use_pipemix() {
   return on;
 }
 
-// The conv GATHER routes measured NET-NEGATIVE on the pipeline (3296 vs
-// 3390 img/s same-box A/B): their ptr16 re-derives the pixel/tap split
-// per stage (two int64 divmods per granule) where the mix stagers hoist
-// it — the glds win doesn't cover the extra VALU at 2 blocks/CU. Default
-// OFF (mix path); MPIAMD_PIPEGATHER=1 re-enables for A/B after the
-// carry-chain SrcMap optimization.
+// The conv GATHER routes measured net-negative on the 4-WAVE pipeline
+// (3296 vs 3390 same-box), but the verdict REVERSED under the 8-wave
+// kernel: 3742 -> 3820 img/s ResNet101 / 6508 ResNet50 same-box — the
+// extra VALU of the gather ptr16 hides behind the doubled wave count.
+// Default ON; MPIAMD_PIPEGATHER=0 reverts to the register-staged mix
+// gathers.
 static inline bool use_pipegather() {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_PIPEGATHER");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   return use_pipemix() && on;
 }
