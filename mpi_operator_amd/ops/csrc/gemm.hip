@@ -211,14 +211,15 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
 // `partial` reduced into bf16 C. Same splits heuristic as the dw path.
 extern "C" int gemm_tn_tn_splits(int M, int N, int K); // defined below
 
-// dx split-K: measured net-negative in round 1 (the runtime-loop
-// splitk_reduce was 73% wave-parked then), but AFTER the templated/unrolled
-// reduce rewrite it wins same-box at BERT-Large bs32: 1257 -> 1274 seq/s,
-// repeatable. Default ON; MPIAMD_DX_SK=0 disables.
+// dx split-K has flipped verdict TWICE with the baseline kernel: lost in
+// round 1 (73%-parked runtime reduce), won after the templated reduce
+// (1257 -> 1274), and LOSES again under the 8-wave pipeline (1542 vs
+// 1553 seq/s same-box — pipe8 fills the chip without the slab+reduce
+// tax). Default OFF; MPIAMD_DX_SK=1 re-enables.
 extern "C" int gemm_nt_tn_splits(int M, int N, int K) {
   static const bool on = [] {
     const char *e = getenv("MPIAMD_DX_SK");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   return on ? gemm_tn_tn_splits(M, N, K) : 1;
 }
