@@ -237,7 +237,15 @@ static Tensor conv2d_wgrad(const Tensor &x, const Tensor &dy, int64_t R,
   // per split so each block amortizes its pipeline prologue; a flat 1024
   // cap overshot mid shapes (b2 1x1 dw: 256 splits of nk=3 ran at 94 TF
   // vs 167 at 64 splits)
-  int splits = std::max(512 / tiles, 1);
+  // target-block budget was tuned on the 4-wave pipeline (2 WG/CU); the
+  // 8-wave kernel fills at half the workgroups — MPIAMD_WGRAD_BUDGET
+  // overrides for A/B (default 512)
+  static const int budget = [] {
+    const char *e = getenv("MPIAMD_WGRAD_BUDGET");
+    int v = e ? atoi(e) : 0;
+    return (v >= 64 && v <= 2048) ? v : 512;
+  }();
+  int splits = std::max(budget / tiles, 1);
   splits = std::min(splits, std::max(nk / 8, 1));
   splits = std::min(splits, 256);
   // multiple of 8 so the split-major grid keeps a panel's N-tile sharers
