@@ -211,17 +211,21 @@ extern "C" hipError_t gemm_nt_tn(const void *a, const void *b, void *c, int M,
 // `partial` reduced into bf16 C. Same splits heuristic as the dw path.
 extern "C" int gemm_tn_tn_splits(int M, int N, int K); // defined below
 
-// dx split-K has flipped verdict TWICE with the baseline kernel: lost in
-// round 1 (73%-parked runtime reduce), won after the templated reduce
-// (1257 -> 1274), and LOSES again under the 8-wave pipeline (1542 vs
-// 1553 seq/s same-box — pipe8 fills the chip without the slab+reduce
-// tax). Default OFF; MPIAMD_DX_SK=1 re-enables.
+// dx split-K, SHAPE-aware after three verdict flips: under the 8-wave
+// pipeline a 256-workgroup dx grid fills the chip and the slab+reduce tax
+// loses (bs32: 1542 with splits vs 1553 without), but SMALL grids still
+// need it badly (bs8's 64-wg dx: 690 with splits vs 614 without; so does
+// bert-base's 192-wg grid). Split only when tiles < 256.
+// MPIAMD_DX_SK: 0 = never split, 1 = always use the dw heuristic.
 extern "C" int gemm_nt_tn_splits(int M, int N, int K) {
-  static const bool on = [] {
+  static const int mode = [] {
     const char *e = getenv("MPIAMD_DX_SK");
-    return e && e[0] == '1';
+    return e ? (e[0] == '1' ? 1 : 0) : -1;
   }();
-  return on ? gemm_tn_tn_splits(M, N, K) : 1;
+  if (mode == 0) return 1;
+  long tiles = ((M + 127) / 128) * ((N + 127) / 128);
+  if (mode != 1 && tiles >= 256) return 1;
+  return gemm_tn_tn_splits(M, N, K);
 }
 
 extern "C" hipError_t gemm_nt_tn_sk(const void *a, const void *b,
