@@ -126,6 +126,74 @@ __global__ void ln_bwd_dx_k(const ushort8 *__restrict__ dy,
   }
 }
 
+// 2-row-interleaved dx variant (N <= 1024 so both rows' octets fit the
+// register cache): each wave owns rows (r, r + stride) with fully
+// independent load/reduce chains interleaved in one body — the 1-row
+// kernel ran one 2-octet load pair per wave between shfl-reduction
+// chains and measured ~3.4 TB/s (latency-, not bandwidth-bound).
+// MPIAMD_LN2 gates the route (A/B).
+__global__ void ln_bwd_dx2_k(const ushort8 *__restrict__ dy,
+                             const ushort8 *__restrict__ x,
+                             const float *__restrict__ gamma,
+                             const float *__restrict__ mean,
+                             const float *__restrict__ rstd,
+                             ushort8 *__restrict__ dx, long M, int C8) {
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  int waves = blockDim.x >> 6;
+  int N = C8 * 8;
+  float inv_n = 1.f / (float)N;
+  long stride = (long)gridDim.x * waves; // row pair: (r, r + stride)
+  for (long rA = (long)blockIdx.x * waves + wave; rA < M; rA += 2 * stride) {
+    long rB = rA + stride;
+    bool hasB = rB < M;
+    const ushort8 *xA = x + rA * C8, *dA = dy + rA * C8;
+    const ushort8 *xB = x + (hasB ? rB : rA) * C8;
+    const ushort8 *dB = dy + (hasB ? rB : rA) * C8;
+    float muA = mean[rA], rsA = rstd[rA];
+    float muB = mean[hasB ? rB : rA], rsB = rstd[hasB ? rB : rA];
+    float s1A = 0.f, s2A = 0.f, s1B = 0.f, s2B = 0.f;
+    ushort8 cxA[2], cdA[2], cxB[2], cdB[2];
+    for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
+      ushort8 vxA = xA[c], vdA = dA[c], vxB = xB[c], vdB = dB[c];
+      if (ci < 2) {
+        cxA[ci] = vxA; cdA[ci] = vdA; cxB[ci] = vxB; cdB[ci] = vdB;
+      }
+      float fxA[8], fdA[8], fxB[8], fdB[8];
+      bf8_to_f8(vxA, fxA); bf8_to_f8(vdA, fdA);
+      bf8_to_f8(vxB, fxB); bf8_to_f8(vdB, fdB);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = gamma[c * 8 + j];
+        float gdA = g * fdA[j], gdB = g * fdB[j];
+        s1A += gdA; s2A += gdA * ((fxA[j] - muA) * rsA);
+        s1B += gdB; s2B += gdB * ((fxB[j] - muB) * rsB);
+      }
+    }
+    s1A = wave_sum(s1A) * inv_n; s2A = wave_sum(s2A) * inv_n;
+    s1B = wave_sum(s1B) * inv_n; s2B = wave_sum(s2B) * inv_n;
+    ushort8 *oA = dx + rA * C8;
+    ushort8 *oB = dx + (hasB ? rB : rA) * C8;
+    for (int c = lane; c < C8; c += 64) {
+      int ci = (c - lane) >> 6;
+      float fxA[8], fdA[8], fxB[8], fdB[8];
+      bf8_to_f8(ci < 2 ? cxA[ci] : xA[c], fxA);
+      bf8_to_f8(ci < 2 ? cdA[ci] : dA[c], fdA);
+      bf8_to_f8(ci < 2 ? cxB[ci] : xB[c], fxB);
+      bf8_to_f8(ci < 2 ? cdB[ci] : dB[c], fdB);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = gamma[c * 8 + j];
+        fdA[j] = rsA * (g * fdA[j] - s1A - ((fxA[j] - muA) * rsA) * s2A);
+        fdB[j] = rsB * (g * fdB[j] - s1B - ((fxB[j] - muB) * rsB) * s2B);
+      }
+      oA[c] = f8_to_bf8(fdA);
+      if (hasB) oB[c] = f8_to_bf8(fdB);
+    }
+  }
+}
+
 // dgamma/dbeta partials, bn_partials-style: thread owns a channel octet,
 // strides rows (coalesced 16 B/lane), per-row mean/rstd scalar loads;
 // [grid][2][N] fp32 slabs reduced by splitk_reduce.
@@ -338,7 +406,19 @@ extern "C" hipError_t ln_bwd(const void *dy, const void *x, const float *gamma,
                              int N, int *grid_out, hipStream_t s) {
   if (N % 8 || N > 2048) return hipErrorInvalidValue;
   int C8 = N / 8;
-  if (ln_spec() && N == 1024)
+  static const bool ln2 = [] { // MPIAMD_LN2: 2-row-interleaved dx (A/B)
+    const char *e = getenv("MPIAMD_LN2");
+    return e && e[0] == '1';
+  }();
+  if (ln2 && N <= 1024) {
+    long waves2 = (M + 1) / 2;
+    int grid2 = (int)((waves2 + 3) / 4);
+    if (grid2 > 1024) grid2 = 1024;
+    if (grid2 < 1) grid2 = 1;
+    ln_bwd_dx2_k<<<grid2, 256, 0, s>>>((const ushort8 *)dy,
+                                       (const ushort8 *)x, gamma, mean, rstd,
+                                       (ushort8 *)dx, M, C8);
+  } else if (ln_spec() && N == 1024)
     ln_bwd_dx_k<128><<<ln_grid(M, 4), 256, 0, s>>>(
         (const ushort8 *)dy, (const ushort8 *)x, gamma, mean, rstd,
         (ushort8 *)dx, M, C8);
