@@ -416,8 +416,29 @@ static hipError_t conv_dgrad_s2(const void *dy, const void *w, void *dx,
       lb.w = (const uint16_t *)w;
       lb.C = C; lb.Q = Kout; lb.K = K; lb.RSC = R * S * C; lb.S = S;
       lb.nth = sa.nth; lb.ntw = sa.ntw;
+      static const bool pipes2 = [] { // MPIAMD_PIPES2=0: A/B lever
+        const char *env = getenv("MPIAMD_PIPES2");
+        return !(env && env[0] == '0');
+      }();
       hipError_t e;
-      if (n_live == 1) {
+      if (use_pipegather() && pipes2) {
+        // pipe route (8-wave under MPIAMD_PIPE8): same parity GEMM with
+        // the glds/tr staging; Stride2*Writer epilogues are pipe-generic
+        NtPipe<DgradS2Src> sa2{{(const uint16_t *)dy, HO, WO, Kout, W2, H2,
+                                K, M, sa.dh0, sa.dw0, sa.nth, sa.ntw}};
+        TnPipe<DgradWS2TnSrc> sb2{{(const uint16_t *)w, C, Kout, K,
+                                   R * S * C, S, r_first, s_first, sa.nth,
+                                   sa.ntw}};
+        if (n_live == 1) {
+          Stride2ZeroWriter wrt{W2, H2, ph, pw, W, H, C};
+          e = launch_pipe_mix_wr(sa2, sb2, dx, (int)M, C, K, wrt, C, false,
+                                 strm);
+        } else {
+          Stride2Writer wrt{W2, H2, ph, pw, W, H, C};
+          e = launch_pipe_mix_wr(sa2, sb2, dx, (int)M, C, K, wrt, C, false,
+                                 strm);
+        }
+      } else if (n_live == 1) {
         Stride2ZeroWriter wrt{W2, H2, ph, pw, W, H, C};
         e = launch_mix_gemm_wr(sa, TnStage<DgradWS2Tn>{lb}, dx, (int)M, C, K,
                                wrt, C, false, strm);

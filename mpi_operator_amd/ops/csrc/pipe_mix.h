@@ -248,6 +248,83 @@ struct DgradWTnSrc {
   }
 };
 
+// stride-2 dgrad parity gather (NT: rows = parity-subgrid pixels, k =
+// (tap, q) with q fastest; tap shifts affine in the tap index). Ported
+// from conv.hip's DgradS2Stage to the SrcMap contract; Q is pow2 on every
+// ResNet shape (shift/mask), ntw <= 4 (16-bit reciprocal).
+struct DgradS2Src {
+  const uint16_t *dy;
+  int HO, WO, Q, W2, H2, K;
+  long M;
+  int dh0, dw0, nth, ntw;
+  long nbase_[2];
+  int hb_[2], wb_[2], koff_[2];
+  bool ok_[2];
+  int qsh_, ntwrecip_;
+  DEV_INLINE void init(int i, int rc, int koff) {
+    long m = rc;
+    ok_[i] = m < M;
+    int w_ = (int)(m % W2);
+    long t = m / W2;
+    int h_ = (int)(t % H2);
+    int n = (int)(t / H2);
+    nbase_[i] = (long)n * HO * WO;
+    hb_[i] = h_;
+    wb_[i] = w_;
+    koff_[i] = koff;
+    qsh_ = ilog2_if_pow2(Q);
+    ntwrecip_ = (1 << 16) / ntw + 1;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    int k = kb + koff_[i];
+    int q, ti;
+    if (qsh_ >= 0) {
+      q = k & (Q - 1);
+      ti = k >> qsh_;
+    } else {
+      q = k % Q;
+      ti = k / Q;
+    }
+    int ih = div_small(ti, ntwrecip_);
+    int iw = ti - ih * ntw;
+    int ho = hb_[i] + dh0 - ih, wo = wb_[i] + dw0 - iw;
+    if (k < K && ok_[i] && (unsigned)ho < (unsigned)HO &&
+        (unsigned)wo < (unsigned)WO)
+      return dy + (nbase_[i] + (long)ho * WO + wo) * Q + q;
+    return nullptr;
+  }
+};
+
+// stride-2 dgrad weight view (TN: cols = C, k = (tap, q)):
+// element (c, k) = w[q][((r0+2*ih)*S + s0+2*iw)*C + c]
+struct DgradWS2TnSrc {
+  const uint16_t *w;
+  int C, Q, K, RSC, S, r0, s0, nth, ntw;
+  int col_[2], koff_[2];
+  int qsh_, ntwrecip_;
+  DEV_INLINE void init(int i, int rc, int koff) {
+    col_[i] = rc;
+    koff_[i] = koff;
+    qsh_ = ilog2_if_pow2(Q);
+    ntwrecip_ = (1 << 16) / ntw + 1;
+  }
+  DEV_INLINE const uint16_t *ptr16(int i, int kb) const {
+    int k = kb + koff_[i];
+    if (k >= K || col_[i] >= C) return nullptr;
+    int q, ti;
+    if (qsh_ >= 0) {
+      q = k & (Q - 1);
+      ti = k >> qsh_;
+    } else {
+      q = k % Q;
+      ti = k / Q;
+    }
+    int ih = div_small(ti, ntwrecip_);
+    int iw = ti - ih * ntw;
+    return w + (long)q * RSC + ((r0 + 2 * ih) * S + s0 + 2 * iw) * C + col_[i];
+  }
+};
+
 // wgrad implicit-im2col gather (TN: k = output pixel m, cols = (r,s,c)).
 // Column decomposition (rr, ss, c) hoisted. The pixel decomposition rides
 // a CARRY CHAIN: pipe_mix stages k-halves in strictly increasing kb order
