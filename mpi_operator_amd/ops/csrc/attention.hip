@@ -85,8 +85,12 @@ __global__ __launch_bounds__(256) void attn_fwd_k(
   }
   __syncthreads();
 
-  // Q fragments straight from global: lane holds row (l&31), k-octet (l>>5)
-  int qrow0 = wave * 32;
+  // Q fragments straight from global: lane holds row (l&31), k-octet (l>>5).
+  // blockIdx.y selects the 128-row q-chunk: a block's 4 waves cover 128
+  // q-rows, so S in (128, 256] launches TWO chunks per (b, h) — the first
+  // S_MAX=256 cut covered only rows 0..127 and left ctx rows 128+ as
+  // uninitialized memory (NaN loss at seq 256; caught by the edge sweep).
+  int qrow0 = blockIdx.y * 128 + wave * 32;
   if (qrow0 >= S) return; // short sequences: idle waves (after the barrier)
   const uint16_t *qp = base + h * D;
   bf16x8a qf[4]; // 4 k-steps of 16 over D=64
@@ -228,11 +232,11 @@ extern "C" hipError_t attn_fwd_launch(const void *qkv, void *ctx, void *probs,
   // LDS: S_MAX=128 → 64 KiB (2 blocks/CU); S_MAX=256 → 128 KiB (1/CU).
   // S > 256 needs a tiled/online-softmax variant (torch fallback upstream).
   if (S <= 128)
-    attn_fwd_k<128><<<B * H, 256, 0, strm>>>(
+    attn_fwd_k<128><<<dim3(B * H, 1), 256, 0, strm>>>(
         (const uint16_t *)qkv, (uint16_t *)ctx, (uint16_t *)probs, mask, B, S,
         H, scale);
-  else if (S <= 256)
-    attn_fwd_k<256><<<B * H, 256, 0, strm>>>(
+  else if (S <= 256) // two 128-row q-chunks per (b, h)
+    attn_fwd_k<256><<<dim3(B * H, 2), 256, 0, strm>>>(
         (const uint16_t *)qkv, (uint16_t *)ctx, (uint16_t *)probs, mask, B, S,
         H, scale);
   else

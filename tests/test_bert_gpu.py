@@ -239,6 +239,29 @@ def test_fused_attention_matches_torch():
     assert g_err < 0.08 * g_scale + 1e-4, (g_err, g_scale)
 
 
+@pytest.mark.parametrize("S", [160, 256])
+def test_fused_attention_seq_over_128(S):
+    """S in (128, 256]: the 256-capacity template runs TWO 128-row
+    q-chunks per (b, h) — the first cut left rows 128+ uninitialized
+    (NaN at seq 256, caught by an end-to-end edge sweep)."""
+    from mpi_operator_amd.ops import functional as Fx
+    torch.manual_seed(19)
+    B, H, D = 2, 4, 64
+    qkv = (torch.randn(B, S, 3, H, D, device="cuda") * 0.5).to(torch.bfloat16)
+    a = qkv.clone().requires_grad_(True)
+    out = Fx.attention(a, H, 1.0 / D ** 0.5)
+    out.float().square().mean().backward()
+    r = qkv.float().clone().requires_grad_(True)
+    q, k, v = (r[:, :, i].transpose(1, 2) for i in range(3))
+    p = torch.softmax(torch.matmul(q, k.transpose(-1, -2)) / D ** 0.5, dim=-1)
+    ref = torch.matmul(p, v).transpose(1, 2).reshape(B, S, H * D)
+    ref.square().mean().backward()
+    assert (out.float() - ref).abs().max().item() < 0.02
+    g_err = (a.grad.float() - r.grad).abs().max().item()
+    g_scale = r.grad.abs().max().item() + 1e-6
+    assert g_err < 0.08 * g_scale + 1e-4, (g_err, g_scale)
+
+
 def test_fused_attention_ragged_seq():
     """S not a multiple of 32: padded key columns must not leak."""
     from mpi_operator_amd.ops import functional as Fx
