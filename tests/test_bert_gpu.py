@@ -383,3 +383,17 @@ def test_composite_layer_matches_decomposed():
         num = (g1[n] - g0[n]).norm().item()
         den = g0[n].norm().item() + 1e-30
         assert num / den < 0.02, (n, num / den)
+
+
+def test_bert_forward_with_attention_mask():
+    """attn_mask falls back to the eager attention path — guard the shape
+    plumbing end-to-end (the fused kernel path requires mask is None)."""
+    from mpi_operator_amd.models.bert import bert_base, to_mi355x_bert
+    torch.manual_seed(23)
+    m = to_mi355x_bert(bert_base(), "cuda")
+    ids = torch.randint(0, m.cfg.vocab_size, (2, 48), device="cuda")
+    mask = torch.ones(2, 48, device="cuda")
+    mask[:, 40:] = 0  # padded tail
+    mlm_logits, nsp_logits = m(ids, attn_mask=mask)
+    assert torch.isfinite(mlm_logits.float()).all()
+    assert torch.isfinite(nsp_logits.float()).all()
