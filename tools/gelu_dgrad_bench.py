@@ -1,4 +1,8 @@
-import torch, time
+import os, sys, time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from mpi_operator_amd.ops import hip_ext
 ext = hip_ext()
 M, N, K = 4096, 4096, 1024
