@@ -285,3 +285,30 @@ def test_pg_min_resources_equal_priority_launcher_first():
     got = _pg_ctrl().calculate_pg_min_resources(2, job)
     assert got["cpu"] == "12"
     assert got["memory"] == "33Gi"
+
+
+# ---- TestResumeMPIJobWithExistingLauncher (:1207): resume must UNSUSPEND
+# the existing launcher Job in place (not delete/recreate) and recreate
+# the worker pods ----
+def test_resume_with_existing_launcher_unsuspends_in_place():
+    client, ctrl = make_controller()
+    job = make_job(workers=2)
+    job["spec"]["runPolicy"] = {"suspend": True}
+    client.seed(MPIJOBS, job)
+    ctrl.sync("default", "test")
+    launcher = client.jobs.get("default", "test-launcher")
+    assert launcher["spec"]["suspend"] is True
+    launcher_uid = launcher["metadata"].get("uid", id(launcher))
+
+    # resume: same launcher object gets suspend=False; workers appear
+    j = client.mpijobs.get("default", "test")
+    j["spec"]["runPolicy"]["suspend"] = False
+    client.mpijobs.update("default", j)
+    ctrl.sync("default", "test")
+    launcher2 = client.jobs.get("default", "test-launcher")
+    assert launcher2["spec"]["suspend"] is False
+    assert launcher2["metadata"].get("uid", id(launcher2)) == launcher_uid
+    workers = client.pods.list("default", builders.worker_selector("test"))
+    assert len(workers) == 2
+    st = client.mpijobs.get("default", "test")["status"]
+    assert t.get_condition(st, c.JOB_SUSPENDED)["status"] == "False"
