@@ -312,3 +312,30 @@ def test_resume_with_existing_launcher_unsuspends_in_place():
     assert len(workers) == 2
     st = client.mpijobs.get("default", "test")["status"]
     assert t.get_condition(st, c.JOB_SUSPENDED)["status"] == "False"
+
+
+# ---- TestLauncherSucceededWithRunningPod (:681): the batch Job's Complete
+# condition is AUTHORITATIVE — a launcher pod still observed Running must
+# not keep the job Active or block the Succeeded transition ----
+def test_launcher_succeeded_while_pod_still_running():
+    from tests.test_controller import _complete_launcher
+    client, ctrl = make_controller()
+    client.seed(MPIJOBS, make_job(workers=1))
+    ctrl.sync("default", "test")
+    launcher = client.jobs.get("default", "test-launcher")
+    launcher["spec"]["selector"] = {"matchLabels": {"controller-uid": "lr1"}}
+    client.jobs.update("default", launcher)
+    # lagging pod observation: still Running after the Job completed
+    client.pods.create("default", {
+        "metadata": {"name": "test-launcher-pod", "namespace": "default",
+                     "labels": {"controller-uid": "lr1"}},
+        "status": {"phase": "Running"}})
+    _complete_launcher(client, succeeded=True)
+    ctrl.sync("default", "test")
+    stt = client.mpijobs.get("default", "test")["status"]
+    assert t.has_condition_true(stt, c.JOB_SUCCEEDED)
+    running = t.get_condition(stt, c.JOB_RUNNING)
+    assert running is None or running["status"] == "False"
+    ls = stt["replicaStatuses"]["Launcher"]
+    assert ls["succeeded"] == 1
+    assert ls.get("active", 0) == 0
