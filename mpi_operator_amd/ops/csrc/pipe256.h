@@ -218,6 +218,8 @@ __global__ __launch_bounds__(512) void pipe256_gemm_k(
 // LDS images, 4-phase schedule and swizzle, but 4(M)×4(N) waves of
 // 64×64 each — 16 waves/CU instead of 8 at the same 128 KiB footprint
 // (the 512-thread form leaves half the wave slots empty at 1 block/CU).
+// Reference role: the forward cuBLAS GEMMs of the tf_cnn_benchmarks /
+// Horovod images (reference README.md:127-130), MI355X-native.
 // Per-thread glds counts HALVE (1 granule per stage): every counted
 // vmcnt is half the 8-wave kernel's. A/B gate: MPIAMD_P256X16.
 // ======================================================================

@@ -683,6 +683,9 @@ static hipError_t launch_pipe_mix_wr(const SA &sa, const SB &sb, void *c,
 // empty at its 64 KiB LDS footprint; PMC: 43-48% parked, 20-30% stall on
 // the dw/dx pool). Cost: B-frags are read by 2 waves each (LDS
 // read:MFMA 1.5 vs 1.0). A/B gated at the launchers (MPIAMD_PIPE8).
+// Reference role: same as pipe_mix_k — the cuDNN/cuBLAS backward GEMMs
+// of the tf_cnn_benchmarks image (reference README.md:127-130), measured
+// +3-4% end-to-end over the 4-wave form on both models.
 // Per-thread VM-op counts HALVE (1 granule per stage instead of 2), so
 // every counted vmcnt is half the 4-wave kernel's.
 // ======================================================================
