@@ -263,3 +263,26 @@ def test_simple_cnn_train_step_gpu():
     loss.backward()
     opt.step()
     assert float(loss) == float(loss)
+
+
+@pytest.mark.parametrize("wd,nesterov", [(0.01, False), (0.01, True), (0.0, True)])
+def test_fused_sgd_kernel_wd_nesterov_matches_torch(wd, nesterov):
+    """The sgd_step_k wd/nesterov algebra vs torch.optim.SGD on fp32 CUDA
+    params (the loss-descent tests only exercise the defaults)."""
+    from mpi_operator_amd.optim import FusedSGD
+    torch.manual_seed(44)
+    a = torch.nn.Linear(64, 32).to("cuda")
+    b = torch.nn.Linear(64, 32).to("cuda")
+    b.load_state_dict(a.state_dict())
+    oa = FusedSGD(a.parameters(), lr=0.1, momentum=0.9, weight_decay=wd,
+                  nesterov=nesterov)
+    ob = torch.optim.SGD(b.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=wd, nesterov=nesterov)
+    x = torch.randn(16, 64, device="cuda")
+    for _ in range(5):
+        oa.zero_grad(); ob.zero_grad()
+        a(x).pow(2).mean().backward()
+        b(x).pow(2).mean().backward()
+        oa.step(); ob.step()
+    for pa, pb in zip(a.parameters(), b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-5), (pa - pb).abs().max()
