@@ -380,9 +380,14 @@ static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
 }
 
 static int bn_apply_grid(long M, int C8) {
+  static const long cap = [] { // MPIAMD_BN_APPLY_GRID: A/B lever
+    const char *e = getenv("MPIAMD_BN_APPLY_GRID");
+    long v = e ? atol(e) : 0;
+    return (v >= 64 && v <= 16384) ? v : 4096L;
+  }();
   int rpb = 256 / C8;
   long g = (M + rpb - 1) / rpb;
-  return (int)(g > 4096 ? 4096 : (g < 1 ? 1 : g));
+  return (int)(g > cap ? cap : (g < 1 ? 1 : g));
 }
 
 extern "C" hipError_t bn_fwd_train_launch(
