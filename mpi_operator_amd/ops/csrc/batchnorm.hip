@@ -380,10 +380,12 @@ static void bn_geom(long M, int C8, int &grid, int &rows_per_block) {
 }
 
 static int bn_apply_grid(long M, int C8) {
-  static const long cap = [] { // MPIAMD_BN_APPLY_GRID: A/B lever
+  // cap 2048 measured best (3846-3854 vs 3833-3834 img/s at 4096, 3800
+  // at 8192, four-sample same-box sweep); MPIAMD_BN_APPLY_GRID overrides
+  static const long cap = [] {
     const char *e = getenv("MPIAMD_BN_APPLY_GRID");
     long v = e ? atol(e) : 0;
-    return (v >= 64 && v <= 16384) ? v : 4096L;
+    return (v >= 64 && v <= 16384) ? v : 2048L;
   }();
   int rpb = 256 / C8;
   long g = (M + rpb - 1) / rpb;
